@@ -1,0 +1,1109 @@
+// elbencho_amd engine implementation.
+//
+// Independent MI355X-native implementation of the behavior of the reference's
+// LocalWorker I/O engine (/root/reference/source/workers/LocalWorker.cpp):
+// same work partitioning and directory layout contract (fair-share block
+// split: fileModeIterateFilesSeq :3597; per-worker random subrange :3511;
+// dir layout "r{rank}/d{dir}/r{rank}-f{file}" :3097; bench path round-robin
+// (rank+dirIdx)%numPaths :3111), re-designed around io_uring + HIP streams.
+
+#include "engine.h"
+
+#include <fcntl.h>
+#include <linux/fs.h>
+#include <sys/ioctl.h>
+#include <sys/stat.h>
+#include <sys/types.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <cerrno>
+#include <cstdio>
+#include <cstring>
+
+#include "uring.h"
+
+namespace eb {
+
+static constexpr uint64_t INTERRUPT_CHECK_INTERVAL = 128;
+
+// ---------------------------------------------------------------------------
+// integrity checksum pattern (CPU side)
+// u64 at 8-aligned file offset o has little-endian value (o + salt).
+// ---------------------------------------------------------------------------
+
+static inline uint64_t checksumValueAt(uint64_t alignedOff, uint64_t salt)
+{
+    return alignedOff + salt;
+}
+
+void fillChecksumCPU(char* buf, uint64_t len, uint64_t fileOff, uint64_t salt)
+{
+    uint64_t pos = fileOff;
+    uint64_t end = fileOff + len;
+
+    // unaligned head: bytes of the u64 containing fileOff
+    while (pos < end && (pos & 7)) {
+        uint64_t v = checksumValueAt(pos & ~7ULL, salt);
+        buf[pos - fileOff] = (char)((v >> (8 * (pos & 7))) & 0xff);
+        pos++;
+    }
+    // aligned middle
+    while (pos + 8 <= end) {
+        uint64_t v = checksumValueAt(pos, salt);
+        std::memcpy(buf + (pos - fileOff), &v, 8);
+        pos += 8;
+    }
+    // tail
+    while (pos < end) {
+        uint64_t v = checksumValueAt(pos & ~7ULL, salt);
+        buf[pos - fileOff] = (char)((v >> (8 * (pos & 7))) & 0xff);
+        pos++;
+    }
+}
+
+uint64_t verifyChecksumCPU(const char* buf, uint64_t len, uint64_t fileOff, uint64_t salt)
+{
+    uint64_t pos = fileOff;
+    uint64_t end = fileOff + len;
+
+    while (pos < end && (pos & 7)) {
+        uint64_t v = checksumValueAt(pos & ~7ULL, salt);
+        if (buf[pos - fileOff] != (char)((v >> (8 * (pos & 7))) & 0xff)) return pos;
+        pos++;
+    }
+    while (pos + 8 <= end) {
+        uint64_t v = checksumValueAt(pos, salt);
+        uint64_t got;
+        std::memcpy(&got, buf + (pos - fileOff), 8);
+        if (got != v) { // find exact byte
+            for (int b = 0; b < 8; b++)
+                if (((got >> (8 * b)) & 0xff) != ((v >> (8 * b)) & 0xff)) return pos + b;
+        }
+        pos += 8;
+    }
+    while (pos < end) {
+        uint64_t v = checksumValueAt(pos & ~7ULL, salt);
+        if (buf[pos - fileOff] != (char)((v >> (8 * (pos & 7))) & 0xff)) return pos;
+        pos++;
+    }
+    return UINT64_MAX;
+}
+
+// ---------------------------------------------------------------------------
+// small helpers
+// ---------------------------------------------------------------------------
+
+static uint64_t nowUSecSince(std::chrono::steady_clock::time_point t0)
+{
+    return (uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
+               std::chrono::steady_clock::now() - t0)
+        .count();
+}
+
+static void throwErrno(const std::string& what, const std::string& path)
+{
+    throw WorkerError(what + " failed. Path: " + path + "; SysErr: " + strerror(errno));
+}
+
+// mkdir -p for one level chain (benchPath/r{rank}) — parents must exist.
+static void mkdirIgnoreExists(const std::string& path)
+{
+    if (mkdir(path.c_str(), 0777) && errno != EEXIST) throwErrno("mkdir", path);
+}
+
+// ---------------------------------------------------------------------------
+// Worker
+// ---------------------------------------------------------------------------
+
+Worker::Worker(Engine& engine, int localRank)
+    : localRank(localRank), globalRank(engine.cfg.rankOffset + localRank), eng(engine)
+{
+}
+
+Worker::~Worker()
+{
+    if (ownHostBufs)
+        for (auto p : hostBufs) free(p);
+}
+
+void Worker::checkInterrupt()
+{
+    if (eng.interruptFlag.load(std::memory_order_relaxed)) throw InterruptedError();
+}
+
+void Worker::allocBuffers()
+{
+    const auto& cfg = eng.cfg;
+    int slots = std::max(1, cfg.ioDepth);
+
+    if (!cfg.gpuIDs.empty()) {
+        setupGpu();
+        for (int i = 0; i < slots; i++) hostBufs.push_back(gpu->hostBuf(i));
+        ownHostBufs = false;
+    } else {
+        for (int i = 0; i < slots; i++) {
+            char* p = nullptr;
+            if (posix_memalign((void**)&p, 4096, cfg.blockSize))
+                throw WorkerError("host I/O buffer allocation failed");
+            hostBufs.push_back(p);
+        }
+        ownHostBufs = true;
+    }
+
+    // pre-fill host buffers with random data (defeats compression; same
+    // contract as reference LocalWorker.cpp:1386-1420)
+    RandAlgoXoshiro256ppSIMD<8> bufFillRng(cfg.benchSeed ^ (0x517cc1b7ULL * (globalRank + 1)));
+    for (auto p : hostBufs) bufFillRng.fillBuf(p, cfg.blockSize);
+
+    if (gpu) { // seed device buffers so a pct<100 refill mixes defined data
+        for (int i = 0; i < slots; i++) gpu->copyH2DAsync(i, cfg.blockSize);
+        gpu->syncStream();
+    }
+}
+
+void Worker::setupGpu()
+{
+    const auto& cfg = eng.cfg;
+    int devId = cfg.gpuIDs[globalRank % cfg.gpuIDs.size()];
+    int slots = std::max(1, cfg.ioDepth);
+
+    if (gpuDeviceCount() <= 0)
+        throw WorkerError("GPU requested (gpuids) but no HIP device is available — "
+                          "refusing silent CPU fallback");
+
+    gpu = std::make_unique<GpuCtx>(devId, slots, cfg.blockSize, cfg.gpuPinnedHostBufs);
+}
+
+void Worker::fairShareSlice(uint64_t totalLen, uint64_t& myStart, uint64_t& myLen) const
+{
+    // contiguous per-rank block slice; last rank takes remainder blocks+tail
+    // (same contract as reference fileModeIterateFilesSeq :3597)
+    const auto& cfg = eng.cfg;
+    uint64_t bs = cfg.blockSize;
+    uint64_t numRanks = cfg.numDataSetThreads;
+    uint64_t numBlocksTotal = (totalLen + bs - 1) / bs;
+    uint64_t perRank = numBlocksTotal / numRanks;
+    uint64_t startBlock = (uint64_t)globalRank * perRank;
+    uint64_t myBlocks = perRank;
+    if ((uint64_t)globalRank == numRanks - 1) myBlocks = numBlocksTotal - startBlock;
+
+    myStart = startBlock * bs;
+    uint64_t endByte = std::min(totalLen, (startBlock + myBlocks) * bs);
+    myLen = (myStart >= endByte) ? 0 : endByte - myStart;
+}
+
+std::unique_ptr<OffsetGen> Worker::makeOffsetGen(uint64_t rangeStart, uint64_t rangeLen)
+{
+    const auto& cfg = eng.cfg;
+    bool isWrite = (eng.currentPhase == Phase::WRITE);
+    uint64_t perWorkerRandAmount =
+        cfg.randAmount ? cfg.randAmount / cfg.numDataSetThreads : rangeLen;
+
+    std::unique_ptr<OffsetGen> gen;
+
+    if (cfg.backward) {
+        gen = std::make_unique<OffsetGenReverseSeq>(cfg.blockSize);
+    } else if (cfg.strided) {
+        gen = std::make_unique<OffsetGenStrided>(cfg.blockSize, globalRank,
+                                                 cfg.numDataSetThreads);
+    } else if (!cfg.random) {
+        gen = std::make_unique<OffsetGenSequential>(cfg.blockSize);
+    } else if (!cfg.randAligned) {
+        gen = std::make_unique<OffsetGenRandom>(cfg.blockSize, *rng, perWorkerRandAmount);
+    } else if (isWrite) {
+        // random aligned writes: full coverage so the resulting file is fully
+        // allocated (reference behavior, LocalWorker.cpp:1177-1185)
+        gen = std::make_unique<OffsetGenRandomAlignedFullCoverage>(
+            cfg.blockSize, cfg.benchSeed ^ (0x9E3779B97F4A7C15ULL * (globalRank + 1)));
+    } else {
+        gen = std::make_unique<OffsetGenRandomAligned>(cfg.blockSize, *rng, perWorkerRandAmount);
+    }
+
+    gen->reset(rangeStart, rangeLen);
+    return gen;
+}
+
+// single synchronous block I/O incl. GPU staging + verify hooks
+ssize_t Worker::blockIO(bool isWrite, int fd, int slot, uint64_t len, uint64_t fileOff)
+{
+    const auto& cfg = eng.cfg;
+    char* buf = hostBufs[slot];
+
+    if (isWrite) {
+        rateLimiter.wait(len);
+        preWriteFill(slot, len, fileOff);
+
+        if (gpu) { // stage GPU->host on this worker's stream
+            gpu->copyD2HAsync(slot, len);
+            gpu->syncStream();
+        }
+
+        ssize_t res = pwrite(fd, buf, len, fileOff);
+        if (res >= 0 && cfg.fsyncPerFile) { /* fsync handled at file close */ }
+        return res;
+    } else {
+        rateLimiter.wait(len);
+        ssize_t res = pread(fd, buf, len, fileOff);
+        if (res < 0) return res;
+
+        if (gpu) { // stage host->GPU (HBM3E resident buffers)
+            gpu->copyH2DAsync(slot, len);
+            gpu->syncStream();
+        }
+
+        postReadCheck(slot, (uint64_t)res, fileOff);
+        return res;
+    }
+}
+
+void Worker::preWriteFill(int slot, uint64_t len, uint64_t fileOff)
+{
+    const auto& cfg = eng.cfg;
+    char* buf = hostBufs[slot];
+
+    if (cfg.verifySalt >= 0) {
+        // integrity fill
+        if (gpu && (fileOff % 8 == 0) && (len % 8 == 0)) {
+            gpu->fillChecksumDev(slot, len, fileOff, (uint64_t)cfg.verifySalt);
+            return; // D2H staging copies it to host buf
+        }
+        fillChecksumCPU(buf, len, fileOff, (uint64_t)cfg.verifySalt);
+        if (gpu) { // keep device buffer coherent for the D2H staging copy
+            gpu->copyH2DAsync(slot, len);
+            gpu->syncStream();
+        }
+        return;
+    }
+
+    if (cfg.blockVarPct > 0) {
+        uint64_t refillLen = (len * cfg.blockVarPct) / 100;
+        if (gpu && (len % 16 == 0)) {
+            gpu->blockVarRefillDev(slot, len, refillLen,
+                                   cfg.benchSeed ^ (0xD1B54A32D192ED03ULL * (globalRank + 1)));
+            return;
+        }
+        // CPU refill: refillLen random bytes, rest a fresh constant u64
+        fillRng->fillBuf(buf, refillLen);
+        if (refillLen < len) {
+            uint64_t c = fillRng->next();
+            uint64_t pos = refillLen;
+            while (pos < len) {
+                uint64_t n = std::min<uint64_t>(8, len - pos);
+                std::memcpy(buf + pos, &c, n);
+                pos += n;
+            }
+        }
+        if (gpu) {
+            gpu->copyH2DAsync(slot, len);
+            gpu->syncStream();
+        }
+    }
+    // blockVarPct == 0: keep the pre-filled random buffer as is
+}
+
+void Worker::postReadCheck(int slot, uint64_t len, uint64_t fileOff)
+{
+    const auto& cfg = eng.cfg;
+    if (cfg.verifySalt < 0) return;
+
+    if (gpu && (fileOff % 8 == 0) && (len % 16 == 0)) {
+        GpuVerifyResult r = gpu->verifyChecksumDev(slot, len, fileOff, (uint64_t)cfg.verifySalt);
+        if (r.numMismatches)
+            throw WorkerError("Data verification failed (GPU). First bad file offset: " +
+                              std::to_string(r.firstBadFileOffset) +
+                              "; mismatching 8-byte words: " + std::to_string(r.numMismatches));
+        return;
+    }
+
+    uint64_t bad = verifyChecksumCPU(hostBufs[slot], len, fileOff, (uint64_t)cfg.verifySalt);
+    if (bad != UINT64_MAX)
+        throw WorkerError("Data verification failed. First bad file offset: " +
+                          std::to_string(bad));
+}
+
+// ---------------------------------------------------------------------------
+// file/bdev mode
+// ---------------------------------------------------------------------------
+
+namespace {
+
+struct FdGuard {
+    std::vector<int> fds;
+    ~FdGuard()
+    {
+        for (int fd : fds)
+            if (fd >= 0) close(fd);
+    }
+};
+
+} // namespace
+
+void Worker::fileModeBlocks(bool isWrite)
+{
+    const auto& cfg = eng.cfg;
+    if (cfg.ioDepth > 1) return fileModeBlocksUring(isWrite);
+
+    const uint64_t fileSize = eng.effFileSize;
+    const uint64_t bs = cfg.blockSize;
+    const size_t numFiles = cfg.paths.size();
+    const uint64_t numBlocksPerFile = (fileSize + bs - 1) / bs;
+
+    // open all files
+    FdGuard fg;
+    int openFlags = (isWrite ? O_WRONLY : O_RDONLY) | (cfg.directIO ? O_DIRECT : 0);
+    if (isWrite && cfg.pathType == PathType::FILE) openFlags |= O_CREAT;
+    for (const auto& p : cfg.paths) {
+        int fd = open(p.c_str(), openFlags, 0644);
+        if (fd < 0) throwErrno("open", p);
+        fg.fds.push_back(fd);
+        if (isWrite && cfg.pathType == PathType::FILE) {
+            if (cfg.truncate && ftruncate(fd, 0)) throwErrno("truncate", p);
+            if (cfg.truncToSize != UINT64_MAX && ftruncate(fd, fileSize))
+                throwErrno("truncate-to-size", p);
+            if (cfg.preallocFile && fallocate(fd, 0, 0, fileSize)) throwErrno("fallocate", p);
+        }
+    }
+
+    // Virtual concatenated range: sequential uses ceil blocks per file (the
+    // tail block is trimmed at I/O time); random/strided use floor blocks per
+    // file (reference fileModeIterateFilesSeq :3597 / ...Rand :3511).
+    const uint64_t mapBPF = (cfg.random || cfg.strided) ? (fileSize / bs) : numBlocksPerFile;
+    const uint64_t virtFileLen = mapBPF * bs;
+
+    auto mapBlock = [&](const BlockSpec& s, uint64_t& fileIdx, uint64_t& inFileOff,
+                        uint64_t& ioLen) -> bool {
+        if (!virtFileLen) return false;
+        fileIdx = s.offset / virtFileLen;
+        if (fileIdx >= numFiles) return false;
+        inFileOff = s.offset - fileIdx * virtFileLen;
+        if (inFileOff >= fileSize) return false;
+        ioLen = std::min(s.len, fileSize - inFileOff);
+        return true;
+    };
+
+    std::unique_ptr<OffsetGen> gen;
+    if (cfg.random || cfg.strided) {
+        // per-worker contiguous subrange of the virtual concatenated range
+        // (whole-block granularity; reference fileModeIterateFilesRand :3511)
+        uint64_t numBlocksTotal = mapBPF * numFiles;
+        uint64_t rangeLen = bs * (numBlocksTotal / cfg.numDataSetThreads);
+        uint64_t rangeOff = (uint64_t)globalRank * rangeLen;
+        if (cfg.strided) {
+            gen = std::make_unique<OffsetGenStrided>(bs, globalRank, cfg.numDataSetThreads);
+            gen->reset(0, bs * numBlocksTotal);
+        } else {
+            gen = makeOffsetGen(rangeOff, rangeLen);
+        }
+    } else {
+        uint64_t myStart, myLen;
+        fairShareSlice(virtFileLen * numFiles, myStart, myLen);
+        if (!myLen) return; // no work this round
+        gen = makeOffsetGen(myStart, myLen);
+    }
+
+    const bool lat = cfg.measureLat;
+    BlockSpec spec;
+    uint64_t opCount = 0;
+
+    while (gen->next(spec)) {
+        if ((opCount++ % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+
+        uint64_t fileIdx, inFileOff, ioLen;
+        if (!mapBlock(spec, fileIdx, inFileOff, ioLen)) continue;
+
+        auto t0 = lat ? Clock::now() : Clock::time_point();
+
+        ssize_t res = blockIO(isWrite, fg.fds[fileIdx], 0, ioLen, inFileOff);
+        if (res < 0)
+            throwErrno(isWrite ? "write" : "read", cfg.paths[fileIdx]);
+        if ((uint64_t)res != ioLen)
+            throw WorkerError(std::string("unexpected short ") + (isWrite ? "write" : "read") +
+                              ". Path: " + cfg.paths[fileIdx] +
+                              "; expected: " + std::to_string(ioLen) +
+                              "; got: " + std::to_string(res));
+
+        if (lat)
+            addIoLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
+                Clock::now() - t0).count());
+
+        liveOps.bytes.fetch_add(ioLen, std::memory_order_relaxed);
+        liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+    }
+}
+
+void Worker::fileModeBlocksUring(bool isWrite)
+{
+    const auto& cfg = eng.cfg;
+    const uint64_t fileSize = eng.effFileSize;
+    const uint64_t bs = cfg.blockSize;
+    const size_t numFiles = cfg.paths.size();
+    const uint64_t numBlocksPerFile = (fileSize + bs - 1) / bs;
+    const int depth = cfg.ioDepth;
+    const bool lat = cfg.measureLat;
+
+    FdGuard fg;
+    int openFlags = (isWrite ? O_WRONLY : O_RDONLY) | (cfg.directIO ? O_DIRECT : 0);
+    if (isWrite && cfg.pathType == PathType::FILE) openFlags |= O_CREAT;
+    for (const auto& p : cfg.paths) {
+        int fd = open(p.c_str(), openFlags, 0644);
+        if (fd < 0) throwErrno("open", p);
+        fg.fds.push_back(fd);
+        if (isWrite && cfg.pathType == PathType::FILE) {
+            if (cfg.truncate && ftruncate(fd, 0)) throwErrno("truncate", p);
+            if (cfg.truncToSize != UINT64_MAX && ftruncate(fd, fileSize))
+                throwErrno("truncate-to-size", p);
+            if (cfg.preallocFile && fallocate(fd, 0, 0, fileSize)) throwErrno("fallocate", p);
+        }
+    }
+
+    const uint64_t mapBPF = (cfg.random || cfg.strided) ? (fileSize / bs) : numBlocksPerFile;
+    const uint64_t virtFileLen = mapBPF * bs;
+
+    auto mapBlock = [&](const BlockSpec& s, uint64_t& fileIdx, uint64_t& inFileOff,
+                        uint64_t& ioLen) -> bool {
+        if (!virtFileLen) return false;
+        fileIdx = s.offset / virtFileLen;
+        if (fileIdx >= numFiles) return false;
+        inFileOff = s.offset - fileIdx * virtFileLen;
+        if (inFileOff >= fileSize) return false;
+        ioLen = std::min(s.len, fileSize - inFileOff);
+        return true;
+    };
+
+    std::unique_ptr<OffsetGen> gen;
+    if (cfg.random || cfg.strided) {
+        uint64_t numBlocksTotal = mapBPF * numFiles;
+        uint64_t rangeLen = bs * (numBlocksTotal / cfg.numDataSetThreads);
+        uint64_t rangeOff = (uint64_t)globalRank * rangeLen;
+        if (cfg.strided) {
+            gen = std::make_unique<OffsetGenStrided>(bs, globalRank, cfg.numDataSetThreads);
+            gen->reset(0, bs * numBlocksTotal);
+        } else {
+            gen = makeOffsetGen(rangeOff, rangeLen);
+        }
+    } else {
+        uint64_t myStart, myLen;
+        fairShareSlice(virtFileLen * numFiles, myStart, myLen);
+        if (!myLen) return;
+        gen = makeOffsetGen(myStart, myLen);
+    }
+
+    IoUring ring;
+    ring.init(depth);
+
+    struct SlotState {
+        uint64_t inFileOff = 0;
+        uint64_t len = 0;
+        Clock::time_point start;
+        int fileIdx = 0;
+    };
+    std::vector<SlotState> slots(depth);
+
+    BlockSpec spec;
+    int inFlight = 0;
+    bool exhausted = false;
+    std::vector<IoUring::Completion> comps(depth);
+
+    auto prepSlot = [&](int slot) -> bool {
+        if (!gen->next(spec)) return false;
+        uint64_t fileIdx, inFileOff, ioLen;
+        if (!mapBlock(spec, fileIdx, inFileOff, ioLen)) return true; // skip this block
+
+        if (isWrite) {
+            rateLimiter.wait(ioLen);
+            preWriteFill(slot, ioLen, inFileOff);
+            if (gpu) {
+                gpu->copyD2HAsync(slot, ioLen);
+                gpu->syncStream();
+            }
+        } else {
+            rateLimiter.wait(ioLen);
+        }
+
+        slots[slot] = {inFileOff, ioLen, lat ? Clock::now() : Clock::time_point(),
+                       (int)fileIdx};
+        if (!ring.prep(isWrite, fg.fds[fileIdx], hostBufs[slot], ioLen, inFileOff,
+                       (uint64_t)slot))
+            throw WorkerError("io_uring SQ unexpectedly full");
+        inFlight++;
+        return true;
+    };
+
+    // seed the queue
+    for (int s = 0; s < depth && !exhausted; s++)
+        if (!prepSlot(s)) exhausted = true;
+
+    uint64_t opCount = 0;
+
+    while (inFlight > 0) {
+        checkInterrupt();
+        ring.submitAndWait(1);
+        unsigned n = ring.reap(comps.data(), depth);
+        for (unsigned i = 0; i < n; i++) {
+            int slot = (int)comps[i].userData;
+            SlotState& st = slots[slot];
+            inFlight--;
+
+            if (comps[i].res < 0)
+                throw WorkerError(std::string("async ") + (isWrite ? "write" : "read") +
+                                  " failed. Path: " + cfg.paths[st.fileIdx] +
+                                  "; SysErr: " + strerror(-comps[i].res));
+            if ((uint64_t)comps[i].res != st.len)
+                throw WorkerError(std::string("unexpected short async ") +
+                                  (isWrite ? "write" : "read") +
+                                  ". Path: " + cfg.paths[st.fileIdx]);
+
+            if (!isWrite) {
+                if (gpu) {
+                    gpu->copyH2DAsync(slot, st.len);
+                    gpu->syncStream();
+                }
+                postReadCheck(slot, st.len, st.inFileOff);
+            }
+
+            if (lat)
+                addIoLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
+                    Clock::now() - st.start).count());
+
+            liveOps.bytes.fetch_add(st.len, std::memory_order_relaxed);
+            liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+
+            if ((opCount++ % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+
+            if (!exhausted && !prepSlot(slot)) exhausted = true;
+        }
+    }
+}
+
+void Worker::fileModeDelete()
+{
+    const auto& cfg = eng.cfg;
+    for (size_t i = 0; i < cfg.paths.size(); i++) {
+        if ((int)(i % cfg.numDataSetThreads) != globalRank) continue;
+        checkInterrupt();
+        if (unlink(cfg.paths[i].c_str())) {
+            if (!cfg.ignoreDelErrors) throwErrno("unlink", cfg.paths[i]);
+        }
+        liveOps.entries.fetch_add(1, std::memory_order_relaxed);
+    }
+}
+
+void Worker::fileModeStat()
+{
+    const auto& cfg = eng.cfg;
+    struct stat st;
+    for (size_t i = 0; i < cfg.paths.size(); i++) {
+        if ((int)(i % cfg.numDataSetThreads) != globalRank) continue;
+        checkInterrupt();
+        if (stat(cfg.paths[i].c_str(), &st)) throwErrno("stat", cfg.paths[i]);
+        liveOps.entries.fetch_add(1, std::memory_order_relaxed);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// dir mode
+// ---------------------------------------------------------------------------
+
+void Worker::dirModeMkdirs()
+{
+    const auto& cfg = eng.cfg;
+    const bool lat = cfg.measureLat;
+    char rel[64];
+
+    // rank root dir under every bench path that this worker will use
+    snprintf(rel, sizeof(rel), "r%d", globalRank);
+    for (const auto& p : cfg.paths) mkdirIgnoreExists(p + "/" + rel);
+
+    for (uint64_t d = 0; d < cfg.numDirs; d++) {
+        if ((d % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+        size_t pathIdx = (globalRank + d) % cfg.paths.size();
+        snprintf(rel, sizeof(rel), "r%d/d%lu", globalRank, (unsigned long)d);
+        std::string full = cfg.paths[pathIdx] + "/" + rel;
+
+        auto t0 = lat ? Clock::now() : Clock::time_point();
+        if (mkdir(full.c_str(), 0777) && errno != EEXIST) throwErrno("mkdir", full);
+        if (lat)
+            addEntryLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
+                Clock::now() - t0).count());
+        liveOps.entries.fetch_add(1, std::memory_order_relaxed);
+    }
+}
+
+void Worker::dirModeRmdirs()
+{
+    const auto& cfg = eng.cfg;
+    const bool lat = cfg.measureLat;
+    char rel[64];
+
+    for (uint64_t d = 0; d < cfg.numDirs; d++) {
+        if ((d % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+        size_t pathIdx = (globalRank + d) % cfg.paths.size();
+        snprintf(rel, sizeof(rel), "r%d/d%lu", globalRank, (unsigned long)d);
+        std::string full = cfg.paths[pathIdx] + "/" + rel;
+
+        auto t0 = lat ? Clock::now() : Clock::time_point();
+        if (rmdir(full.c_str())) {
+            if (!cfg.ignoreDelErrors) throwErrno("rmdir", full);
+        }
+        if (lat)
+            addEntryLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
+                Clock::now() - t0).count());
+        liveOps.entries.fetch_add(1, std::memory_order_relaxed);
+    }
+
+    // rank root dirs (not counted as entries)
+    snprintf(rel, sizeof(rel), "r%d", globalRank);
+    for (const auto& p : cfg.paths) {
+        std::string full = p + "/" + rel;
+        if (rmdir(full.c_str()) && errno != ENOENT && errno != ENOTEMPTY) {
+            if (!cfg.ignoreDelErrors) throwErrno("rmdir", full);
+        }
+    }
+}
+
+void Worker::dirModeFiles(Phase phase)
+{
+    const auto& cfg = eng.cfg;
+    const bool lat = cfg.measureLat;
+    const bool isWrite = (phase == Phase::WRITE);
+    const bool isRead = (phase == Phase::READ);
+    const bool haveSubdirs = (cfg.numDirs > 0);
+    const uint64_t numDirs = haveSubdirs ? cfg.numDirs : 1;
+    const uint64_t fileSize = cfg.fileSize;
+    char rel[128];
+
+    int openFlags = 0;
+    if (isWrite) openFlags = O_CREAT | O_WRONLY;
+    if (isRead) openFlags = O_RDONLY;
+    if (cfg.directIO) openFlags |= O_DIRECT;
+
+    // offsets within one file
+    std::unique_ptr<OffsetGen> gen;
+
+    for (uint64_t d = 0; d < numDirs; d++) {
+        for (uint64_t f = 0; f < cfg.numFiles; f++) {
+            if ((f % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+
+            size_t pathIdx = (globalRank + d) % cfg.paths.size();
+            if (haveSubdirs)
+                snprintf(rel, sizeof(rel), "r%d/d%lu/r%d-f%lu", globalRank, (unsigned long)d,
+                         globalRank, (unsigned long)f);
+            else
+                snprintf(rel, sizeof(rel), "r%d-f%lu", globalRank, (unsigned long)f);
+
+            std::string full = cfg.paths[pathIdx] + "/" + rel;
+
+            auto tEntry0 = lat ? Clock::now() : Clock::time_point();
+
+            switch (phase) {
+                case Phase::WRITE:
+                case Phase::READ: {
+                    int fd = open(full.c_str(), openFlags, 0644);
+                    if (fd < 0) throwErrno("open", full);
+
+                    try {
+                        if (isWrite) {
+                            if (cfg.truncate && ftruncate(fd, 0)) throwErrno("truncate", full);
+                            if (cfg.truncToSize != UINT64_MAX && ftruncate(fd, fileSize))
+                                throwErrno("truncate-to-size", full);
+                            if (cfg.preallocFile && fallocate(fd, 0, 0, fileSize))
+                                throwErrno("fallocate", full);
+                        }
+
+                        if (!gen) gen = makeOffsetGen(0, fileSize);
+                        else gen->reset(0, fileSize);
+
+                        BlockSpec spec;
+                        while (gen->next(spec)) {
+                            auto t0 = lat ? Clock::now() : Clock::time_point();
+                            ssize_t res = blockIO(isWrite, fd, 0, spec.len, spec.offset);
+                            if (res < 0) throwErrno(isWrite ? "write" : "read", full);
+                            if ((uint64_t)res != spec.len)
+                                throw WorkerError("unexpected short I/O on " + full);
+                            if (lat)
+                                addIoLat((uint64_t)std::chrono::duration_cast<
+                                    std::chrono::microseconds>(Clock::now() - t0).count());
+                            liveOps.bytes.fetch_add(spec.len, std::memory_order_relaxed);
+                            liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+                        }
+
+                        if (isWrite && cfg.fsyncPerFile && fsync(fd)) throwErrno("fsync", full);
+                    } catch (...) {
+                        close(fd);
+                        throw;
+                    }
+                    close(fd);
+                    break;
+                }
+                case Phase::STAT: {
+                    struct stat st;
+                    if (stat(full.c_str(), &st)) throwErrno("stat", full);
+                    break;
+                }
+                case Phase::RMFILES: {
+                    if (unlink(full.c_str())) {
+                        if (!cfg.ignoreDelErrors) throwErrno("unlink", full);
+                    }
+                    break;
+                }
+                default:
+                    throw WorkerError("bad dir mode phase");
+            }
+
+            if (lat)
+                addEntryLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
+                    Clock::now() - tEntry0).count());
+            liveOps.entries.fetch_add(1, std::memory_order_relaxed);
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// sync / dropcaches
+// ---------------------------------------------------------------------------
+
+void Worker::anyModeSync()
+{
+    if (localRank != 0) return; // once per instance
+    sync();
+}
+
+void Worker::anyModeDropCaches()
+{
+    if (localRank != 0) return;
+    int fd = open("/proc/sys/vm/drop_caches", O_WRONLY);
+    if (fd < 0) throwErrno("open", "/proc/sys/vm/drop_caches");
+    if (write(fd, "3\n", 2) != 2) {
+        close(fd);
+        throwErrno("write", "/proc/sys/vm/drop_caches");
+    }
+    close(fd);
+}
+
+// ---------------------------------------------------------------------------
+// thread main / phase dispatch
+// ---------------------------------------------------------------------------
+
+void Worker::runPhase()
+{
+    const auto& cfg = eng.cfg;
+
+    switch (eng.currentPhase) {
+        case Phase::MKDIRS:
+            dirModeMkdirs();
+            break;
+        case Phase::RMDIRS:
+            dirModeRmdirs();
+            break;
+        case Phase::WRITE:
+            if (cfg.pathType == PathType::DIR)
+                dirModeFiles(Phase::WRITE);
+            else
+                fileModeBlocks(true);
+            break;
+        case Phase::READ:
+            if (cfg.pathType == PathType::DIR)
+                dirModeFiles(Phase::READ);
+            else
+                fileModeBlocks(false);
+            break;
+        case Phase::STAT:
+            if (cfg.pathType == PathType::DIR)
+                dirModeFiles(Phase::STAT);
+            else
+                fileModeStat();
+            break;
+        case Phase::RMFILES:
+            if (cfg.pathType == PathType::DIR)
+                dirModeFiles(Phase::RMFILES);
+            else
+                fileModeDelete();
+            break;
+        case Phase::SYNC:
+            anyModeSync();
+            break;
+        case Phase::DROPCACHES:
+            anyModeDropCaches();
+            break;
+        default:
+            throw WorkerError(std::string("engine phase not implemented: ") +
+                              phaseName(eng.currentPhase));
+    }
+}
+
+void Worker::threadMain()
+{
+    bool hadError = false;
+
+    try {
+        const auto& cfg = eng.cfg;
+
+        rng.reset(makeRandAlgo(cfg.randAlgo,
+                               cfg.benchSeed ^ (0xBF58476D1CE4E5B9ULL * (globalRank + 1))));
+        fillRng.reset(makeRandAlgo(cfg.blockVarAlgo,
+                                   cfg.benchSeed ^ (0x94D049BB133111EBULL * (globalRank + 1))));
+
+        // buffers only needed for data phases
+        if (eng.currentPhase == Phase::WRITE || eng.currentPhase == Phase::READ)
+            allocBuffers();
+
+        rateLimiter.init(eng.currentPhase == Phase::WRITE ? eng.cfg.limitWriteBps
+                                                          : eng.cfg.limitReadBps);
+
+        { // start gate: all workers begin simultaneously
+            std::unique_lock<std::mutex> lk(eng.gateMtx);
+            eng.gateCv.wait(lk, [&] { return eng.gateOpen; });
+        }
+
+        runPhase();
+    } catch (const InterruptedError&) {
+        error = "interrupted";
+        hadError = true;
+    } catch (const std::exception& e) {
+        error = e.what();
+        hadError = true;
+    }
+
+    elapsedUSec = nowUSecSince(eng.phaseStart);
+    eng.onWorkerDone(*this, hadError);
+}
+
+// ---------------------------------------------------------------------------
+// Engine
+// ---------------------------------------------------------------------------
+
+Engine::Engine(EngineConfig cfgIn) : cfg(std::move(cfgIn)) {}
+
+Engine::~Engine()
+{
+    interrupt();
+    for (auto& t : threads)
+        if (t.joinable()) t.join();
+}
+
+void Engine::prepare()
+{
+    resolvedFileSizes.clear();
+    effFileSize = cfg.fileSize;
+
+    if (cfg.pathType == PathType::DIR) return;
+
+    for (const auto& p : cfg.paths) {
+        struct stat st;
+        if (stat(p.c_str(), &st)) {
+            if (cfg.pathType == PathType::FILE && errno == ENOENT) {
+                resolvedFileSizes.push_back(cfg.fileSize);
+                continue; // will be created by write phase
+            }
+            throwErrno("stat", p);
+        }
+        if (S_ISBLK(st.st_mode)) {
+            int fd = open(p.c_str(), O_RDONLY);
+            if (fd < 0) throwErrno("open", p);
+            uint64_t sz = 0;
+            if (ioctl(fd, BLKGETSIZE64, &sz)) {
+                close(fd);
+                throwErrno("BLKGETSIZE64", p);
+            }
+            close(fd);
+            resolvedFileSizes.push_back(sz);
+        } else {
+            resolvedFileSizes.push_back(st.st_size);
+        }
+    }
+
+    if (!cfg.fileSize && !resolvedFileSizes.empty()) {
+        // no -s given: use detected size (min across paths for uniform stripe)
+        uint64_t minSz = UINT64_MAX;
+        for (auto s : resolvedFileSizes) minSz = std::min(minSz, s);
+        effFileSize = (minSz == UINT64_MAX) ? 0 : minSz;
+    }
+}
+
+void Engine::startPhase(Phase phase)
+{
+    if (phaseRunning) throw std::runtime_error("phase already running");
+
+    currentPhase = phase;
+    interruptFlag.store(false);
+    workersDone.store(0);
+    workersWithError.store(0);
+    stonewallTriggered.store(false);
+    gateOpen = false;
+
+    workers.clear();
+    threads.clear();
+    for (int i = 0; i < cfg.numThreads; i++) workers.push_back(std::make_unique<Worker>(*this, i));
+
+    for (auto& w : workers) threads.emplace_back(&Worker::threadMain, w.get());
+
+    { // release the gate; timestamp = phase start
+        std::lock_guard<std::mutex> lk(gateMtx);
+        phaseStart = std::chrono::steady_clock::now();
+        gateOpen = true;
+    }
+    gateCv.notify_all();
+    phaseRunning = true;
+}
+
+void Engine::onWorkerDone(Worker& w, bool hadError)
+{
+    std::lock_guard<std::mutex> lk(doneMtx);
+
+    if (hadError) {
+        workersWithError.fetch_add(1);
+        interruptFlag.store(true); // peers stop cooperatively (reference behavior)
+    } else if (!stonewallTriggered.load()) {
+        // first successful finisher triggers the stonewall: snapshot every
+        // worker's live counters ("first done" result)
+        bool didWork = w.liveOps.entries.load(std::memory_order_relaxed) ||
+                       w.liveOps.bytes.load(std::memory_order_relaxed) ||
+                       w.liveOps.iops.load(std::memory_order_relaxed);
+        if (didWork) {
+            uint64_t elapsed = nowUSecSince(phaseStart);
+            for (auto& peer : workers) {
+                peer->stonewallOps.takeFrom(peer->liveOps);
+                peer->stonewallElapsedUSec = elapsed;
+            }
+            stonewallTriggered.store(true);
+        }
+    }
+
+    workersDone.fetch_add(1);
+    doneCv.notify_all();
+}
+
+bool Engine::waitPhaseDone(int64_t timeoutMs)
+{
+    std::unique_lock<std::mutex> lk(doneMtx);
+    auto pred = [&] { return workersDone.load() >= (int)workers.size(); };
+    if (timeoutMs < 0) {
+        doneCv.wait(lk, pred);
+        return true;
+    }
+    return doneCv.wait_for(lk, std::chrono::milliseconds(timeoutMs), pred);
+}
+
+void Engine::interrupt() { interruptFlag.store(true); }
+
+Engine::LivePoll Engine::poll()
+{
+    LivePoll lp{};
+    lp.workersTotal = (int)workers.size();
+    lp.workersDone = workersDone.load();
+    lp.workersWithError = workersWithError.load();
+    lp.stonewallTriggered = stonewallTriggered.load();
+    lp.elapsedUSec = phaseRunning ? nowUSecSince(phaseStart) : 0;
+
+    for (auto& w : workers) {
+        lp.entries += w->liveOps.entries.load(std::memory_order_relaxed);
+        lp.bytes += w->liveOps.bytes.load(std::memory_order_relaxed);
+        lp.iops += w->liveOps.iops.load(std::memory_order_relaxed);
+        lp.latNumIOs += w->liveIoLatNum.load(std::memory_order_relaxed);
+        lp.latSumIOs += w->liveIoLatSum.load(std::memory_order_relaxed);
+        lp.latNumEntries += w->liveEntryLatNum.load(std::memory_order_relaxed);
+        lp.latSumEntries += w->liveEntryLatSum.load(std::memory_order_relaxed);
+    }
+    return lp;
+}
+
+std::vector<WorkerResult> Engine::finishPhase()
+{
+    for (auto& t : threads)
+        if (t.joinable()) t.join();
+    threads.clear();
+    phaseRunning = false;
+
+    std::vector<WorkerResult> results;
+    for (auto& w : workers) {
+        WorkerResult r;
+        r.rank = w->globalRank;
+        r.elapsedUSec = w->elapsedUSec;
+        r.total.takeFrom(w->liveOps);
+        r.stonewall = stonewallTriggered.load() ? w->stonewallOps : r.total;
+        r.stonewallElapsedUSec =
+            stonewallTriggered.load() ? w->stonewallElapsedUSec : w->elapsedUSec;
+        r.ioLatVec = w->ioLat.toVec();
+        r.entryLatVec = w->entryLat.toVec();
+        r.error = w->error;
+        results.push_back(std::move(r));
+    }
+
+    workers.clear();
+    currentPhase = Phase::IDLE;
+    return results;
+}
+
+std::pair<uint64_t, uint64_t> Engine::plannedWork(Phase phase) const
+{
+    uint64_t entries = 0, bytes = 0;
+
+    if (cfg.pathType == PathType::DIR) {
+        uint64_t numDirs = cfg.numDirs ? cfg.numDirs : 1;
+        switch (phase) {
+            case Phase::MKDIRS:
+            case Phase::RMDIRS:
+                entries = cfg.numDirs * cfg.numThreads;
+                break;
+            case Phase::WRITE:
+            case Phase::READ:
+                entries = numDirs * cfg.numFiles * cfg.numThreads;
+                bytes = entries * cfg.fileSize;
+                break;
+            case Phase::STAT:
+            case Phase::RMFILES:
+                entries = numDirs * cfg.numFiles * cfg.numThreads;
+                break;
+            default:
+                break;
+        }
+    } else {
+        switch (phase) {
+            case Phase::WRITE:
+            case Phase::READ: {
+                uint64_t totalLen = effFileSize * cfg.paths.size();
+                if (cfg.random) {
+                    uint64_t amount = cfg.randAmount ? cfg.randAmount : totalLen;
+                    bytes = (amount / cfg.numDataSetThreads) * cfg.numThreads;
+                } else {
+                    // fair-share block slices over the padded virtual range,
+                    // minus bytes trimmed off per-file tail blocks
+                    uint64_t bs = cfg.blockSize;
+                    uint64_t bpf = bs ? (effFileSize + bs - 1) / bs : 0;
+                    uint64_t tailTrim = bpf * bs - effFileSize; // per file
+                    uint64_t numBlocksTotal = bpf * cfg.paths.size();
+                    uint64_t perRank = numBlocksTotal / cfg.numDataSetThreads;
+                    auto tailsBelow = [&](uint64_t x) -> uint64_t {
+                        return (bpf && x >= bpf) ? ((x - bpf) / bpf + 1) : 0;
+                    };
+                    for (int i = 0; i < cfg.numThreads; i++) {
+                        uint64_t grank = cfg.rankOffset + i;
+                        uint64_t startBlock = grank * perRank;
+                        uint64_t myBlocks = perRank;
+                        if (grank == (uint64_t)cfg.numDataSetThreads - 1)
+                            myBlocks = (numBlocksTotal > startBlock)
+                                           ? numBlocksTotal - startBlock
+                                           : 0;
+                        uint64_t endBlock = startBlock + myBlocks;
+                        uint64_t tails = tailsBelow(endBlock) - tailsBelow(startBlock);
+                        bytes += myBlocks * bs - tails * tailTrim;
+                    }
+                }
+                break;
+            }
+            case Phase::STAT:
+            case Phase::RMFILES:
+                for (size_t i = 0; i < cfg.paths.size(); i++)
+                    if ((int)(i % cfg.numDataSetThreads) >= cfg.rankOffset &&
+                        (int)(i % cfg.numDataSetThreads) < cfg.rankOffset + cfg.numThreads)
+                        entries++;
+                break;
+            default:
+                break;
+        }
+    }
+    return {entries, bytes};
+}
+
+} // namespace eb

@@ -1,0 +1,230 @@
+// The elbencho_amd I/O engine: multithreaded block/file/dir benchmark core.
+//
+// MI355X-native counterpart of the reference's LocalWorker/WorkerManager
+// machinery (/root/reference/source/workers/LocalWorker.{h,cpp},
+// WorkerManager.{h,cpp}) — independent design:
+//   - phase lifecycle driven from Python (start/poll/wait/finish), worker
+//     threads spawned per phase behind a start gate,
+//   - io_uring (raw syscalls) instead of libaio for async depth,
+//   - GPU staging through GpuCtx (HIP streams + gfx950 kernels, see gpu.h),
+//   - stonewall ("first done") snapshots taken by the first finisher reading
+//     peers' relaxed atomics.
+
+#pragma once
+
+#include <chrono>
+#include <condition_variable>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include "common.h"
+#include "gpu.h"
+#include "histogram.h"
+#include "offsetgen.h"
+#include "rand.h"
+#include "rate.h"
+
+namespace eb {
+
+struct EngineConfig {
+    std::vector<std::string> paths;
+    PathType pathType = PathType::FILE;
+
+    int numThreads = 1;
+    int rankOffset = 0;        // global rank of this instance's worker 0
+    int numDataSetThreads = 1; // total workers across all instances sharing the dataset
+
+    uint64_t numDirs = 0;  // dir mode: dirs per thread
+    uint64_t numFiles = 0; // dir mode: files per dir (per thread)
+    uint64_t fileSize = 0;
+    uint64_t blockSize = 1ULL << 20;
+    int ioDepth = 1;
+
+    bool directIO = false;
+    bool random = false;
+    bool randAligned = true;
+    uint64_t randAmount = 0; // bytes per instance; 0 = dataset size
+    bool strided = false;
+    bool backward = false;
+
+    bool truncate = false;
+    uint64_t truncToSize = UINT64_MAX; // UINT64_MAX = off
+    bool preallocFile = false;
+    bool fsyncPerFile = false;
+
+    int64_t verifySalt = -1; // -1 = off
+    bool verifyDirect = false;
+    int blockVarPct = 100;
+    std::string blockVarAlgo = "fast";
+    std::string randAlgo = "fast";
+
+    std::vector<int> gpuIDs; // empty = CPU buffers only
+    bool gpuPinnedHostBufs = true;
+
+    bool measureLat = false;
+    uint64_t limitReadBps = 0;
+    uint64_t limitWriteBps = 0;
+
+    bool ignoreDelErrors = false;
+    bool dirSharing = false;
+    bool infiniteLoop = false;
+
+    uint64_t benchSeed = 0x243F6A8885A308D3ULL; // per-run; Python sets per iteration
+};
+
+struct WorkerResult {
+    int rank = 0; // global rank
+    uint64_t elapsedUSec = 0;
+    LiveOpsSnapshot total;
+    LiveOpsSnapshot stonewall;
+    uint64_t stonewallElapsedUSec = 0;
+    std::vector<uint64_t> ioLatVec;
+    std::vector<uint64_t> entryLatVec;
+    std::string error;
+};
+
+class Engine;
+
+// One benchmark worker = one OS thread (+ its GPU stream when GPU mode).
+class Worker {
+public:
+    Worker(Engine& engine, int localRank);
+    ~Worker();
+
+    void threadMain();
+
+    AtomicLiveOps liveOps;
+    LiveOpsSnapshot stonewallOps;
+    uint64_t stonewallElapsedUSec = 0;
+
+    // histograms are worker-private until the thread is joined; the live
+    // stats poller reads only these running-average atomics
+    std::atomic<uint64_t> liveIoLatNum{0}, liveIoLatSum{0};
+    std::atomic<uint64_t> liveEntryLatNum{0}, liveEntryLatSum{0};
+
+    LatencyHistogram ioLat;
+    LatencyHistogram entryLat;
+
+    uint64_t elapsedUSec = 0;
+    std::string error;
+    int localRank;
+    int globalRank;
+
+private:
+    using Clock = std::chrono::steady_clock;
+
+    void runPhase();
+    void checkInterrupt();
+
+    // phase bodies
+    void fileModeBlocks(bool isWrite);
+    void fileModeBlocksUring(bool isWrite);
+    void fileModeDelete();
+    void fileModeStat();
+    void dirModeMkdirs();
+    void dirModeRmdirs();
+    void dirModeFiles(Phase phase);
+    void anyModeSync();
+    void anyModeDropCaches();
+
+    // per-block helpers (sync path)
+    void addIoLat(uint64_t us)
+    {
+        ioLat.add(us);
+        liveIoLatNum.fetch_add(1, std::memory_order_relaxed);
+        liveIoLatSum.fetch_add(us, std::memory_order_relaxed);
+    }
+    void addEntryLat(uint64_t us)
+    {
+        entryLat.add(us);
+        liveEntryLatNum.fetch_add(1, std::memory_order_relaxed);
+        liveEntryLatSum.fetch_add(us, std::memory_order_relaxed);
+    }
+    void preWriteFill(int slot, uint64_t len, uint64_t fileOff);
+    void postReadCheck(int slot, uint64_t len, uint64_t fileOff);
+    ssize_t blockIO(bool isWrite, int fd, int slot, uint64_t len, uint64_t fileOff);
+
+    // setup
+    void allocBuffers();
+    void setupGpu();
+    std::unique_ptr<OffsetGen> makeOffsetGen(uint64_t myRangeStart, uint64_t myRangeLen);
+    void fairShareSlice(uint64_t totalLen, uint64_t& myStart, uint64_t& myLen) const;
+
+    Engine& eng;
+    std::vector<char*> hostBufs; // CPU-owned unless GPU mode (then GpuCtx owns)
+    bool ownHostBufs = false;
+    std::unique_ptr<GpuCtx> gpu;
+    std::unique_ptr<RandAlgo> rng;        // offsets
+    std::unique_ptr<RandAlgo> fillRng;    // block variance fill
+    RateLimiter rateLimiter;
+    uint64_t interruptCheckCounter = 0;
+};
+
+// Shared per-phase state + the engine facade exposed to Python.
+class Engine {
+public:
+    explicit Engine(EngineConfig cfg);
+    ~Engine();
+
+    // Resolve path type specifics (bdev sizes, existing file sizes).
+    void prepare();
+
+    void startPhase(Phase phase);
+    bool waitPhaseDone(int64_t timeoutMs); // true when all workers finished
+    void interrupt();
+
+    struct LivePoll {
+        uint64_t entries, bytes, iops;
+        int workersDone, workersTotal, workersWithError;
+        uint64_t elapsedUSec;
+        bool stonewallTriggered;
+        uint64_t latNumIOs, latSumIOs, latNumEntries, latSumEntries;
+    };
+    LivePoll poll();
+
+    std::vector<WorkerResult> finishPhase(); // joins threads, returns results
+
+    // planned work for dryrun/progress: {entries, bytes} for this instance
+    std::pair<uint64_t, uint64_t> plannedWork(Phase phase) const;
+
+    const EngineConfig& config() const { return cfg; }
+
+    // ---- shared state visible to workers ----
+    EngineConfig cfg;
+    Phase currentPhase = Phase::IDLE;
+    std::atomic<bool> interruptFlag{false};
+    std::atomic<int> workersDone{0};
+    std::atomic<int> workersWithError{0};
+    std::atomic<bool> stonewallTriggered{false};
+    std::chrono::steady_clock::time_point phaseStart;
+
+    std::mutex gateMtx;
+    std::condition_variable gateCv;
+    bool gateOpen = false;
+
+    std::mutex doneMtx;
+    std::condition_variable doneCv;
+
+    std::vector<std::unique_ptr<Worker>> workers;
+
+    // resolved at prepare()
+    std::vector<uint64_t> resolvedFileSizes; // per path (file/bdev mode)
+    uint64_t effFileSize = 0;                // uniform stripe unit
+
+    void onWorkerDone(Worker& w, bool hadError);
+
+private:
+    std::vector<std::thread> threads;
+    bool phaseRunning = false;
+};
+
+// CPU-side integrity checksum helpers (pattern: u64 at 8-aligned file offset
+// o has value o + salt, little-endian; arbitrary alignment handled bytewise).
+void fillChecksumCPU(char* buf, uint64_t len, uint64_t fileOff, uint64_t salt);
+// returns UINT64_MAX when ok, else file offset of first mismatching byte
+uint64_t verifyChecksumCPU(const char* buf, uint64_t len, uint64_t fileOff, uint64_t salt);
+
+} // namespace eb

@@ -1,0 +1,78 @@
+// GPU buffer management + CDNA4 kernel wrappers — host-facing interface.
+//
+// MI355X-native replacement of the reference's CUDA/cuFile call sites
+// (/root/reference/source/workers/LocalWorker.cpp:1427-1537 allocGPUIOBuffer,
+// :2269-2310 curand block-variance refill, :2437-2486 staging memcpys).
+// Design differences (deliberate, per BASELINE.json north star):
+//   - random fill and integrity fill/verify run as hand-written gfx950 HIP
+//     kernels directly in HBM3E (the reference does fill on CPU or curand,
+//     verify always on CPU),
+//   - staging uses hipMemcpyAsync on a per-worker non-blocking stream with
+//     pinned host buffers (the reference uses synchronous cudaMemcpy),
+//   - there is no cuFile/GDS on ROCm: the "direct" storage<->HBM path is
+//     O_DIRECT into pinned host bounce buffers behind the same seam.
+//
+// All HIP API usage lives in gpu_kernels.hip; this header keeps the engine
+// translation-unit HIP-free.
+
+#pragma once
+
+#include <cstdint>
+#include <string>
+
+namespace eb {
+
+struct GpuVerifyResult {
+    uint64_t numMismatches;
+    uint64_t firstBadFileOffset; // UINT64_MAX if none
+};
+
+// Returns number of visible HIP devices; 0 when no GPU or no driver.
+int gpuDeviceCount();
+
+// Human-readable device name (empty if unavailable).
+std::string gpuDeviceName(int deviceId);
+
+// Per-worker GPU context: one HIP stream, `numSlots` device buffers of
+// `bufSize` bytes (one per io-depth slot) and matching pinned host buffers.
+class GpuCtx {
+public:
+    GpuCtx(int deviceId, int numSlots, uint64_t bufSize, bool pinnedHostBufs);
+    ~GpuCtx();
+    GpuCtx(const GpuCtx&) = delete;
+    GpuCtx& operator=(const GpuCtx&) = delete;
+
+    int deviceId() const { return devId; }
+    char* hostBuf(int slot) const;
+    uint64_t bufSize() const { return slotSize; }
+
+    // --- async staging copies on this worker's stream ---
+    void copyH2DAsync(int slot, uint64_t len);
+    void copyD2HAsync(int slot, uint64_t len);
+    void syncStream();
+
+    // --- device-side buffer ops (hand-written gfx950 kernels) ---
+
+    // Fill device buffer with xoshiro256++ random data (replaces curand).
+    void fillRandDev(int slot, uint64_t len, uint64_t seed);
+
+    // Integrity-checksum fill: u64 at 8-aligned file offset o gets value
+    // o + salt. Requires fileOff % 8 == 0 and len % 8 == 0.
+    void fillChecksumDev(int slot, uint64_t len, uint64_t fileOff, uint64_t salt);
+
+    // Verify the checksum pattern on-device (LDS/wave-reduced mismatch count
+    // + first bad offset). Requires 8-aligned fileOff/len. Synchronizes.
+    GpuVerifyResult verifyChecksumDev(int slot, uint64_t len, uint64_t fileOff, uint64_t salt);
+
+    // Block-variance refill: first refillLen bytes random, remainder filled
+    // with one random u64 constant (defeats dedup). 8-aligned lengths.
+    void blockVarRefillDev(int slot, uint64_t len, uint64_t refillLen, uint64_t seed);
+
+private:
+    struct Impl;
+    Impl* impl;
+    int devId;
+    uint64_t slotSize;
+};
+
+} // namespace eb

@@ -1,0 +1,175 @@
+// pybind11 bindings for the elbencho_amd core engine (_core).
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "engine.h"
+#include "gpu.h"
+#include "histogram.h"
+
+namespace py = pybind11;
+using namespace eb;
+
+namespace {
+
+EngineConfig configFromDict(const py::dict& d)
+{
+    EngineConfig c;
+
+    auto getU64 = [&](const char* k, uint64_t def) -> uint64_t {
+        return d.contains(k) ? d[k].cast<uint64_t>() : def;
+    };
+    auto getI = [&](const char* k, int64_t def) -> int64_t {
+        return d.contains(k) ? d[k].cast<int64_t>() : def;
+    };
+    auto getB = [&](const char* k, bool def) -> bool {
+        return d.contains(k) ? d[k].cast<bool>() : def;
+    };
+    auto getS = [&](const char* k, std::string def) -> std::string {
+        return d.contains(k) ? d[k].cast<std::string>() : def;
+    };
+
+    if (d.contains("paths")) c.paths = d["paths"].cast<std::vector<std::string>>();
+    std::string pt = getS("path_type", "file");
+    c.pathType = (pt == "dir") ? PathType::DIR : (pt == "bdev") ? PathType::BLOCKDEV
+                                                                : PathType::FILE;
+
+    c.numThreads = (int)getI("threads", 1);
+    c.rankOffset = (int)getI("rank_offset", 0);
+    c.numDataSetThreads = (int)getI("num_dataset_threads", c.numThreads);
+    c.numDirs = getU64("dirs", 0);
+    c.numFiles = getU64("files", 0);
+    c.fileSize = getU64("file_size", 0);
+    c.blockSize = getU64("block_size", 1ULL << 20);
+    c.ioDepth = (int)getI("iodepth", 1);
+    c.directIO = getB("direct", false);
+    c.random = getB("random", false);
+    c.randAligned = getB("rand_aligned", true);
+    c.randAmount = getU64("rand_amount", 0);
+    c.strided = getB("strided", false);
+    c.backward = getB("backward", false);
+    c.truncate = getB("truncate", false);
+    c.truncToSize = d.contains("trunc_to_size") && !d["trunc_to_size"].is_none()
+                        ? d["trunc_to_size"].cast<uint64_t>()
+                        : UINT64_MAX;
+    c.preallocFile = getB("prealloc", false);
+    c.fsyncPerFile = getB("fsync", false);
+    c.verifySalt = getI("verify_salt", -1);
+    c.verifyDirect = getB("verify_direct", false);
+    c.blockVarPct = (int)getI("blockvar_pct", 100);
+    c.blockVarAlgo = getS("blockvar_algo", "fast");
+    c.randAlgo = getS("rand_algo", "balanced_single");
+    if (d.contains("gpu_ids")) c.gpuIDs = d["gpu_ids"].cast<std::vector<int>>();
+    c.gpuPinnedHostBufs = getB("gpu_pinned", true);
+    c.measureLat = getB("lat", false);
+    c.limitReadBps = getU64("limit_read_bps", 0);
+    c.limitWriteBps = getU64("limit_write_bps", 0);
+    c.ignoreDelErrors = getB("ignore_del_errors", false);
+    c.dirSharing = getB("dir_sharing", false);
+    c.infiniteLoop = getB("inf_loop", false);
+    c.benchSeed = getU64("bench_seed", 0x243F6A8885A308D3ULL);
+
+    return c;
+}
+
+py::dict resultToDict(const WorkerResult& r)
+{
+    py::dict d;
+    d["rank"] = r.rank;
+    d["elapsed_usec"] = r.elapsedUSec;
+    d["entries"] = r.total.entries;
+    d["bytes"] = r.total.bytes;
+    d["iops"] = r.total.iops;
+    d["stonewall_entries"] = r.stonewall.entries;
+    d["stonewall_bytes"] = r.stonewall.bytes;
+    d["stonewall_iops"] = r.stonewall.iops;
+    d["stonewall_elapsed_usec"] = r.stonewallElapsedUSec;
+    d["io_lat"] = r.ioLatVec;
+    d["entry_lat"] = r.entryLatVec;
+    d["error"] = r.error;
+    return d;
+}
+
+} // namespace
+
+PYBIND11_MODULE(_core, m)
+{
+    m.doc() = "elbencho_amd native I/O engine (MI355X / gfx950)";
+
+    m.def("gpu_device_count", &gpuDeviceCount);
+    m.def("gpu_device_name", &gpuDeviceName);
+
+    m.def("hist_bucket_lower_bound", &LatencyHistogram::bucketLowerBound);
+    m.def("hist_num_buckets", [] { return (int)LatencyHistogram::NUM_BUCKETS; });
+
+    // CPU checksum helpers exposed for tests (numerics parity with the GPU
+    // kernels is tested by comparing against these)
+    m.def("fill_checksum", [](uint64_t len, uint64_t fileOff, uint64_t salt) {
+        std::string buf(len, '\0');
+        fillChecksumCPU(buf.data(), len, fileOff, salt);
+        return py::bytes(buf);
+    });
+    m.def("verify_checksum", [](py::bytes data, uint64_t fileOff, uint64_t salt) {
+        std::string buf = data;
+        return verifyChecksumCPU(buf.data(), buf.size(), fileOff, salt);
+    });
+
+    py::class_<Engine>(m, "Engine")
+        .def(py::init([](const py::dict& cfg) {
+            return std::make_unique<Engine>(configFromDict(cfg));
+        }))
+        .def("prepare", &Engine::prepare, py::call_guard<py::gil_scoped_release>())
+        .def("start_phase",
+             [](Engine& e, int phaseCode) { e.startPhase((Phase)phaseCode); })
+        .def("wait_phase_done", &Engine::waitPhaseDone,
+             py::call_guard<py::gil_scoped_release>())
+        .def("interrupt", &Engine::interrupt)
+        .def("poll",
+             [](Engine& e) {
+                 Engine::LivePoll lp = e.poll();
+                 py::dict d;
+                 d["entries"] = lp.entries;
+                 d["bytes"] = lp.bytes;
+                 d["iops"] = lp.iops;
+                 d["workers_done"] = lp.workersDone;
+                 d["workers_total"] = lp.workersTotal;
+                 d["workers_with_error"] = lp.workersWithError;
+                 d["elapsed_usec"] = lp.elapsedUSec;
+                 d["stonewall_triggered"] = lp.stonewallTriggered;
+                 d["lat_num_ios"] = lp.latNumIOs;
+                 d["lat_sum_ios"] = lp.latSumIOs;
+                 d["lat_num_entries"] = lp.latNumEntries;
+                 d["lat_sum_entries"] = lp.latSumEntries;
+                 return d;
+             })
+        .def("finish_phase",
+             [](Engine& e) {
+                 std::vector<WorkerResult> rs;
+                 {
+                     py::gil_scoped_release rel;
+                     rs = e.finishPhase();
+                 }
+                 py::list out;
+                 for (auto& r : rs) out.append(resultToDict(r));
+                 return out;
+             })
+        .def("planned_work", [](Engine& e, int phaseCode) {
+            auto pw = e.plannedWork((Phase)phaseCode);
+            return py::make_tuple(pw.first, pw.second);
+        });
+
+    // phase code constants (wire-stable)
+    py::dict phases;
+    phases["IDLE"] = (int)Phase::IDLE;
+    phases["TERMINATE"] = (int)Phase::TERMINATE;
+    phases["MKDIRS"] = (int)Phase::MKDIRS;
+    phases["WRITE"] = (int)Phase::WRITE;
+    phases["READ"] = (int)Phase::READ;
+    phases["STAT"] = (int)Phase::STAT;
+    phases["RMFILES"] = (int)Phase::RMFILES;
+    phases["RMDIRS"] = (int)Phase::RMDIRS;
+    phases["SYNC"] = (int)Phase::SYNC;
+    phases["DROPCACHES"] = (int)Phase::DROPCACHES;
+    phases["NETBENCH"] = (int)Phase::NETBENCH;
+    m.attr("PHASES") = phases;
+}

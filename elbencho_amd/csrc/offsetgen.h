@@ -1,0 +1,269 @@
+// Offset stream generators for block I/O.
+//
+// Behavior parity with the reference's offset generator family
+// (/root/reference/source/toolkits/offsetgen/OffsetGenerator.h and
+// OffsetGenRandomAlignedFullCoverageV2.h): sequential, reverse-sequential,
+// random (unaligned), random block-aligned, random aligned full-coverage
+// (LCG permutation), and strided. Independent implementation.
+
+#pragma once
+
+#include <cstdint>
+#include <memory>
+
+#include "rand.h"
+
+namespace eb {
+
+struct BlockSpec {
+    uint64_t offset;
+    uint64_t len;
+};
+
+// A stream of (offset, len) blocks within one byte range.
+class OffsetGen {
+public:
+    virtual ~OffsetGen() = default;
+
+    // Reset for a new range (e.g. per file in dir mode).
+    virtual void reset(uint64_t rangeStart, uint64_t rangeLen) = 0;
+
+    // True if another block is available; fills spec.
+    virtual bool next(BlockSpec& spec) = 0;
+
+    // Total bytes this generator will produce for the current range.
+    virtual uint64_t totalBytes() const = 0;
+};
+
+class OffsetGenSequential final : public OffsetGen {
+public:
+    explicit OffsetGenSequential(uint64_t blockSize) : blockSize(blockSize) {}
+
+    void reset(uint64_t rangeStart, uint64_t rangeLen) override
+    {
+        pos = rangeStart;
+        end = rangeStart + rangeLen;
+        total = rangeLen;
+    }
+
+    bool next(BlockSpec& spec) override
+    {
+        if (pos >= end) return false;
+        spec.offset = pos;
+        spec.len = (end - pos < blockSize) ? (end - pos) : blockSize;
+        pos += spec.len;
+        return true;
+    }
+
+    uint64_t totalBytes() const override { return total; }
+
+private:
+    uint64_t blockSize;
+    uint64_t pos = 0, end = 0, total = 0;
+};
+
+// Backwards sequential: last block first (a possibly-short tail block is
+// emitted first, then full blocks walking toward rangeStart).
+class OffsetGenReverseSeq final : public OffsetGen {
+public:
+    explicit OffsetGenReverseSeq(uint64_t blockSize) : blockSize(blockSize) {}
+
+    void reset(uint64_t rangeStart, uint64_t rangeLen) override
+    {
+        start = rangeStart;
+        remaining = rangeLen;
+        total = rangeLen;
+    }
+
+    bool next(BlockSpec& spec) override
+    {
+        if (!remaining) return false;
+        uint64_t tail = remaining % blockSize;
+        spec.len = tail ? tail : blockSize;
+        remaining -= spec.len;
+        spec.offset = start + remaining;
+        return true;
+    }
+
+    uint64_t totalBytes() const override { return total; }
+
+private:
+    uint64_t blockSize;
+    uint64_t start = 0, remaining = 0, total = 0;
+};
+
+// Random unaligned offsets; emits `amount` bytes in blockSize pieces at
+// byte-granular random offsets (each block fully inside the range).
+class OffsetGenRandom final : public OffsetGen {
+public:
+    OffsetGenRandom(uint64_t blockSize, RandAlgo& algo, uint64_t amount)
+        : blockSize(blockSize), algo(algo), amount(amount) {}
+
+    void reset(uint64_t rangeStart, uint64_t rangeLen) override
+    {
+        start = rangeStart;
+        len = rangeLen;
+        left = amount ? amount : rangeLen;
+        if (blockSize > rangeLen) left = 0; // cannot place a block
+    }
+
+    bool next(BlockSpec& spec) override
+    {
+        if (!left) return false;
+        spec.len = (left < blockSize) ? left : blockSize;
+        uint64_t maxOff = len - spec.len;
+        spec.offset = start + (maxOff ? (algo.next() % (maxOff + 1)) : 0);
+        left -= spec.len;
+        return true;
+    }
+
+    uint64_t totalBytes() const override { return amount ? amount : len; }
+
+private:
+    uint64_t blockSize;
+    RandAlgo& algo;
+    uint64_t amount;
+    uint64_t start = 0, len = 0, left = 0;
+};
+
+// Random block-aligned offsets.
+class OffsetGenRandomAligned final : public OffsetGen {
+public:
+    OffsetGenRandomAligned(uint64_t blockSize, RandAlgo& algo, uint64_t amount)
+        : blockSize(blockSize), algo(algo), amount(amount) {}
+
+    void reset(uint64_t rangeStart, uint64_t rangeLen) override
+    {
+        start = rangeStart;
+        numBlocks = rangeLen / blockSize;
+        left = amount ? amount : (numBlocks * blockSize);
+        if (!numBlocks) left = 0;
+        total = left;
+    }
+
+    bool next(BlockSpec& spec) override
+    {
+        if (!left) return false;
+        spec.len = (left < blockSize) ? left : blockSize;
+        spec.offset = start + (algo.next() % numBlocks) * blockSize;
+        left -= spec.len;
+        return true;
+    }
+
+    uint64_t totalBytes() const override { return total; }
+
+private:
+    uint64_t blockSize;
+    RandAlgo& algo;
+    uint64_t amount;
+    uint64_t start = 0, numBlocks = 0, left = 0, total = 0;
+};
+
+// Random aligned with full coverage: every block of the range is visited
+// exactly once, in a pseudo-random permutation order.
+//
+// Design (independent; same contract as the reference's V2 generator): a
+// power-of-two-modulus LCG is a bijection over [0, 2^k); with 2^k >= numBlocks
+// we walk the LCG cycle and reject values >= numBlocks ("cycle walking").
+// Expected rejections < 1 per emitted block since 2^k < 2*numBlocks.
+class OffsetGenRandomAlignedFullCoverage final : public OffsetGen {
+public:
+    OffsetGenRandomAlignedFullCoverage(uint64_t blockSize, uint64_t seed)
+        : blockSize(blockSize), seed(seed) {}
+
+    void reset(uint64_t rangeStart, uint64_t rangeLen) override
+    {
+        start = rangeStart;
+        numBlocks = rangeLen / blockSize;
+        tailLen = rangeLen - numBlocks * blockSize;
+        emitted = 0;
+        total = rangeLen;
+
+        // modulus = smallest power of two >= numBlocks
+        mod = 1;
+        while (mod < numBlocks) mod <<= 1;
+        modMask = mod - 1;
+
+        // LCG x' = (a*x + c) mod 2^k is full-period iff a % 4 == 1 and c odd
+        // (Hull–Dobell). Derive multiplier/increment from the seed so
+        // different workers/iterations get different permutations.
+        uint64_t sm = seed;
+        a = (splitmix64(sm) & modMask & ~3ULL) | 1ULL; // a % 4 == 1
+        if (mod >= 4) a |= 0; // keep a < mod implicitly via mask
+        c = splitmix64(sm) | 1ULL; // odd
+        c &= modMask ? modMask : 0;
+        c |= 1ULL;
+        x = splitmix64(sm) & modMask;
+    }
+
+    bool next(BlockSpec& spec) override
+    {
+        if (emitted >= numBlocks) {
+            if (tailLen && !tailEmitted) { // short tail block, emitted last
+                tailEmitted = true;
+                spec.offset = start + numBlocks * blockSize;
+                spec.len = tailLen;
+                return true;
+            }
+            return false;
+        }
+        // walk the LCG until a value < numBlocks comes up
+        do {
+            x = (a * x + c) & modMask;
+        } while (x >= numBlocks);
+        spec.offset = start + x * blockSize;
+        spec.len = blockSize;
+        emitted++;
+        return true;
+    }
+
+    uint64_t totalBytes() const override { return total; }
+
+private:
+    uint64_t blockSize;
+    uint64_t seed;
+    uint64_t start = 0, numBlocks = 0, tailLen = 0, emitted = 0, total = 0;
+    bool tailEmitted = false;
+    uint64_t mod = 1, modMask = 0, a = 1, c = 1, x = 0;
+};
+
+// Strided: rank r of N dataset threads starts at r*blockSize and strides by
+// N*blockSize through the shared range (interleaves ranks blockwise).
+class OffsetGenStrided final : public OffsetGen {
+public:
+    OffsetGenStrided(uint64_t blockSize, uint64_t rank, uint64_t numRanks)
+        : blockSize(blockSize), rank(rank), numRanks(numRanks ? numRanks : 1) {}
+
+    void reset(uint64_t rangeStart, uint64_t rangeLen) override
+    {
+        start = rangeStart;
+        end = rangeStart + rangeLen;
+        pos = rangeStart + rank * blockSize;
+        stride = numRanks * blockSize;
+        // fair share of the range for this rank
+        uint64_t numBlocks = rangeLen / blockSize;
+        uint64_t myBlocks = numBlocks / numRanks + ((numBlocks % numRanks) > rank ? 1 : 0);
+        total = myBlocks * blockSize;
+        // rank handling the byte tail: the one whose stride position lands on it
+        uint64_t tail = rangeLen - numBlocks * blockSize;
+        if (tail && (numBlocks % numRanks) == rank) total += tail;
+    }
+
+    bool next(BlockSpec& spec) override
+    {
+        if (pos >= end) return false;
+        spec.offset = pos;
+        spec.len = (end - pos < blockSize) ? (end - pos) : blockSize;
+        pos += stride;
+        return true;
+    }
+
+    uint64_t totalBytes() const override { return total; }
+
+private:
+    uint64_t blockSize;
+    uint64_t rank, numRanks;
+    uint64_t start = 0, end = 0, pos = 0, stride = 0, total = 0;
+};
+
+} // namespace eb

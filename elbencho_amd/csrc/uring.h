@@ -1,0 +1,183 @@
+// Minimal raw io_uring wrapper (no liburing dependency).
+//
+// MI355X-native replacement for the reference's libaio engine
+// (/root/reference/source/workers/LocalWorker.cpp:1828-2070 aioBlockSized):
+// we use io_uring as the primary async engine per BASELINE.json's north star
+// ("libaio/io_uring"). Linux AIO is also provided (aio.h) as fallback.
+//
+// Self-contained: io_uring_setup/io_uring_enter syscalls + ring mmaps,
+// straight from the uapi contract in <linux/io_uring.h>.
+
+#pragma once
+
+#include <linux/io_uring.h>
+#include <sys/mman.h>
+#include <sys/syscall.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <cerrno>
+#include <cstdint>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+
+namespace eb {
+
+class IoUring {
+public:
+    struct Completion {
+        uint64_t userData;
+        int32_t res; // bytes or -errno
+    };
+
+    IoUring() = default;
+    IoUring(const IoUring&) = delete;
+    IoUring& operator=(const IoUring&) = delete;
+    ~IoUring() { destroy(); }
+
+    void init(unsigned entries)
+    {
+        struct io_uring_params p;
+        std::memset(&p, 0, sizeof(p));
+
+        ringFd = (int)syscall(__NR_io_uring_setup, entries, &p);
+        if (ringFd < 0)
+            throw std::runtime_error(std::string("io_uring_setup failed: ") + strerror(errno));
+
+        sqEntries = p.sq_entries;
+        cqEntries = p.cq_entries;
+
+        size_t sqRingSz = p.sq_off.array + p.sq_entries * sizeof(unsigned);
+        size_t cqRingSz = p.cq_off.cqes + p.cq_entries * sizeof(struct io_uring_cqe);
+
+        bool singleMmap = p.features & IORING_FEAT_SINGLE_MMAP;
+        if (singleMmap && cqRingSz > sqRingSz) sqRingSz = cqRingSz;
+
+        sqRing = mmap(nullptr, sqRingSz, PROT_READ | PROT_WRITE, MAP_SHARED | MAP_POPULATE,
+                      ringFd, IORING_OFF_SQ_RING);
+        if (sqRing == MAP_FAILED)
+            throw std::runtime_error("io_uring SQ ring mmap failed");
+        sqRingSize = sqRingSz;
+
+        if (singleMmap) {
+            cqRing = sqRing;
+            cqRingSize = 0; // shared mapping
+        } else {
+            cqRing = mmap(nullptr, cqRingSz, PROT_READ | PROT_WRITE, MAP_SHARED | MAP_POPULATE,
+                          ringFd, IORING_OFF_CQ_RING);
+            if (cqRing == MAP_FAILED)
+                throw std::runtime_error("io_uring CQ ring mmap failed");
+            cqRingSize = cqRingSz;
+        }
+
+        sqesSize = p.sq_entries * sizeof(struct io_uring_sqe);
+        sqes = (struct io_uring_sqe*)mmap(nullptr, sqesSize, PROT_READ | PROT_WRITE,
+                                          MAP_SHARED | MAP_POPULATE, ringFd, IORING_OFF_SQES);
+        if (sqes == MAP_FAILED)
+            throw std::runtime_error("io_uring SQE array mmap failed");
+
+        auto base = (char*)sqRing;
+        sqHead = (std::atomic<unsigned>*)(base + p.sq_off.head);
+        sqTail = (std::atomic<unsigned>*)(base + p.sq_off.tail);
+        sqMask = *(unsigned*)(base + p.sq_off.ring_mask);
+        sqArray = (unsigned*)(base + p.sq_off.array);
+
+        auto cbase = (char*)cqRing;
+        cqHead = (std::atomic<unsigned>*)(cbase + p.cq_off.head);
+        cqTail = (std::atomic<unsigned>*)(cbase + p.cq_off.tail);
+        cqMask = *(unsigned*)(cbase + p.cq_off.ring_mask);
+        cqes = (struct io_uring_cqe*)(cbase + p.cq_off.cqes);
+    }
+
+    bool valid() const { return ringFd >= 0; }
+
+    // Queue one read or write; does not submit to the kernel yet.
+    // Returns false if the SQ is full.
+    bool prep(bool isWrite, int fd, void* buf, uint64_t len, uint64_t fileOff, uint64_t userData)
+    {
+        unsigned tail = sqTail->load(std::memory_order_relaxed);
+        unsigned head = sqHead->load(std::memory_order_acquire);
+        if (tail - head >= sqEntries) return false;
+
+        unsigned idx = tail & sqMask;
+        struct io_uring_sqe* sqe = &sqes[idx];
+        std::memset(sqe, 0, sizeof(*sqe));
+        sqe->opcode = isWrite ? IORING_OP_WRITE : IORING_OP_READ;
+        sqe->fd = fd;
+        sqe->addr = (uint64_t)buf;
+        sqe->len = (uint32_t)len;
+        sqe->off = fileOff;
+        sqe->user_data = userData;
+
+        sqArray[idx] = idx;
+        sqTail->store(tail + 1, std::memory_order_release);
+        pending++;
+        return true;
+    }
+
+    // Submit queued SQEs; optionally wait for at least `waitNr` completions.
+    int submitAndWait(unsigned waitNr)
+    {
+        unsigned toSubmit = pending;
+        int ret = (int)syscall(__NR_io_uring_enter, ringFd, toSubmit, waitNr,
+                               waitNr ? IORING_ENTER_GETEVENTS : 0, nullptr, 0);
+        if (ret < 0) {
+            if (errno == EINTR) return 0;
+            throw std::runtime_error(std::string("io_uring_enter failed: ") + strerror(errno));
+        }
+        pending -= (unsigned)ret;
+        return ret;
+    }
+
+    // Reap up to maxEvents completions without blocking.
+    unsigned reap(Completion* out, unsigned maxEvents)
+    {
+        unsigned head = cqHead->load(std::memory_order_relaxed);
+        unsigned tail = cqTail->load(std::memory_order_acquire);
+        unsigned n = 0;
+        while (head != tail && n < maxEvents) {
+            const struct io_uring_cqe* cqe = &cqes[head & cqMask];
+            out[n].userData = cqe->user_data;
+            out[n].res = cqe->res;
+            n++;
+            head++;
+        }
+        cqHead->store(head, std::memory_order_release);
+        return n;
+    }
+
+    unsigned entries() const { return sqEntries; }
+
+    void destroy()
+    {
+        if (sqes && sqes != MAP_FAILED) munmap(sqes, sqesSize);
+        if (cqRing && cqRing != MAP_FAILED && cqRingSize) munmap(cqRing, cqRingSize);
+        if (sqRing && sqRing != MAP_FAILED) munmap(sqRing, sqRingSize);
+        if (ringFd >= 0) close(ringFd);
+        sqes = nullptr;
+        cqRing = nullptr;
+        sqRing = nullptr;
+        ringFd = -1;
+    }
+
+private:
+    int ringFd = -1;
+    unsigned sqEntries = 0, cqEntries = 0, pending = 0;
+    void* sqRing = nullptr;
+    void* cqRing = nullptr;
+    struct io_uring_sqe* sqes = nullptr;
+    size_t sqRingSize = 0, cqRingSize = 0, sqesSize = 0;
+
+    std::atomic<unsigned>* sqHead = nullptr;
+    std::atomic<unsigned>* sqTail = nullptr;
+    unsigned sqMask = 0;
+    unsigned* sqArray = nullptr;
+
+    std::atomic<unsigned>* cqHead = nullptr;
+    std::atomic<unsigned>* cqTail = nullptr;
+    unsigned cqMask = 0;
+    struct io_uring_cqe* cqes = nullptr;
+};
+
+} // namespace eb
