@@ -1,0 +1,198 @@
+#!/usr/bin/env python3
+"""Flagship benchmark step for the driver contract.
+
+Measures the headline metric of BASELINE.json: sequential-read GiB/s into
+GPU HBM (config: "Large-file seq read, 16 threads, 4 MiB blocks, --gpuids N
+hipMemcpyAsync into HBM on MI355X") on synthetic files on tmpfs.
+
+One "step" = one full sequential READ pass of this rank's dataset into the
+rank's GPU HBM through the native engine (16 C++ I/O threads, pinned host
+buffers, hipMemcpyAsync staging on per-thread streams). Multi-GPU runs are
+launched by the driver as one rank per GPU via torch.distributed.run; ranks
+sync with RCCL barriers around the timed region and the slowest rank's time
+is used (MAX all-reduce).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import shutil
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+BASELINE_GIBS = 11.0  # reference's best recorded seq write: 94.46 Gbps ~= 11.0 GiB/s
+                      # (BASELINE.md, 8x NVMe RAID-0 testbed, other hardware)
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=5)
+    p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--threads", type=int, default=int(os.environ.get("EB_BENCH_THREADS", "16")))
+    p.add_argument("--block", type=int,
+                   default=int(os.environ.get("EB_BENCH_BLOCK", str(4 * 1024 * 1024))))
+    p.add_argument("--filesize", type=int,
+                   default=int(os.environ.get("EB_BENCH_FILESIZE", str(2 * 1024 ** 3))))
+    p.add_argument("--dir", default=os.environ.get("EB_BENCH_DIR", "/dev/shm/elbencho_amd_bench"))
+    p.add_argument("--workload", default="seqread", choices=["seqread", "seqwrite", "randread"])
+    p.add_argument("--iodepth", type=int, default=int(os.environ.get("EB_BENCH_IODEPTH", "1")))
+    return p.parse_args()
+
+
+def main() -> int:
+    args = parse_args()
+
+    import torch
+
+    from elbencho_amd import load_core
+
+    core = load_core()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    use_gpu = torch.cuda.is_available()
+    if use_gpu and core.gpu_device_count() < 1:
+        raise RuntimeError("torch sees a GPU but the HIP engine does not — broken build")
+
+    sync = None
+    if world > 1:
+        from elbencho_amd import parallel
+
+        sync = parallel.init_from_env()
+
+    device = None
+    if use_gpu:
+        import torch
+
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+
+    # --- per-rank dataset on tmpfs ---
+    os.makedirs(args.dir, exist_ok=True)
+    path = os.path.join(args.dir, f"bench_r{rank}.bin")
+
+    base_cfg = dict(
+        path_type="file",
+        threads=args.threads,
+        num_dataset_threads=args.threads,  # each rank owns its file entirely
+        rank_offset=0,
+        file_size=args.filesize,
+        block_size=args.block,
+        iodepth=args.iodepth,
+        lat=False,
+        blockvar_pct=0,  # setup fill is random already; steps measure I/O, not RNG
+        bench_seed=0x9E3779B97F4A7C15 ^ rank,
+    )
+
+    # setup: create the synthetic file (not timed)
+    wcfg = dict(base_cfg, paths=[path])
+    weng = core.Engine(wcfg)
+    weng.prepare()
+
+    def run_pass(eng, phase):
+        eng.start_phase(core.PHASES[phase])
+        eng.wait_phase_done(-1)
+        res = eng.finish_phase()
+        errs = [r["error"] for r in res if r["error"]]
+        if errs:
+            raise RuntimeError(f"bench phase failed: {errs}")
+        return sum(r["bytes"] for r in res)
+
+    run_pass(weng, "WRITE")
+    assert os.path.getsize(path) == args.filesize
+
+    # measured engine: GPU-staged when a GPU is present
+    mcfg = dict(base_cfg, paths=[path])
+    if use_gpu:
+        mcfg["gpu_ids"] = [local_rank]
+    if args.workload == "randread":
+        mcfg["random"] = True
+        mcfg["block_size"] = 4096
+    meng = core.Engine(mcfg)
+    meng.prepare()
+
+    phase = "WRITE" if args.workload == "seqwrite" else "READ"
+
+    # --- warmup ---
+    for _ in range(args.warmup):
+        run_pass(meng, phase)
+
+    # --- timed steps, barrier + device sync on both sides ---
+    if sync:
+        sync.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+
+    bytes_done = 0
+    for _ in range(args.steps):
+        bytes_done += run_pass(meng, phase)
+
+    if use_gpu:
+        torch.cuda.synchronize()
+    if sync:
+        sync.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if sync:  # slowest rank defines the job time
+        elapsed = sync.allreduce_max([elapsed])[0]
+        total_bytes = sync.allreduce_sum([float(bytes_done)])[0]
+    else:
+        total_bytes = float(bytes_done)
+
+    value = total_bytes / elapsed / (1024 ** 3)  # whole-job GiB/s
+    ms_per_step = elapsed * 1000.0 / args.steps
+
+    if rank == 0:
+        doc = {
+            "metric": "seq-read GiB/s into GPU HBM",
+            "value": round(value, 3),
+            "unit": "GiB/s",
+            "n_gpus": world if use_gpu else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / BASELINE_GIBS, 3),
+            "dtype": "raw-bytes",
+            "data": "synthetic (tmpfs files, random fill)",
+            "config": {
+                "model": "storage-benchmark seq-read into HBM",
+                "workload": args.workload,
+                "threads_per_gpu": args.threads,
+                "block_size": args.block,
+                "file_size_per_rank": args.filesize,
+                "iodepth": args.iodepth,
+                "bench_dir": args.dir,
+                "gpu_staged": use_gpu,
+                "parallelism": f"dp{world}" if world > 1 else "single",
+            },
+        }
+        print(json.dumps(doc), flush=True)
+
+    # cleanup (tmpfs is shared RAM — do not leak multi-GiB files)
+    try:
+        os.unlink(path)
+    except OSError:
+        pass
+
+    if sync:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,487 @@
+"""Command line interface — elbencho-compatible option names.
+
+The option vocabulary matches the reference's CLI (see
+/root/reference/docs/usage/help-all.md; option registry ProgArgs.cpp:216-860)
+so existing elbencho invocations and wrapper scripts work unchanged.
+Independent implementation on argparse.
+"""
+
+from __future__ import annotations
+
+import argparse
+import sys
+
+from elbencho_amd import VERSION
+from elbencho_amd.config import BenchConfig, ConfigError, parse_gpu_ids
+from elbencho_amd.units import parse_size
+
+
+class _HelpFormatter(argparse.HelpFormatter):
+    def __init__(self, prog):
+        super().__init__(prog, max_help_position=28, width=100)
+
+
+def build_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(
+        prog="elbencho-amd",
+        description="elbencho-amd - A distributed storage benchmark for files, objects and "
+                    "blocks, built MI355X-native (HIP/gfx950 + RCCL over xGMI).",
+        formatter_class=_HelpFormatter,
+        add_help=True,
+    )
+
+    p.add_argument("paths", nargs="*", metavar="PATH",
+                   help="Benchmark paths (files, block devices, directories or s3:// buckets). "
+                        "Supports bracket expansion, e.g. /mnt/file[1-4].")
+
+    g = p.add_argument_group("benchmark phases")
+    g.add_argument("-w", "--write", action="store_true", help="Write/create files or objects.")
+    g.add_argument("-r", "--read", action="store_true", help="Read files or objects.")
+    g.add_argument("--stat", action="store_true", help="Stat files / HEAD objects.")
+    g.add_argument("-F", "--delfiles", action="store_true", help="Delete files/objects.")
+    g.add_argument("-d", "--mkdirs", action="store_true", help="Create directories.")
+    g.add_argument("-D", "--deldirs", action="store_true", help="Delete directories.")
+    g.add_argument("--sync", action="store_true",
+                   help="Sync dirty page cache before/after each phase.")
+    g.add_argument("--dropcache", action="store_true",
+                   help="Drop page/dentry/inode caches before/after each phase (needs root).")
+
+    g = p.add_argument_group("workload")
+    g.add_argument("-t", "--threads", type=int, default=1, metavar="N",
+                   help="Number of I/O worker threads. (Default: 1)")
+    g.add_argument("-n", "--dirs", type=int, default=0, metavar="N",
+                   help="Number of directories per thread (dir mode).")
+    g.add_argument("-N", "--files", type=int, default=0, metavar="N",
+                   help="Number of files per directory, per thread (dir mode).")
+    g.add_argument("-s", "--size", default="0", metavar="SIZE",
+                   help="File/object size (supports base2 suffixes, e.g. \"4M\").")
+    g.add_argument("-b", "--block", default="1M", metavar="SIZE",
+                   help="Bytes per read/write operation. (Default: 1M)")
+    g.add_argument("--iodepth", type=int, default=1, metavar="N",
+                   help="Depth of I/O queue per thread via io_uring. (Default: 1)")
+    g.add_argument("-i", "--iterations", type=int, default=1, metavar="N",
+                   help="Repeat all benchmark phases N times. (Default: 1)")
+    g.add_argument("--infloop", action="store_true",
+                   help="Let I/O threads restart their work until interrupted.")
+
+    g = p.add_argument_group("access pattern")
+    g.add_argument("--direct", action="store_true", help="Use direct IO (O_DIRECT).")
+    g.add_argument("--rand", action="store_true", help="Random offsets.")
+    g.add_argument("--norandalign", action="store_true",
+                   help="Do not align random offsets to block size.")
+    g.add_argument("--randamount", default="0", metavar="SIZE",
+                   help="Total amount for random I/O across all threads. "
+                        "(Default: full file/device size)")
+    g.add_argument("--randalgo", default="balanced_single", metavar="ALGO",
+                   help="Random offset algorithm: fast, balanced_single, balanced, strong.")
+    g.add_argument("--strided", action="store_true",
+                   help="Interleave threads blockwise over shared files.")
+    g.add_argument("--backward", action="store_true", help="Backwards sequential access.")
+    g.add_argument("--trunc", action="store_true", help="Truncate files to 0 before writing.")
+    g.add_argument("--trunctosize", action="store_true",
+                   help="Truncate files to full size before writing.")
+    g.add_argument("--preallocfile", action="store_true",
+                   help="Preallocate file space via fallocate before writing.")
+    g.add_argument("--sharesize", default="0", metavar="SIZE",
+                   help="Custom tree: files larger than this are shared between threads. "
+                        "(Default: 0)")
+
+    g = p.add_argument_group("data integrity & variance")
+    g.add_argument("--verify", type=int, default=-1, metavar="SALT",
+                   help="Write/check offset-based block checksums with the given salt.")
+    g.add_argument("--verifydirect", action="store_true",
+                   help="Verify each block by read-back immediately after writing.")
+    g.add_argument("--readinline", action="store_true",
+                   help="Read files immediately after writing in the same phase.")
+    g.add_argument("--blockvarpct", type=int, default=100, metavar="PCT",
+                   help="Percentage of each block refilled with random data between writes. "
+                        "(Default: 100)")
+    g.add_argument("--blockvaralgo", default="fast", metavar="ALGO",
+                   help="RNG for --blockvarpct: fast, balanced, strong. On GPUs the "
+                        "gfx950 xoshiro256++ fill kernel is used. (Default: fast)")
+
+    g = p.add_argument_group("GPU (MI355X)")
+    g.add_argument("--gpuids", default="", metavar="IDS",
+                   help="Comma-separated HIP device ids; buffers live in HBM3E and are "
+                        "staged via hipMemcpyAsync on per-thread streams (round-robin "
+                        "thread assignment).")
+    g.add_argument("--gpuperservice", action="store_true",
+                   help="Assign GPUs round-robin to service instances instead of threads.")
+    g.add_argument("--gds", action="store_true",
+                   help="Direct storage<->HBM path: O_DIRECT + pinned staging buffers "
+                        "(ROCm has no cuFile; this is the MI355X-native equivalent).")
+    g.add_argument("--cuhostbufreg", action="store_true",
+                   help="Pin host I/O buffers for DMA transfers (default on when GPUs "
+                        "are used; kept for CLI compatibility).")
+    g.add_argument("--cufile", action="store_true", help=argparse.SUPPRESS)
+    g.add_argument("--cufiledriveropen", action="store_true", help=argparse.SUPPRESS)
+    g.add_argument("--gdsbufreg", action="store_true", help=argparse.SUPPRESS)
+
+    g = p.add_argument_group("results & latency")
+    g.add_argument("--lat", action="store_true", help="Measure IO and entry latencies.")
+    g.add_argument("--lathisto", action="store_true", help="Show latency histogram buckets.")
+    g.add_argument("--latpercent", action="store_true", help="Show latency percentiles.")
+    g.add_argument("--latpercent9s", type=int, default=0, metavar="N",
+                   help="Number of nines after p99 to show (e.g. 2 -> p99.9, p99.99).")
+    g.add_argument("--allelapsed", action="store_true",
+                   help="Show elapsed time of each worker thread.")
+    g.add_argument("--cpu", action="store_true", help="Show CPU utilization.")
+    g.add_argument("--csvfile", default="", metavar="PATH", help="Append results as CSV.")
+    g.add_argument("--jsonfile", default="", metavar="PATH",
+                   help="Append results as JSON lines.")
+    g.add_argument("--resfile", default="", metavar="PATH",
+                   help="Append human-readable results to file.")
+    g.add_argument("--nocsvlabels", action="store_true",
+                   help="Do not write the header line to new CSV files.")
+    g.add_argument("--label", default="", metavar="STR",
+                   help="Custom label stored with the results.")
+    g.add_argument("--dirstats", action="store_true",
+                   help="Show dirs/s in dir-mode file write/read phases.")
+    g.add_argument("--liveint", type=int, default=2000, metavar="MS",
+                   help="Update interval of live statistics in ms. (Default: 2000)")
+    g.add_argument("--nolive", action="store_true", help="Disable live statistics.")
+    g.add_argument("--live1", action="store_true",
+                   help="Single-line live statistics. (Default on TTY)")
+    g.add_argument("--live1n", action="store_true",
+                   help="Live statistics on a new line per update.")
+    g.add_argument("--livecsv", default="", metavar="PATH",
+                   help="Stream live statistics to a CSV file.")
+    g.add_argument("--livecsvex", action="store_true",
+                   help="Extended live CSV (per-worker rows).")
+    g.add_argument("--opslog", default="", metavar="PATH",
+                   help="Log every I/O operation as JSON lines.")
+    g.add_argument("--opsloglock", action="store_true",
+                   help="Serialize ops log writes across processes via flock.")
+    g.add_argument("--log", type=int, default=0, metavar="LEVEL",
+                   help="Log level: 0 normal, 1 verbose, 2 debug.")
+
+    g = p.add_argument_group("limits & timing")
+    g.add_argument("--timelimit", type=int, default=0, metavar="SECS",
+                   help="Time limit per phase in seconds.")
+    g.add_argument("--limitread", default="0", metavar="RATE",
+                   help="Read bandwidth limit per thread (bytes/s, supports suffixes).")
+    g.add_argument("--limitwrite", default="0", metavar="RATE",
+                   help="Write bandwidth limit per thread.")
+    g.add_argument("--phasedelay", type=int, default=0, metavar="SECS",
+                   help="Delay between phases in seconds.")
+    g.add_argument("--start", type=int, default=0, metavar="EPOCHSECS",
+                   help="Synchronized start time (UNIX epoch seconds).")
+
+    g = p.add_argument_group("error handling & checks")
+    g.add_argument("--nodelerr", action="store_true",
+                   help="Do not treat deletion of non-existing entries as error.")
+    g.add_argument("--no0usecerr", action="store_true",
+                   help="Do not warn on sub-microsecond phase completion.")
+    g.add_argument("--nodiocheck", action="store_true",
+                   help="Skip direct IO alignment checks.")
+    g.add_argument("--nopathexp", action="store_true",
+                   help="Disable bracket expansion of paths.")
+    g.add_argument("--dryrun", action="store_true",
+                   help="Show planned work per phase without doing I/O.")
+
+    g = p.add_argument_group("CPU/NUMA placement")
+    g.add_argument("--zones", default="", metavar="LIST",
+                   help="Comma-separated NUMA zones to bind threads to (round-robin).")
+    g.add_argument("--zone", dest="zones", help=argparse.SUPPRESS)
+    g.add_argument("--cores", default="", metavar="LIST",
+                   help="Comma-separated CPU cores to bind threads to (round-robin).")
+    g.add_argument("--core", dest="cores", help=argparse.SUPPRESS)
+
+    g = p.add_argument_group("distributed mode")
+    g.add_argument("--hosts", default="", metavar="LIST",
+                   help="Comma-separated service hosts (host[:port], supports bracket "
+                        "ranges) to run as master against.")
+    g.add_argument("--hostsfile", default="", metavar="PATH",
+                   help="File with one service host per line.")
+    g.add_argument("--service", action="store_true",
+                   help="Run as service for distributed mode (control via master).")
+    g.add_argument("--port", type=int, default=1611, metavar="N",
+                   help="TCP port of service. (Default: 1611)")
+    g.add_argument("--foreground", action="store_true",
+                   help="Run service in foreground instead of daemonizing.")
+    g.add_argument("--nosvcshare", action="store_true",
+                   help="Benchmark paths are not shared between service hosts.")
+    g.add_argument("--numhosts", type=int, default=-1, metavar="N",
+                   help="Use only the first N hosts of --hosts.")
+    g.add_argument("--rotatehosts", type=int, default=0, metavar="N",
+                   help="Rotate hosts list by N between phases.")
+    g.add_argument("--svcupint", type=int, default=500, metavar="MS",
+                   help="Service status poll interval in ms. (Default: 500)")
+    g.add_argument("--svcelapsed", action="store_true",
+                   help="Show elapsed time per service host.")
+    g.add_argument("--svcpwfile", default="", metavar="PATH",
+                   help="Shared-secret file to authorize master<->service communication.")
+    g.add_argument("--svcwait", action="store_true", help=argparse.SUPPRESS)
+    g.add_argument("--svcping", action="store_true", help=argparse.SUPPRESS)
+    g.add_argument("--interrupt", action="store_true",
+                   help="Interrupt the current phase on the given service hosts.")
+    g.add_argument("--quit", action="store_true",
+                   help="Tell service hosts to quit.")
+    g.add_argument("--rankoffset", type=int, default=0, metavar="N",
+                   help="Offset for worker thread ranks (standalone mode).")
+    g.add_argument("--dirsharing", action="store_true",
+                   help="Threads share the dirs of rank 0 in dir mode.")
+    g.add_argument("--nofdsharing", action="store_true",
+                   help="Each thread opens its own file descriptors in file/bdev mode.")
+
+    g = p.add_argument_group("custom tree")
+    g.add_argument("--treefile", default="", metavar="PATH",
+                   help="Benchmark a custom tree of dirs/files defined in this file.")
+    g.add_argument("--treeroundup", default="0", metavar="SIZE",
+                   help="Round file sizes in the tree file up to a multiple of this size.")
+    g.add_argument("--treescan", default="", metavar="PATH",
+                   help="Scan an existing directory tree into a tree file.")
+    g.add_argument("--treerand", action="store_true",
+                   help="Process custom tree files in random order.")
+
+    g = p.add_argument_group("mixed read/write")
+    g.add_argument("--rwmixpct", type=int, default=0, metavar="PCT",
+                   help="Percentage of blocks to read instead of write in a write phase.")
+    g.add_argument("--rwmixthr", type=int, default=0, metavar="N",
+                   help="Number of threads of a write phase dedicated to reads.")
+    g.add_argument("--rwmixthrpct", type=int, default=0, metavar="PCT",
+                   help="Percentage of threads of a write phase dedicated to reads.")
+
+    g = p.add_argument_group("network benchmark")
+    g.add_argument("--netbench", action="store_true",
+                   help="Network benchmark between services (first given paths are "
+                        "servers; requires --servers).")
+    g.add_argument("--servers", default="", metavar="LIST",
+                   help="Comma-separated netbench server hosts.")
+    g.add_argument("--serversfile", default="", metavar="PATH",
+                   help="File with one netbench server per line.")
+    g.add_argument("--respsize", default="1", metavar="SIZE",
+                   help="Netbench server response size per received block. (Default: 1)")
+    g.add_argument("--sendbuf", default="0", metavar="SIZE", help="Socket send buffer size.")
+    g.add_argument("--recvbuf", default="0", metavar="SIZE", help="Socket recv buffer size.")
+    g.add_argument("--netdevs", default="", metavar="LIST",
+                   help="Comma-separated network devices to bind outgoing conns to.")
+
+    g = p.add_argument_group("S3 object storage")
+    g.add_argument("--s3endpoints", default="", metavar="LIST",
+                   help="Comma-separated S3 endpoint URLs.")
+    g.add_argument("--s3key", default="", metavar="KEY", help="S3 access key.")
+    g.add_argument("--s3secret", default="", metavar="SECRET", help="S3 secret key.")
+    g.add_argument("--s3region", default="", metavar="REGION", help="S3 region.")
+    g.add_argument("--s3objprefix", default="", metavar="STR", help="S3 object name prefix.")
+    g.add_argument("--s3randobj", action="store_true",
+                   help="Read at random offsets of random objects.")
+    g.add_argument("--s3listobj", type=int, default=0, metavar="N",
+                   help="List up to N objects per bucket.")
+    g.add_argument("--s3listverify", action="store_true",
+                   help="Verify object listing completeness.")
+    g.add_argument("--s3multidel", type=int, default=0, metavar="N",
+                   help="Delete objects in multi-delete batches of N.")
+    g.add_argument("--s3fastget", action="store_true",
+                   help="Discard downloaded objects instead of keeping them in RAM.")
+    g.add_argument("--s3fastput", action="store_true", help=argparse.SUPPRESS)
+    g.add_argument("--s3sign", type=int, default=0, help=argparse.SUPPRESS)
+    g.add_argument("--s3maxconns", type=int, default=0, help=argparse.SUPPRESS)
+    g.add_argument("--s3ignoreerrors", action="store_true", help=argparse.SUPPRESS)
+    g.add_argument("--s3nocompress", action="store_true", help=argparse.SUPPRESS)
+
+    g = p.add_argument_group("misc")
+    g.add_argument("-c", "--configfile", default="", metavar="PATH",
+                   help="Read options from a config file (key=value lines).")
+    g.add_argument("--version", action="version",
+                   version=f"elbencho-amd {VERSION} (MI355X/gfx950 native)")
+
+    return p
+
+
+def args_to_config(args: argparse.Namespace) -> BenchConfig:
+    cfg = BenchConfig()
+    cfg.paths = list(args.paths)
+    cfg.run_write = args.write
+    cfg.run_read = args.read
+    cfg.run_stat = args.stat
+    cfg.run_delfiles = args.delfiles
+    cfg.run_mkdirs = args.mkdirs
+    cfg.run_deldirs = args.deldirs
+    cfg.run_sync = args.sync
+    cfg.run_dropcaches = args.dropcache
+
+    cfg.threads = args.threads
+    cfg.dirs = args.dirs
+    cfg.files = args.files
+    cfg.file_size = parse_size(args.size)
+    cfg.block_size = parse_size(args.block)
+    cfg.iodepth = args.iodepth
+    cfg.iterations = args.iterations
+    cfg.inf_loop = args.infloop
+
+    cfg.direct = args.direct
+    cfg.random = args.rand
+    cfg.rand_aligned = not args.norandalign
+    cfg.rand_amount = parse_size(args.randamount)
+    cfg.rand_algo = args.randalgo
+    cfg.strided = args.strided
+    cfg.backward = args.backward
+    cfg.truncate = args.trunc
+    cfg.trunc_to_size = args.trunctosize
+    cfg.prealloc = args.preallocfile
+    cfg.sharesize = parse_size(args.sharesize)
+
+    cfg.verify = args.verify
+    cfg.verify_direct = args.verifydirect
+    cfg.read_inline = args.readinline
+    cfg.blockvar_pct = args.blockvarpct
+    cfg.blockvar_algo = args.blockvaralgo
+
+    cfg.gpu_ids = parse_gpu_ids(args.gpuids)
+    cfg.gpu_per_service = args.gpuperservice
+    cfg.gpu_direct = args.gds
+    if args.gds:
+        cfg.direct = True  # direct storage<->HBM path implies O_DIRECT
+
+    cfg.lat = args.lat or args.lathisto or args.latpercent
+    cfg.lat_histo = args.lathisto
+    cfg.lat_percent = args.latpercent
+    cfg.lat_percent_9s = args.latpercent9s
+    cfg.all_elapsed = args.allelapsed
+    cfg.cpu_util = args.cpu
+    cfg.csv_file = args.csvfile
+    cfg.json_file = args.jsonfile
+    cfg.res_file = args.resfile
+    cfg.no_csv_labels = args.nocsvlabels
+    cfg.label = args.label
+    cfg.show_dir_stats = args.dirstats
+    cfg.live_int_ms = args.liveint
+    cfg.no_live = args.nolive
+    cfg.live1 = args.live1
+    cfg.live_csv = args.livecsv
+    cfg.live_csv_ex = args.livecsvex
+    cfg.ops_log_path = args.opslog
+    cfg.ops_log_lock = args.opsloglock
+    cfg.log_level = args.log
+
+    cfg.timelimit = args.timelimit
+    cfg.limit_read = parse_size(args.limitread)
+    cfg.limit_write = parse_size(args.limitwrite)
+    cfg.phase_delay_secs = args.phasedelay
+    cfg.start_time = args.start
+
+    cfg.ignore_del_errors = args.nodelerr
+    cfg.ignore_0usec_errors = args.no0usecerr
+    cfg.no_dio_check = args.nodiocheck
+    cfg.no_path_expansion = args.nopathexp
+    cfg.dryrun = args.dryrun
+
+    cfg.numa_zones = args.zones or ""
+    cfg.cpu_cores = args.cores or ""
+
+    hosts = []
+    if args.hostsfile:
+        with open(args.hostsfile) as f:
+            hosts = [ln.strip() for ln in f if ln.strip() and not ln.startswith("#")]
+    elif args.hosts:
+        hosts = [h for h in args.hosts.split(",") if h]
+    cfg.hosts = hosts
+    cfg.service_mode = args.service
+    cfg.service_port = args.port
+    cfg.foreground = args.foreground
+    cfg.no_svc_share = args.nosvcshare
+    cfg.num_hosts = args.numhosts
+    cfg.rotate_hosts = args.rotatehosts
+    cfg.svc_update_int_ms = args.svcupint
+    cfg.svc_elapsed = args.svcelapsed
+    cfg.svc_pw_file = args.svcpwfile
+    cfg.interrupt_services = args.interrupt
+    cfg.quit_services = args.quit
+    cfg.rank_offset = args.rankoffset
+    cfg.dir_sharing = args.dirsharing
+    cfg.no_fd_sharing = args.nofdsharing
+
+    cfg.treefile = args.treefile
+    cfg.tree_round_up = parse_size(args.treeroundup)
+    cfg.tree_scan = args.treescan
+    cfg.tree_rand = args.treerand
+
+    cfg.rwmix_pct = args.rwmixpct
+    cfg.rwmix_threads = args.rwmixthr
+    cfg.rwmix_thread_pct = args.rwmixthrpct
+
+    cfg.netbench = args.netbench
+    cfg.servers = [s for s in args.servers.split(",") if s] if args.servers else []
+    if args.serversfile:
+        with open(args.serversfile) as f:
+            cfg.servers = [ln.strip() for ln in f if ln.strip()]
+    cfg.resp_size = parse_size(args.respsize)
+    cfg.send_buf = parse_size(args.sendbuf)
+    cfg.recv_buf = parse_size(args.recvbuf)
+    cfg.netdevs = [x for x in args.netdevs.split(",") if x] if args.netdevs else []
+
+    cfg.s3_endpoints = [x for x in args.s3endpoints.split(",") if x] if args.s3endpoints else []
+    cfg.s3_key = args.s3key
+    cfg.s3_secret = args.s3secret
+    cfg.s3_region = args.s3region
+    cfg.s3_obj_prefix = args.s3objprefix
+    cfg.s3_rand_obj = args.s3randobj
+    cfg.s3_list_obj = args.s3listobj
+    cfg.s3_list_verify = args.s3listverify
+    cfg.s3_multi_del = args.s3multidel
+    cfg.s3_fastget = args.s3fastget
+
+    cfg.config_file = args.configfile
+
+    cfg.finalize()
+    return cfg
+
+
+def apply_config_file(argv: list[str]) -> list[str]:
+    """Prepend options from -c/--configfile (key=value lines) to argv."""
+    cfgpath = None
+    for i, a in enumerate(argv):
+        if a in ("-c", "--configfile") and i + 1 < len(argv):
+            cfgpath = argv[i + 1]
+        elif a.startswith("--configfile="):
+            cfgpath = a.split("=", 1)[1]
+    if not cfgpath:
+        return argv
+    extra: list[str] = []
+    with open(cfgpath) as f:
+        for ln in f:
+            ln = ln.strip()
+            if not ln or ln.startswith("#"):
+                continue
+            if "=" in ln:
+                k, v = ln.split("=", 1)
+                k, v = k.strip(), v.strip()
+                if v.lower() == "true":
+                    extra.append(f"--{k}")
+                elif v.lower() == "false":
+                    pass
+                else:
+                    extra.extend([f"--{k}", v])
+            else:
+                extra.append(f"--{ln}")
+    # command line options override config file (argparse: later wins)
+    return extra + argv
+
+
+def main(argv: list[str] | None = None) -> int:
+    argv = list(sys.argv[1:] if argv is None else argv)
+    argv = apply_config_file(argv)
+    parser = build_parser()
+    args = parser.parse_args(argv)
+
+    try:
+        cfg = args_to_config(args)
+    except ConfigError as e:
+        print(f"ERROR: {e}", file=sys.stderr)
+        return 1
+
+    from elbencho_amd.coordinator import Coordinator
+
+    try:
+        return Coordinator(cfg).main()
+    except ConfigError as e:
+        print(f"ERROR: {e}", file=sys.stderr)
+        return 1
+    except RuntimeError as e:
+        print(f"ERROR: {e}", file=sys.stderr)
+        return 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,383 @@
+"""Benchmark configuration — the ProgArgs equivalent.
+
+Holds the full option surface (CLI names match the reference elbencho,
+/root/reference/docs/usage/help-all.md), performs unit parsing, bracket
+expansion of paths, path-type inference, cross-option validation, and
+serializes to/from a JSON-able dict for master->service transfer (the
+config system IS the wire schema, like reference ProgArgs.cpp:3921/:3754).
+Independent implementation.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import os
+import re
+import stat as statmod
+from dataclasses import dataclass, field
+from typing import Any, Optional
+
+from elbencho_amd.units import parse_size
+
+# benchmark path type
+PATH_DIR = "dir"
+PATH_FILE = "file"
+PATH_BDEV = "bdev"
+
+S3_PATH_PREFIX = "s3://"
+
+
+class ConfigError(ValueError):
+    pass
+
+
+def expand_path_brackets(path: str) -> list[str]:
+    """Expand "name[1-4]" / "name[3,5]" style patterns (one bracket pair),
+    mirroring the reference's path expansion (ProgArgs.cpp:1805)."""
+    m = re.search(r"\[([0-9,\-]+)\]", path)
+    if not m:
+        return [path]
+    nums: list[int] = []
+    for part in m.group(1).split(","):
+        if "-" in part:
+            lo, hi = part.split("-", 1)
+            nums.extend(range(int(lo), int(hi) + 1))
+        else:
+            nums.append(int(part))
+    pre, post = path[: m.start()], path[m.end() :]
+    out = []
+    for n in nums:
+        out.extend(expand_path_brackets(f"{pre}{n}{post}"))
+    return out
+
+
+def expand_hosts(hosts: list[str]) -> list[str]:
+    """Expand host lists with port ranges: "localhost:[1711-1712]"."""
+    out = []
+    for h in hosts:
+        out.extend(expand_path_brackets(h))
+    return out
+
+
+@dataclass
+class BenchConfig:
+    # --- paths & mode ---
+    paths: list[str] = field(default_factory=list)
+    path_type: str = PATH_FILE  # dir|file|bdev (auto-inferred)
+    bench_mode: str = "posix"  # posix|s3|netbench
+
+    # --- phases ---
+    run_mkdirs: bool = False       # -d/--mkdirs
+    run_write: bool = False        # -w/--write
+    run_read: bool = False         # -r/--read
+    run_stat: bool = False         # --stat
+    run_delfiles: bool = False     # -F/--delfiles
+    run_deldirs: bool = False      # -D/--deldirs
+    run_sync: bool = False         # --sync
+    run_dropcaches: bool = False   # --dropcache
+    run_statdirs: bool = False     # --statdirs (s3)
+
+    # --- workload shape ---
+    threads: int = 1               # -t
+    dirs: int = 0                  # -n (per thread)
+    files: int = 0                 # -N (per dir per thread)
+    file_size: int = 0             # -s
+    block_size: int = 1 << 20      # -b
+    iodepth: int = 1               # --iodepth
+    iterations: int = 1            # -i
+    inf_loop: bool = False         # --infloop
+
+    # --- access pattern ---
+    direct: bool = False           # --direct
+    random: bool = False           # --rand
+    rand_aligned: bool = True      # (--norandalign clears)
+    rand_amount: int = 0           # --randamount (total across workers; 0=dataset)
+    rand_algo: str = "balanced_single"  # --randalgo
+    strided: bool = False          # --strided
+    backward: bool = False         # --backward
+    truncate: bool = False         # --trunc
+    trunc_to_size: bool = False    # --trunctosize
+    prealloc: bool = False         # --preallocfile
+    sharesize: int = 0             # --sharesize (custom tree)
+
+    # --- integrity / variance ---
+    verify: int = -1               # --verify SALT (-1 off)
+    verify_direct: bool = False    # --verifydirect
+    read_inline: bool = False      # --readinline
+    blockvar_pct: int = 100        # --blockvarpct
+    blockvar_algo: str = "fast"    # --blockvaralgo
+
+    # --- GPU ---
+    gpu_ids: list[int] = field(default_factory=list)  # --gpuids
+    gpu_per_service: bool = False  # --gpuperservice
+    gpu_direct: bool = False       # --gds analogue: O_DIRECT + pinned staging
+    gpu_pinned: bool = True        # (--cuhostbufreg analogue; pinned host bufs)
+
+    # --- latency / stats / output ---
+    lat: bool = False              # --lat
+    lat_histo: bool = False        # --lathisto
+    lat_percent: bool = False      # --latpercent
+    lat_percent_9s: int = 0        # --latpercent9s
+    all_elapsed: bool = False      # --allelapsed
+    cpu_util: bool = False         # --cpu
+    live_int_ms: int = 2000        # --liveint
+    no_live: bool = False          # --nolive
+    live1: bool = False            # --live1
+    live_csv: str = ""             # --livecsv
+    live_csv_ex: bool = False      # --livecsvex
+    csv_file: str = ""             # --csvfile
+    json_file: str = ""            # --jsonfile
+    res_file: str = ""             # --resfile
+    no_csv_labels: bool = False    # --nocsvlabels
+    label: str = ""                # --label
+    show_dir_stats: bool = False   # --dirstats
+    log_level: int = 0             # --log
+    ops_log_path: str = ""         # --opslog
+    ops_log_lock: bool = False     # --opsloglock
+
+    # --- limits / timing ---
+    timelimit: int = 0             # --timelimit (secs per phase)
+    limit_read: int = 0            # --limitread (bytes/s per thread)
+    limit_write: int = 0           # --limitwrite
+    phase_delay_secs: int = 0      # --phasedelay
+    start_time: int = 0            # --start (epoch secs)
+
+    # --- error handling / checks ---
+    ignore_del_errors: bool = False  # --nodelerr
+    ignore_0usec_errors: bool = False  # --no0usecerr
+    no_dio_check: bool = False     # --nodiocheck
+    no_path_expansion: bool = False  # --nopathexp
+    dryrun: bool = False           # --dryrun
+
+    # --- placement ---
+    numa_zones: str = ""           # --zones
+    cpu_cores: str = ""            # --cores
+
+    # --- distributed ---
+    hosts: list[str] = field(default_factory=list)  # --hosts
+    service_mode: bool = False     # --service
+    service_port: int = 1611       # --port
+    foreground: bool = False       # --foreground
+    no_svc_share: bool = False     # --nosvcshare
+    num_hosts: int = -1            # --numhosts
+    rotate_hosts: int = 0          # --rotatehosts
+    svc_update_int_ms: int = 500   # --svcupint
+    svc_elapsed: bool = False      # --svcelapsed
+    svc_pw_file: str = ""          # --svcpwfile
+    interrupt_services: bool = False  # --interrupt
+    quit_services: bool = False    # --quit
+    rank_offset: int = 0           # --rankoffset
+    dir_sharing: bool = False      # --dirsharing
+    no_fd_sharing: bool = False    # --nofdsharing
+
+    # --- custom tree ---
+    treefile: str = ""             # --treefile
+    tree_round_up: int = 0         # --treeroundup
+    tree_scan: str = ""            # --treescan
+    tree_rand: bool = False        # --treerand
+
+    # --- rwmix ---
+    rwmix_pct: int = 0             # --rwmixpct
+    rwmix_threads: int = 0         # --rwmixthr
+    rwmix_thread_pct: int = 0      # --rwmixthrpct
+
+    # --- netbench ---
+    netbench: bool = False         # --netbench
+    servers: list[str] = field(default_factory=list)  # --servers
+    resp_size: int = 1             # --respsize
+    send_buf: int = 0              # --sendbuf
+    recv_buf: int = 0              # --recvbuf
+    netdevs: list[str] = field(default_factory=list)  # --netdevs
+
+    # --- S3 ---
+    s3_endpoints: list[str] = field(default_factory=list)  # --s3endpoints
+    s3_key: str = ""               # --s3key
+    s3_secret: str = ""            # --s3secret
+    s3_region: str = ""            # --s3region
+    s3_no_compress: bool = False
+    s3_fastget: bool = False
+    s3_fastput: bool = False
+    s3_list_obj: int = 0           # --s3listobj
+    s3_list_verify: bool = False   # --s3listverify
+    s3_multi_del: int = 0          # --s3multidel
+    s3_rand_obj: bool = False      # --s3randobj
+    s3_obj_prefix: str = ""        # --s3objprefix
+    s3_sign_policy: int = 0        # --s3sign
+    s3_max_conns: int = 0          # --s3maxconns
+    s3_ignore_errors: bool = False
+
+    # --- misc ---
+    config_file: str = ""          # -c/--configfile
+    bench_seed: int = 0            # internal; derived per run
+
+    # internal derived values (not user options)
+    num_dataset_threads: int = 0
+
+    # ------------------------------------------------------------------
+    def finalize(self) -> None:
+        """Expand paths, infer mode/path type, derive values, validate."""
+        if not self.no_path_expansion:
+            expanded: list[str] = []
+            for p in self.paths:
+                expanded.extend(expand_path_brackets(p))
+            self.paths = expanded
+        self.hosts = expand_hosts(self.hosts)
+
+        if self.paths and self.paths[0].startswith(S3_PATH_PREFIX):
+            self.bench_mode = "s3"
+        elif self.s3_endpoints:
+            self.bench_mode = "s3"
+        elif self.netbench:
+            self.bench_mode = "netbench"
+
+        if self.bench_mode == "posix" and self.paths:
+            self.path_type = self._infer_path_type()
+
+        # derived: total dataset threads across hosts sharing the dataset
+        num_hosts = len(self.hosts) if self.hosts else 1
+        if self.num_hosts >= 0:
+            num_hosts = min(num_hosts, self.num_hosts)
+        share = not self.no_svc_share or self.path_type == PATH_DIR
+        if self.hosts and not self.no_svc_share and self.path_type != PATH_DIR:
+            self.num_dataset_threads = self.threads * num_hosts
+        elif not self.hosts:
+            self.num_dataset_threads = self.threads
+        else:
+            self.num_dataset_threads = self.threads  # per-host private dataset
+        _ = share
+
+        self.validate()
+
+    def _infer_path_type(self) -> str:
+        types = set()
+        for p in self.paths:
+            try:
+                st = os.stat(p)
+            except FileNotFoundError:
+                types.add(PATH_FILE)  # will be created
+                continue
+            if statmod.S_ISDIR(st.st_mode):
+                types.add(PATH_DIR)
+            elif statmod.S_ISBLK(st.st_mode):
+                types.add(PATH_BDEV)
+            else:
+                types.add(PATH_FILE)
+        if len(types) > 1:
+            raise ConfigError(f"mixed path types are not supported: {sorted(types)}")
+        return types.pop() if types else PATH_FILE
+
+    def validate(self) -> None:
+        if self.service_mode:
+            return  # service gets its config from the master
+
+        needs_paths = any([self.run_mkdirs, self.run_write, self.run_read, self.run_stat,
+                           self.run_delfiles, self.run_deldirs])
+        if needs_paths and not self.paths and self.bench_mode != "netbench":
+            raise ConfigError("benchmark paths are required")
+
+        if self.threads < 1:
+            raise ConfigError("number of threads must be >= 1")
+        if self.block_size < 1 and (self.run_write or self.run_read):
+            raise ConfigError("block size must be >= 1")
+        if self.path_type == PATH_DIR and (self.run_write or self.run_read):
+            if not self.treefile and self.files < 1:
+                raise ConfigError("dir mode read/write requires -N/--files >= 1")
+            if self.file_size and self.block_size > self.file_size:
+                self.block_size = self.file_size
+        if self.path_type != PATH_DIR and (self.run_write or self.run_read):
+            if not self.file_size and self.path_type == PATH_FILE and self.run_write:
+                raise ConfigError("file mode write requires -s/--size")
+        if self.iodepth < 1:
+            raise ConfigError("iodepth must be >= 1")
+        if not (0 <= self.blockvar_pct <= 100):
+            raise ConfigError("blockvarpct must be in 0..100")
+        if self.rwmix_pct and not (0 <= self.rwmix_pct <= 100):
+            raise ConfigError("rwmixpct must be in 0..100")
+        if self.direct and not self.no_dio_check and (self.run_write or self.run_read):
+            if self.block_size % 512:
+                raise ConfigError("direct IO requires block size aligned to 512 bytes "
+                                  "(--nodiocheck to skip this check)")
+            if self.file_size and self.path_type != PATH_DIR and self.file_size % 512:
+                raise ConfigError("direct IO requires file size aligned to 512 bytes in "
+                                  "file/bdev mode (--nodiocheck to skip this check)")
+        if self.verify >= 0 and self.random and not self.rand_aligned:
+            raise ConfigError("--verify cannot be used with unaligned random offsets")
+        if self.verify >= 0 and self.blockvar_pct and False:
+            pass  # verify overrides block variance; no error
+
+    # ------------------------------------------------------------------
+    def phase_list(self) -> list[str]:
+        """Ordered phase names for one iteration (reference order,
+        Coordinator.cpp:311-334)."""
+        order = [
+            ("MKDIRS", self.run_mkdirs),
+            ("WRITE", self.run_write),
+            ("STAT", self.run_stat),
+            ("READ", self.run_read),
+            ("RMFILES", self.run_delfiles),
+            ("RMDIRS", self.run_deldirs),
+        ]
+        return [name for name, enabled in order if enabled]
+
+    # ------------------------------------------------------------------
+    def to_wire(self) -> dict[str, Any]:
+        """Serialize for master->service transfer (JSON-able)."""
+        d = dataclasses.asdict(self)
+        # master-only options never sent to services
+        for k in ("hosts", "service_mode", "quit_services", "interrupt_services",
+                  "csv_file", "json_file", "res_file", "live_csv", "config_file"):
+            d.pop(k, None)
+        return d
+
+    @classmethod
+    def from_wire(cls, d: dict[str, Any]) -> "BenchConfig":
+        cfg = cls()
+        for k, v in d.items():
+            if hasattr(cfg, k):
+                setattr(cfg, k, v)
+        return cfg
+
+    # ------------------------------------------------------------------
+    def engine_dict(self) -> dict[str, Any]:
+        """Config dict for the native engine (_core.Engine)."""
+        return dict(
+            paths=self.paths,
+            path_type=self.path_type,
+            threads=self.threads,
+            rank_offset=self.rank_offset,
+            num_dataset_threads=self.num_dataset_threads or self.threads,
+            dirs=self.dirs,
+            files=self.files,
+            file_size=self.file_size,
+            block_size=self.block_size,
+            iodepth=self.iodepth,
+            direct=self.direct,
+            random=self.random,
+            rand_aligned=self.rand_aligned,
+            rand_amount=self.rand_amount,
+            strided=self.strided,
+            backward=self.backward,
+            truncate=self.truncate,
+            trunc_to_size=self.file_size if self.trunc_to_size else None,
+            prealloc=self.prealloc,
+            verify_salt=self.verify,
+            verify_direct=self.verify_direct,
+            blockvar_pct=self.blockvar_pct,
+            blockvar_algo=self.blockvar_algo,
+            rand_algo=self.rand_algo,
+            gpu_ids=self.gpu_ids,
+            gpu_pinned=self.gpu_pinned,
+            lat=self.lat,
+            limit_read_bps=self.limit_read,
+            limit_write_bps=self.limit_write,
+            ignore_del_errors=self.ignore_del_errors,
+            dir_sharing=self.dir_sharing,
+            inf_loop=self.inf_loop,
+            bench_seed=self.bench_seed or 0x243F6A8885A308D3,
+        )
+
+
+def parse_gpu_ids(s: str) -> list[int]:
+    if not s:
+        return []
+    return [int(x) for x in s.replace(",", " ").split()]

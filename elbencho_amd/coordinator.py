@@ -1,0 +1,252 @@
+"""Run control: phase ordering, iterations, live stats loop, time limit.
+
+Reference analogue: /root/reference/source/Coordinator.cpp (phase order
+:311-334, sync/dropcaches interleave :278, SIGINT handling :420) and the
+WorkerManager phase barrier. Independent implementation: the local engine is
+the native _core.Engine; remote services are driven via elbencho_amd.remote;
+multi-GPU lockstep uses RCCL via elbencho_amd.parallel when torch.distributed
+is initialized.
+"""
+
+from __future__ import annotations
+
+import signal
+import sys
+import time
+import uuid
+from typing import Any, Optional
+
+from elbencho_amd import load_core
+from elbencho_amd.config import BenchConfig, PATH_DIR
+from elbencho_amd.stats import (CpuUtil, LiveCsvWriter, LiveStatsPrinter, PhaseResults,
+                                WorkerStats, aggregate_phase, append_csv_result,
+                                append_json_result, print_phase_results,
+                                print_results_table_header)
+
+PHASE_CODES: dict[str, int] = {}
+
+
+def phase_code(name: str) -> int:
+    global PHASE_CODES
+    if not PHASE_CODES:
+        PHASE_CODES = dict(load_core().PHASES)
+    return PHASE_CODES[name]
+
+
+class LocalRunner:
+    """Drives the native engine for one instance (standalone or service)."""
+
+    def __init__(self, cfg: BenchConfig):
+        self.cfg = cfg
+        self.core = load_core()
+        self.engine = self.core.Engine(cfg.engine_dict())
+        self.engine.prepare()
+
+    def start(self, phase_name: str) -> None:
+        self.engine.start_phase(phase_code(phase_name))
+
+    def poll(self) -> dict[str, Any]:
+        return self.engine.poll()
+
+    def wait(self, timeout_ms: int) -> bool:
+        return self.engine.wait_phase_done(timeout_ms)
+
+    def interrupt(self) -> None:
+        self.engine.interrupt()
+
+    def finish(self) -> list[WorkerStats]:
+        return [WorkerStats.from_engine(d) for d in self.engine.finish_phase()]
+
+    def planned_work(self, phase_name: str) -> tuple[int, int]:
+        return tuple(self.engine.planned_work(phase_code(phase_name)))
+
+
+class Coordinator:
+    def __init__(self, cfg: BenchConfig, out=None):
+        self.cfg = cfg
+        self.out = out or sys.stdout
+        self.interrupted = False
+        self._sigint_count = 0
+        self.runner: Any = None
+        self.dist = None  # parallel.PhaseSync when torch.distributed is up
+
+    # ------------------------------------------------------------------
+    def main(self) -> int:
+        cfg = self.cfg
+
+        if cfg.quit_services or cfg.interrupt_services:
+            from elbencho_amd.remote import send_control
+            send_control(cfg, quit=cfg.quit_services)
+            return 0
+
+        if cfg.service_mode:
+            from elbencho_amd.service import run_service
+            return run_service(cfg)
+
+        try:
+            signal.signal(signal.SIGINT, self._on_sigint)
+        except ValueError:
+            pass  # not in main thread (tests)
+
+        if cfg.hosts:
+            from elbencho_amd.remote import RemoteRunner
+            self.runner = RemoteRunner(cfg)
+        else:
+            import os
+            if "RANK" in os.environ and "WORLD_SIZE" in os.environ:
+                # launched under torch.distributed.run: one instance per GPU,
+                # RCCL barrier + all-reduce replace the HTTP master poll
+                from elbencho_amd import parallel
+                self.dist = parallel.init_from_env()
+                if self.dist:
+                    rank, world = self.dist.rank, self.dist.world_size
+                    if cfg.gpu_ids:
+                        local = int(os.environ.get("LOCAL_RANK", rank))
+                        cfg.gpu_ids = [cfg.gpu_ids[local % len(cfg.gpu_ids)]]
+                    cfg.rank_offset = rank * cfg.threads
+                    if not cfg.no_svc_share and cfg.path_type != PATH_DIR:
+                        cfg.num_dataset_threads = cfg.threads * world
+                    else:
+                        cfg.num_dataset_threads = cfg.threads
+            self.runner = LocalRunner(cfg)
+
+        try:
+            if cfg.dryrun:
+                self._print_dryrun()
+                return 0
+            return self.run_benchmarks()
+        finally:
+            closer = getattr(self.runner, "close", None)
+            if closer:
+                closer()
+
+    # ------------------------------------------------------------------
+    def _on_sigint(self, signum, frame):
+        self._sigint_count += 1
+        if self._sigint_count == 1:
+            print("\nReceived interrupt. Finishing current phase gracefully "
+                  "(interrupt again to abort)...", file=sys.stderr)
+            self.interrupted = True
+            if self.runner:
+                self.runner.interrupt()
+        else:
+            print("\nAborting.", file=sys.stderr)
+            sys.exit(130)
+
+    # ------------------------------------------------------------------
+    def _print_dryrun(self) -> None:
+        cfg = self.cfg
+        print("DRY RUN (no I/O will be done)", file=self.out)
+        for name in cfg.phase_list():
+            entries, nbytes = self.runner.planned_work(name)
+            print(f"  {name}: entries={entries} bytes={nbytes}", file=self.out)
+
+    # ------------------------------------------------------------------
+    def run_benchmarks(self) -> int:
+        cfg = self.cfg
+        phases = cfg.phase_list()
+        if not phases and not (cfg.run_sync or cfg.run_dropcaches):
+            print("No benchmark phase selected (e.g. -w to write or -r to read).",
+                  file=sys.stderr)
+            return 1
+
+        if cfg.start_time:
+            delay = cfg.start_time - time.time()
+            if delay > 0:
+                time.sleep(delay)
+
+        rc = 0
+        for it in range(cfg.iterations):
+            if cfg.iterations > 1:
+                print(f"[Starting iteration {it + 1} of {cfg.iterations}...]", file=self.out)
+            print_results_table_header(self.out)
+
+            self._run_sync_and_dropcaches()
+
+            for i, name in enumerate(phases):
+                if self.interrupted:
+                    rc = 130
+                    break
+                ok = self.run_phase(name)
+                if not ok:
+                    rc = 1
+                    break
+                self._run_sync_and_dropcaches()
+                if i < len(phases) - 1 and cfg.phase_delay_secs:
+                    time.sleep(cfg.phase_delay_secs)
+            if rc:
+                break
+        return rc
+
+    def _run_sync_and_dropcaches(self) -> None:
+        if self.cfg.run_sync:
+            self.run_phase("SYNC", quiet=False)
+        if self.cfg.run_dropcaches:
+            self.run_phase("DROPCACHES", quiet=False)
+
+    # ------------------------------------------------------------------
+    def run_phase(self, name: str, quiet: bool = False) -> bool:
+        cfg = self.cfg
+        phase_id = str(uuid.uuid4())
+        start_time = time.time()
+
+        planned_entries, planned_bytes = self.runner.planned_work(name)
+
+        cpu_first_meter = CpuUtil()  # phase start -> stonewall
+        cpu_last_meter = CpuUtil()   # phase start -> phase end
+        live = LiveStatsPrinter(cfg, name, planned_entries, planned_bytes)
+        live_csv = LiveCsvWriter(cfg.live_csv, cfg, name) if cfg.live_csv else None
+
+        if self.dist:
+            self.dist.barrier()  # lockstep phase start across GPU ranks
+
+        self.runner.start(name)
+
+        deadline = time.monotonic() + cfg.timelimit if cfg.timelimit else None
+        poll_int = max(0.05, cfg.live_int_ms / 1000.0)
+        cpu_first: Optional[int] = None
+
+        while not self.runner.wait(int(poll_int * 1000)):
+            p = self.runner.poll()
+            if cpu_first is None and p.get("stonewall_triggered"):
+                cpu_first = cpu_first_meter.percent_since_last()
+            live.update(p)
+            if live_csv:
+                live_csv.update(p)
+            if deadline and time.monotonic() > deadline:
+                print(f"\nPhase time limit reached ({cfg.timelimit}s); "
+                      "interrupting workers...", file=sys.stderr)
+                self.runner.interrupt()
+                deadline = None
+
+        live.finish()
+        if live_csv:
+            live_csv.close()
+
+        cpu_last = cpu_last_meter.percent_since_last()
+        if cpu_first is None:
+            cpu_first = cpu_last
+        workers = self.runner.finish()
+
+        results = aggregate_phase(name, phase_id, start_time, workers,
+                                  cpu_first, cpu_last)
+
+        if self.dist:
+            results = self.dist.allreduce_results(results)
+
+        if not quiet or results.errors:
+            if self.dist is None or self.dist.rank == 0:
+                print_phase_results(cfg, results, self.out)
+
+        if cfg.csv_file and (self.dist is None or self.dist.rank == 0):
+            append_csv_result(cfg, results, cfg.csv_file)
+        if cfg.json_file and (self.dist is None or self.dist.rank == 0):
+            append_json_result(cfg, results, cfg.json_file)
+        if cfg.res_file and (self.dist is None or self.dist.rank == 0):
+            with open(cfg.res_file, "a") as f:
+                print_phase_results(cfg, results, f)
+
+        interrupted_only = all(e.endswith("interrupted") for e in results.errors)
+        if results.errors and not (self.interrupted and interrupted_only):
+            return False
+        return True

@@ -125,6 +125,13 @@ Worker::~Worker()
 {
     if (ownHostBufs)
         for (auto p : hostBufs) free(p);
+
+    if (gpu) { // return the GPU context for reuse by the next phase
+        std::lock_guard<std::mutex> lk(eng.gpuCacheMtx);
+        if ((size_t)localRank >= eng.gpuCtxCache.size())
+            eng.gpuCtxCache.resize(localRank + 1);
+        eng.gpuCtxCache[localRank] = std::move(gpu);
+    }
 }
 
 void Worker::checkInterrupt()
@@ -172,7 +179,19 @@ void Worker::setupGpu()
         throw WorkerError("GPU requested (gpuids) but no HIP device is available — "
                           "refusing silent CPU fallback");
 
-    gpu = std::make_unique<GpuCtx>(devId, slots, cfg.blockSize, cfg.gpuPinnedHostBufs);
+    { // reuse a cached context from a previous phase when shapes match
+        std::lock_guard<std::mutex> lk(eng.gpuCacheMtx);
+        if ((size_t)localRank < eng.gpuCtxCache.size() && eng.gpuCtxCache[localRank] &&
+            eng.gpuCtxCache[localRank]->deviceId() == devId &&
+            eng.gpuCtxCache[localRank]->numSlots() >= slots &&
+            eng.gpuCtxCache[localRank]->bufSize() >= cfg.blockSize)
+            gpu = std::move(eng.gpuCtxCache[localRank]);
+    }
+
+    if (gpu)
+        gpu->bindThread();
+    else
+        gpu = std::make_unique<GpuCtx>(devId, slots, cfg.blockSize, cfg.gpuPinnedHostBufs);
 }
 
 void Worker::fairShareSlice(uint64_t totalLen, uint64_t& myStart, uint64_t& myLen) const

@@ -160,7 +160,6 @@ private:
     std::unique_ptr<RandAlgo> rng;        // offsets
     std::unique_ptr<RandAlgo> fillRng;    // block variance fill
     RateLimiter rateLimiter;
-    uint64_t interruptCheckCounter = 0;
 };
 
 // Shared per-phase state + the engine facade exposed to Python.
@@ -213,6 +212,11 @@ public:
     // resolved at prepare()
     std::vector<uint64_t> resolvedFileSizes; // per path (file/bdev mode)
     uint64_t effFileSize = 0;                // uniform stripe unit
+
+    // GPU contexts cached across phases (hipMalloc + pinned allocs are
+    // expensive; reusing them keeps phase setup off the measured path)
+    std::mutex gpuCacheMtx;
+    std::vector<std::unique_ptr<GpuCtx>> gpuCtxCache;
 
     void onWorkerDone(Worker& w, bool hadError);
 

@@ -45,6 +45,11 @@ public:
     int deviceId() const { return devId; }
     char* hostBuf(int slot) const;
     uint64_t bufSize() const { return slotSize; }
+    int numSlots() const { return slots; }
+
+    // bind the calling thread to this context's device (hipSetDevice);
+    // required when a cached context is reused by a new worker thread
+    void bindThread();
 
     // --- async staging copies on this worker's stream ---
     void copyH2DAsync(int slot, uint64_t len);
@@ -73,6 +78,7 @@ private:
     Impl* impl;
     int devId;
     uint64_t slotSize;
+    int slots;
 };
 
 } // namespace eb

@@ -235,7 +235,7 @@ struct GpuCtx::Impl {
 };
 
 GpuCtx::GpuCtx(int deviceId, int numSlots, uint64_t bufSize, bool pinnedHostBufs)
-    : impl(new Impl), devId(deviceId), slotSize(bufSize)
+    : impl(new Impl), devId(deviceId), slotSize(bufSize), slots(numSlots)
 {
     HIP_CHECK(hipSetDevice(deviceId));
     HIP_CHECK(hipStreamCreateWithFlags(&impl->stream, hipStreamNonBlocking));
@@ -278,6 +278,8 @@ GpuCtx::~GpuCtx()
 }
 
 char* GpuCtx::hostBuf(int slot) const { return impl->hostBufs[slot]; }
+
+void GpuCtx::bindThread() { HIP_CHECK(hipSetDevice(devId)); }
 
 void GpuCtx::copyH2DAsync(int slot, uint64_t len)
 {

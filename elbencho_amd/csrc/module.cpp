@@ -114,6 +114,77 @@ PYBIND11_MODULE(_core, m)
         return verifyChecksumCPU(buf.data(), buf.size(), fileOff, salt);
     });
 
+    // --- offset generator test hook ---
+    m.def("gen_offsets",
+          [](const std::string& kind, uint64_t blockSize, uint64_t rangeStart,
+             uint64_t rangeLen, uint64_t seed, uint64_t rank, uint64_t nranks,
+             uint64_t amount, uint64_t maxCount) {
+              RandAlgoXoshiro256ss rng(seed);
+              std::unique_ptr<OffsetGen> gen;
+              if (kind == "seq")
+                  gen = std::make_unique<OffsetGenSequential>(blockSize);
+              else if (kind == "reverse")
+                  gen = std::make_unique<OffsetGenReverseSeq>(blockSize);
+              else if (kind == "random")
+                  gen = std::make_unique<OffsetGenRandom>(blockSize, rng, amount);
+              else if (kind == "random_aligned")
+                  gen = std::make_unique<OffsetGenRandomAligned>(blockSize, rng, amount);
+              else if (kind == "full_coverage")
+                  gen = std::make_unique<OffsetGenRandomAlignedFullCoverage>(blockSize, seed);
+              else if (kind == "strided")
+                  gen = std::make_unique<OffsetGenStrided>(blockSize, rank, nranks);
+              else
+                  throw std::runtime_error("unknown offset generator: " + kind);
+              gen->reset(rangeStart, rangeLen);
+              std::vector<std::pair<uint64_t, uint64_t>> out;
+              BlockSpec spec;
+              while (gen->next(spec) && out.size() < maxCount)
+                  out.emplace_back(spec.offset, spec.len);
+              return py::make_tuple(out, gen->totalBytes());
+          },
+          py::arg("kind"), py::arg("block_size"), py::arg("range_start"),
+          py::arg("range_len"), py::arg("seed") = 1, py::arg("rank") = 0,
+          py::arg("nranks") = 1, py::arg("amount") = 0,
+          py::arg("max_count") = 1u << 22);
+
+    // --- GPU kernel test helpers (numerics parity vs the CPU reference) ---
+    m.def("gpu_fill_checksum", [](uint64_t len, uint64_t fileOff, uint64_t salt, int dev) {
+        GpuCtx ctx(dev, 1, len, true);
+        ctx.fillChecksumDev(0, len, fileOff, salt);
+        ctx.copyD2HAsync(0, len);
+        ctx.syncStream();
+        return py::bytes(ctx.hostBuf(0), len);
+    }, py::arg("len"), py::arg("file_off"), py::arg("salt"), py::arg("dev") = 0);
+
+    m.def("gpu_verify_checksum", [](py::bytes data, uint64_t fileOff, uint64_t salt, int dev) {
+        std::string buf = data;
+        GpuCtx ctx(dev, 1, buf.size(), true);
+        std::memcpy(ctx.hostBuf(0), buf.data(), buf.size());
+        ctx.copyH2DAsync(0, buf.size());
+        ctx.syncStream();
+        GpuVerifyResult r = ctx.verifyChecksumDev(0, buf.size(), fileOff, salt);
+        return py::make_tuple(r.numMismatches, r.firstBadFileOffset);
+    }, py::arg("data"), py::arg("file_off"), py::arg("salt"), py::arg("dev") = 0);
+
+    m.def("gpu_fill_rand", [](uint64_t len, uint64_t seed, int dev) {
+        GpuCtx ctx(dev, 1, len, true);
+        ctx.fillRandDev(0, len, seed);
+        ctx.copyD2HAsync(0, len);
+        ctx.syncStream();
+        return py::bytes(ctx.hostBuf(0), len);
+    }, py::arg("len"), py::arg("seed"), py::arg("dev") = 0);
+
+    m.def("gpu_blockvar_refill", [](uint64_t len, uint64_t refillLen, uint64_t seed, int dev) {
+        GpuCtx ctx(dev, 1, len, true);
+        std::memset(ctx.hostBuf(0), 0, len);
+        ctx.copyH2DAsync(0, len);
+        ctx.syncStream();
+        ctx.blockVarRefillDev(0, len, refillLen, seed);
+        ctx.copyD2HAsync(0, len);
+        ctx.syncStream();
+        return py::bytes(ctx.hostBuf(0), len);
+    }, py::arg("len"), py::arg("refill_len"), py::arg("seed"), py::arg("dev") = 0);
+
     py::class_<Engine>(m, "Engine")
         .def(py::init([](const py::dict& cfg) {
             return std::make_unique<Engine>(configFromDict(cfg));

@@ -1,0 +1,136 @@
+"""RCCL-over-xGMI phase synchronization for multi-GPU lockstep runs.
+
+This replaces the reference's master-polls-services statistics loop for
+intra-node GPU workers: a torch.distributed barrier
+gives lockstep phase starts across the node's 8 MI355X GPUs, and the
+LiveOps counters + latency histogram buckets are aggregated with a single
+all-reduce per phase end. The payloads are tiny (a few KB), so latency —
+not xGMI link bandwidth — dominates; we batch everything into one SUM
+all-reduce plus one MIN/MAX pair.
+
+Backend: "nccl" (= RCCL on ROCm) with device tensors on GPU ranks, "gloo"
+with CPU tensors in CPU tests. Collectives run only at phase boundaries,
+never inside the measured I/O loop (SURVEY.md §7 hard part (f)).
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from elbencho_amd.stats import PhaseResults
+
+
+def init_from_env(device: Optional[torch.device] = None) -> "PhaseSync | None":
+    """Initialize torch.distributed from torchrun env vars; None if absent."""
+    if "RANK" not in os.environ or "WORLD_SIZE" not in os.environ:
+        return None
+    if not dist.is_initialized():
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29511")
+        dist.init_process_group(
+            backend=backend,
+            timeout=datetime.timedelta(seconds=int(os.environ.get("EB_DIST_TIMEOUT", "600"))),
+        )
+    if device is None and torch.cuda.is_available():
+        local_rank = int(os.environ.get("LOCAL_RANK", dist.get_rank()))
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    return PhaseSync(device)
+
+
+class PhaseSync:
+    """Barrier + stats aggregation over an initialized process group."""
+
+    def __init__(self, device: Optional[torch.device] = None):
+        if not dist.is_initialized():
+            raise RuntimeError("torch.distributed is not initialized")
+        self.rank = dist.get_rank()
+        self.world_size = dist.get_world_size()
+        if device is None:
+            device = (torch.device("cuda", torch.cuda.current_device())
+                      if dist.get_backend() == "nccl" else torch.device("cpu"))
+        self.device = device
+
+    # ------------------------------------------------------------------
+    def barrier(self) -> None:
+        """Lockstep phase start across ranks (replaces /startphase round-trip)."""
+        if dist.get_backend() == "nccl":
+            dist.barrier(device_ids=[self.device.index])
+        else:
+            dist.barrier()
+
+    # ------------------------------------------------------------------
+    def allreduce_results(self, r: PhaseResults) -> PhaseResults:
+        """Aggregate a per-rank PhaseResults into the whole-job result.
+
+        One SUM all-reduce carries counters + both histograms' buckets; one
+        MIN and one MAX carry the time/min/max fields — 3 small collectives
+        per phase end, off the measured I/O path.
+        """
+        hist_len = len(r.io_lat.vec)
+
+        # --- SUM: counters + histogram counts/sums + bucket arrays ---
+        sum_vals = [
+            r.entries, r.bytes, r.iops,
+            r.sw_entries, r.sw_bytes, r.sw_iops,
+            r.io_lat.vec[0], r.io_lat.vec[1],
+            r.entry_lat.vec[0], r.entry_lat.vec[1],
+        ]
+        sum_vals += r.io_lat.vec[4:]
+        sum_vals += r.entry_lat.vec[4:]
+        t_sum = torch.tensor(sum_vals, dtype=torch.float64, device=self.device)
+        dist.all_reduce(t_sum, op=dist.ReduceOp.SUM)
+
+        # --- MIN: first-finish elapsed + histogram mins ---
+        t_min = torch.tensor(
+            [r.first_finish_usec if r.first_finish_usec else 2**53,
+             r.io_lat.vec[2], r.entry_lat.vec[2]],
+            dtype=torch.float64, device=self.device)
+        dist.all_reduce(t_min, op=dist.ReduceOp.MIN)
+
+        # --- MAX: last-finish elapsed + histogram maxes ---
+        t_max = torch.tensor(
+            [r.last_finish_usec, r.io_lat.vec[3], r.entry_lat.vec[3]],
+            dtype=torch.float64, device=self.device)
+        dist.all_reduce(t_max, op=dist.ReduceOp.MAX)
+
+        s = [int(x) for x in t_sum.tolist()]
+        mn = [int(x) for x in t_min.tolist()]
+        mx = [int(x) for x in t_max.tolist()]
+
+        out = PhaseResults(phase_name=r.phase_name, phase_id=r.phase_id,
+                           start_time=r.start_time)
+        out.entries, out.bytes, out.iops = s[0], s[1], s[2]
+        out.sw_entries, out.sw_bytes, out.sw_iops = s[3], s[4], s[5]
+        out.first_finish_usec = mn[0] if mn[0] < 2**53 else 0
+        out.last_finish_usec = mx[0]
+        out.cpu_first, out.cpu_last = r.cpu_first, r.cpu_last
+
+        nb = hist_len - 4
+        out.io_lat.vec = [s[6], s[7], mn[1], mx[1]] + s[10:10 + nb]
+        out.entry_lat.vec = [s[8], s[9], mn[2], mx[2]] + s[10 + nb:10 + 2 * nb]
+        out.errors = list(r.errors)
+        out.worker_elapsed_usec = list(r.worker_elapsed_usec)
+        return out
+
+    # ------------------------------------------------------------------
+    def allreduce_sum(self, values: list[float]) -> list[float]:
+        t = torch.tensor(values, dtype=torch.float64, device=self.device)
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        return t.tolist()
+
+    def allreduce_max(self, values: list[float]) -> list[float]:
+        t = torch.tensor(values, dtype=torch.float64, device=self.device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        return t.tolist()
+
+    def broadcast_obj(self, obj, src: int = 0):
+        lst = [obj]
+        dist.broadcast_object_list(lst, src=src)
+        return lst[0]

@@ -1,0 +1,278 @@
+"""Master-side remote service driver.
+
+Reference analogue: /root/reference/source/workers/RemoteWorker.cpp — one
+thread per service host, POST /preparephase with the serialized config,
+GET /startphase, adaptive /status poll loop (:447-584, 25ms -> svcupint),
+GET /benchresult collection, stonewall propagation. Independent
+implementation on http.client.
+"""
+
+from __future__ import annotations
+
+import hashlib
+import http.client
+import json
+import threading
+import time
+import urllib.parse
+from dataclasses import dataclass, field
+from typing import Any, Optional
+
+from elbencho_amd import HTTP_PROTOCOL_VERSION
+from elbencho_amd.config import BenchConfig
+from elbencho_amd.stats import WorkerStats
+
+DEFAULT_PORT = 1611
+FIRST_POLL_MS = 25  # adaptive poll start (reference RemoteWorker.cpp:699)
+
+
+def split_host(h: str) -> tuple[str, int]:
+    if ":" in h:
+        host, port = h.rsplit(":", 1)
+        return host, int(port)
+    return h, DEFAULT_PORT
+
+
+class ServiceClient:
+    """HTTP client for one service host."""
+
+    def __init__(self, hostport: str, auth: str = "", timeout: float = 30.0):
+        self.hostport = hostport
+        self.host, self.port = split_host(hostport)
+        self.auth = auth
+        self.timeout = timeout
+
+    def _headers(self) -> dict[str, str]:
+        h = {}
+        if self.auth:
+            h["X-Service-Auth"] = self.auth
+        return h
+
+    def get(self, path: str, timeout: float | None = None) -> Any:
+        conn = http.client.HTTPConnection(self.host, self.port,
+                                          timeout=timeout or self.timeout)
+        try:
+            conn.request("GET", path, headers=self._headers())
+            resp = conn.getresponse()
+            body = resp.read()
+            if resp.status != 200:
+                raise RuntimeError(self._frame_err(path, resp.status, body))
+            ctype = resp.getheader("Content-Type", "")
+            return json.loads(body) if "json" in ctype else body.decode()
+        finally:
+            conn.close()
+
+    def post(self, path: str, obj: Any) -> Any:
+        conn = http.client.HTTPConnection(self.host, self.port, timeout=self.timeout)
+        try:
+            body = json.dumps(obj).encode()
+            hdrs = self._headers()
+            hdrs["Content-Type"] = "application/json"
+            conn.request("POST", path, body=body, headers=hdrs)
+            resp = conn.getresponse()
+            rbody = resp.read()
+            if resp.status != 200:
+                raise RuntimeError(self._frame_err(path, resp.status, rbody))
+            return json.loads(rbody)
+        finally:
+            conn.close()
+
+    def _frame_err(self, path: str, status: int, body: bytes) -> str:
+        try:
+            msg = json.loads(body).get("error", body.decode())
+        except (ValueError, UnicodeDecodeError):
+            msg = body.decode(errors="replace")
+        return f"Service {self.hostport}{path} returned HTTP {status}: {msg}"
+
+
+@dataclass
+class HostState:
+    client: ServiceClient
+    last_status: dict = field(default_factory=dict)
+    done: bool = False
+    error: str = ""
+
+
+class RemoteRunner:
+    """Coordinator backend driving N remote services (master mode)."""
+
+    def __init__(self, cfg: BenchConfig):
+        self.cfg = cfg
+        auth = ""
+        if cfg.svc_pw_file:
+            with open(cfg.svc_pw_file) as f:
+                auth = hashlib.sha256(f.read().strip().encode()).hexdigest()
+
+        hosts = cfg.hosts
+        if cfg.num_hosts >= 0:
+            hosts = hosts[: cfg.num_hosts]
+        self.hosts = [HostState(ServiceClient(h, auth)) for h in hosts]
+        self._phase = ""
+        self._bench_id = ""
+        self._poll_ms = FIRST_POLL_MS
+        self._wait_for_services()
+
+    # ------------------------------------------------------------------
+    def _wait_for_services(self, timeout: float = 10.0) -> None:
+        deadline = time.monotonic() + timeout
+        for hs in self.hosts:
+            while True:
+                try:
+                    ver = hs.client.get("/protocolversion", timeout=2.0)
+                    if ver != HTTP_PROTOCOL_VERSION:
+                        raise RuntimeError(
+                            f"Service {hs.client.hostport} protocol version {ver!r} does "
+                            f"not match master {HTTP_PROTOCOL_VERSION!r}")
+                    break
+                except (ConnectionError, OSError) as e:
+                    if time.monotonic() > deadline:
+                        raise RuntimeError(
+                            f"Service {hs.client.hostport} is not reachable: {e}") from e
+                    time.sleep(0.25)
+
+    # ------------------------------------------------------------------
+    def _for_all(self, fn) -> list[Any]:
+        """Run fn(hostIdx, HostState) on all hosts in parallel; raise first error."""
+        results: list[Any] = [None] * len(self.hosts)
+        errors: list[str] = []
+
+        def run(i: int, hs: HostState):
+            try:
+                results[i] = fn(i, hs)
+            except Exception as e:  # noqa: BLE001
+                errors.append(str(e))
+
+        threads = [threading.Thread(target=run, args=(i, hs))
+                   for i, hs in enumerate(self.hosts)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        if errors:
+            raise RuntimeError("; ".join(errors))
+        return results
+
+    # ------------------------------------------------------------------
+    def start(self, phase_name: str) -> None:
+        import uuid
+
+        cfg = self.cfg
+        self._phase = phase_name
+        self._bench_id = str(uuid.uuid4())
+        self._poll_ms = FIRST_POLL_MS
+        shared = not cfg.no_svc_share and cfg.path_type != "dir"
+        num_hosts = len(self.hosts)
+
+        def prep(i: int, hs: HostState):
+            wire = cfg.to_wire()
+            wire["threads"] = cfg.threads
+            wire["rank_offset"] = i * cfg.threads
+            wire["num_dataset_threads"] = (cfg.threads * num_hosts) if shared else cfg.threads
+            hs.done = False
+            hs.error = ""
+            hs.last_status = {}
+            return hs.client.post("/preparephase",
+                                  {"protocol_version": HTTP_PROTOCOL_VERSION, "config": wire})
+
+        self._for_all(prep)
+        self._for_all(lambda i, hs: hs.client.get(
+            f"/startphase?phase={urllib.parse.quote(phase_name)}&benchid={self._bench_id}"))
+
+    # ------------------------------------------------------------------
+    def wait(self, timeout_ms: int) -> bool:
+        """Poll all services once (after an adaptive sleep); True when all done."""
+        time.sleep(min(self._poll_ms, timeout_ms if timeout_ms > 0 else self._poll_ms) / 1000.0)
+        self._poll_ms = min(self._poll_ms * 2, max(self.cfg.svc_update_int_ms, 50))
+
+        def poll(i: int, hs: HostState):
+            if hs.done:
+                return
+            st = hs.client.get("/status")
+            if st.get("bench_id") and st["bench_id"] != self._bench_id:
+                raise RuntimeError(
+                    f"Service {hs.client.hostport} reports foreign benchmark ID "
+                    f"(service hijacked?): {st['bench_id']} != {self._bench_id}")
+            hs.last_status = st
+            if st.get("workers_total") and st["workers_done"] >= st["workers_total"]:
+                hs.done = True
+
+        self._for_all(poll)
+        return all(hs.done for hs in self.hosts)
+
+    # ------------------------------------------------------------------
+    def poll(self) -> dict[str, Any]:
+        agg = {"entries": 0, "bytes": 0, "iops": 0, "workers_done": 0, "workers_total": 0,
+               "workers_with_error": 0, "elapsed_usec": 0, "stonewall_triggered": False,
+               "lat_num_ios": 0, "lat_sum_ios": 0, "lat_num_entries": 0,
+               "lat_sum_entries": 0}
+        for hs in self.hosts:
+            st = hs.last_status
+            if not st or st.get("idle"):
+                continue
+            for k in ("entries", "bytes", "iops", "workers_done", "workers_total",
+                      "workers_with_error", "lat_num_ios", "lat_sum_ios",
+                      "lat_num_entries", "lat_sum_entries"):
+                agg[k] += st.get(k, 0)
+            agg["elapsed_usec"] = max(agg["elapsed_usec"], st.get("elapsed_usec", 0))
+            agg["stonewall_triggered"] |= bool(st.get("stonewall_triggered"))
+        return agg
+
+    # ------------------------------------------------------------------
+    def interrupt(self) -> None:
+        try:
+            self._for_all(lambda i, hs: hs.client.get("/interruptphase"))
+        except RuntimeError:
+            pass
+
+    # ------------------------------------------------------------------
+    def finish(self) -> list[WorkerStats]:
+        def collect(i: int, hs: HostState):
+            return hs.client.get("/benchresult")
+
+        all_workers: list[WorkerStats] = []
+        for res in self._for_all(collect):
+            for w in res["workers"]:
+                ws = WorkerStats(**{k: w[k] for k in w if k in WorkerStats.__dataclass_fields__})
+                all_workers.append(ws)
+        return all_workers
+
+    # ------------------------------------------------------------------
+    def planned_work(self, phase_name: str) -> tuple[int, int]:
+        cfg = self.cfg
+        if cfg.path_type == "dir":
+            n_hosts = len(self.hosts)
+            dirs = cfg.dirs or 1
+            if phase_name in ("MKDIRS", "RMDIRS"):
+                return cfg.dirs * cfg.threads * n_hosts, 0
+            if phase_name in ("WRITE", "READ"):
+                e = dirs * cfg.files * cfg.threads * n_hosts
+                return e, e * cfg.file_size
+            if phase_name in ("STAT", "RMFILES"):
+                return dirs * cfg.files * cfg.threads * n_hosts, 0
+            return 0, 0
+        if phase_name in ("WRITE", "READ"):
+            total = cfg.file_size * len(cfg.paths)
+            if cfg.no_svc_share:
+                total *= len(self.hosts)
+            if cfg.random and cfg.rand_amount:
+                total = cfg.rand_amount
+            return 0, total
+        return 0, 0
+
+    def close(self) -> None:
+        pass
+
+
+def send_control(cfg: BenchConfig, quit: bool = False) -> None:
+    """--interrupt / --quit: stop the phase (and optionally the services)."""
+    auth = ""
+    if cfg.svc_pw_file:
+        with open(cfg.svc_pw_file) as f:
+            auth = hashlib.sha256(f.read().strip().encode()).hexdigest()
+    for h in cfg.hosts:
+        client = ServiceClient(h, auth)
+        try:
+            client.get(f"/interruptphase?quit={'1' if quit else '0'}")
+            print(f"Service {h}: {'quit' if quit else 'interrupt'} requested.")
+        except (ConnectionError, OSError, RuntimeError) as e:
+            print(f"Service {h}: {e}")

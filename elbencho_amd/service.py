@@ -1,0 +1,256 @@
+"""HTTP service mode — the per-host agent of distributed runs.
+
+Reference analogue: /root/reference/source/HTTPServiceSWS.cpp (endpoints
+/info /protocolversion /status /benchresult /preparephase /startphase
+/interruptphase, Common.h:229-246) and HTTPService.cpp (daemonize, port
+check). Independent implementation on http.server; wire format is JSON.
+
+The MI355X-native twist: intra-node GPU workers are synchronized via RCCL
+barriers (elbencho_amd.parallel) when the service is launched per-GPU; the
+HTTP plane is kept for multi-node CLI compatibility.
+"""
+
+from __future__ import annotations
+
+import hashlib
+import json
+import os
+import socket
+import sys
+import threading
+import time
+import urllib.parse
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+from elbencho_amd import HTTP_PROTOCOL_VERSION, VERSION
+from elbencho_amd.config import BenchConfig
+from elbencho_amd.coordinator import LocalRunner
+from elbencho_amd.stats import CpuUtil
+
+
+def _auth_hash(pw: str) -> str:
+    return hashlib.sha256(pw.encode()).hexdigest()
+
+
+class ServiceState:
+    def __init__(self, base_cfg: BenchConfig):
+        self.base_cfg = base_cfg  # service-side overrides (paths, GPUs, port)
+        self.lock = threading.Lock()
+        self.runner: LocalRunner | None = None
+        self.cfg: BenchConfig | None = None
+        self.bench_id = ""
+        self.phase_name = "IDLE"
+        self.results: list[dict] | None = None
+        self.error: str = ""
+        self.cpu = CpuUtil()
+        self.quit_requested = threading.Event()
+        self.auth = _auth_hash(self._read_pw(base_cfg.svc_pw_file)) if base_cfg.svc_pw_file \
+            else ""
+
+    @staticmethod
+    def _read_pw(path: str) -> str:
+        with open(path) as f:
+            return f.read().strip()
+
+    # ------------------------------------------------------------------
+    def prepare_phase(self, wire_cfg: dict) -> dict:
+        with self.lock:
+            cfg = BenchConfig.from_wire(wire_cfg)
+            # service-side overrides (reference HTTPService.cpp:141-161)
+            if self.base_cfg.paths:
+                cfg.paths = list(self.base_cfg.paths)
+            if self.base_cfg.gpu_ids:
+                cfg.gpu_ids = list(self.base_cfg.gpu_ids)
+            cfg.service_mode = False
+            cfg.hosts = []
+            self.cfg = cfg
+            self.runner = LocalRunner(cfg)
+            self.results = None
+            self.error = ""
+            self.phase_name = "IDLE"
+            return {"protocol_version": HTTP_PROTOCOL_VERSION,
+                    "path_type": cfg.path_type,
+                    "num_threads": cfg.threads}
+
+    def start_phase(self, phase_name: str, bench_id: str) -> None:
+        with self.lock:
+            if not self.runner:
+                raise RuntimeError("no prepared phase (call /preparephase first)")
+            self.bench_id = bench_id
+            self.phase_name = phase_name
+            self.results = None
+            self.cpu = CpuUtil()
+            self.runner.start(phase_name)
+
+    def status(self) -> dict:
+        with self.lock:
+            if not self.runner:
+                return {"bench_id": self.bench_id, "phase_name": self.phase_name,
+                        "workers_done": 0, "workers_total": 0, "idle": True}
+            p = self.runner.poll()
+            p["bench_id"] = self.bench_id
+            p["phase_name"] = self.phase_name
+            p["cpu_util_pct"] = 0
+            p["idle"] = False
+            return p
+
+    def bench_result(self) -> dict:
+        with self.lock:
+            if not self.runner:
+                raise RuntimeError("no active benchmark")
+            if self.results is None:
+                self.runner.wait(-1)
+                workers = self.runner.finish()
+                self.results = [w.__dict__ for w in workers]
+                self.phase_name = "IDLE"
+            return {"bench_id": self.bench_id, "workers": self.results}
+
+    def interrupt(self) -> None:
+        with self.lock:
+            if self.runner:
+                self.runner.interrupt()
+
+
+class Handler(BaseHTTPRequestHandler):
+    state: ServiceState  # set by run_service
+    protocol_version = "HTTP/1.1"
+
+    # --- helpers ---
+    def _send(self, code: int, body: bytes, ctype: str = "application/json") -> None:
+        self.send_response(code)
+        self.send_header("Content-Type", ctype)
+        self.send_header("Content-Length", str(len(body)))
+        self.end_headers()
+        self.wfile.write(body)
+
+    def _send_json(self, obj, code: int = 200) -> None:
+        self._send(code, json.dumps(obj).encode())
+
+    def _send_err(self, msg: str, code: int = 500) -> None:
+        self._send_json({"error": msg}, code)
+
+    def _check_auth(self) -> bool:
+        if not self.state.auth:
+            return True
+        if self.headers.get("X-Service-Auth", "") == self.state.auth:
+            return True
+        self._send_err("authorization failed (service password mismatch)", 403)
+        return False
+
+    def log_message(self, fmt, *args):  # quiet
+        pass
+
+    # --- GET endpoints ---
+    def do_GET(self):
+        if not self._check_auth():
+            return
+        url = urllib.parse.urlparse(self.path)
+        q = urllib.parse.parse_qs(url.query)
+        try:
+            if url.path == "/info":
+                self._send(200, f"elbencho-amd service v{VERSION} on "
+                                f"{socket.gethostname()}\n".encode(), "text/plain")
+            elif url.path == "/protocolversion":
+                self._send(200, HTTP_PROTOCOL_VERSION.encode(), "text/plain")
+            elif url.path == "/status":
+                self._send_json(self.state.status())
+            elif url.path == "/benchresult":
+                self._send_json(self.state.bench_result())
+            elif url.path == "/startphase":
+                phase = q.get("phase", ["IDLE"])[0]
+                bench_id = q.get("benchid", [""])[0]
+                self.state.start_phase(phase, bench_id)
+                self._send_json({"ok": True})
+            elif url.path == "/interruptphase":
+                self.state.interrupt()
+                if q.get("quit", ["0"])[0] == "1":
+                    self.state.quit_requested.set()
+                self._send_json({"ok": True})
+            else:
+                self._send_err(f"unknown endpoint: {url.path}", 404)
+        except Exception as e:  # noqa: BLE001 — report to master
+            self._send_err(str(e))
+
+    # --- POST endpoints ---
+    def do_POST(self):
+        if not self._check_auth():
+            return
+        url = urllib.parse.urlparse(self.path)
+        length = int(self.headers.get("Content-Length", "0"))
+        body = self.rfile.read(length)
+        try:
+            if url.path == "/preparephase":
+                req = json.loads(body)
+                ver = req.get("protocol_version", "")
+                if ver != HTTP_PROTOCOL_VERSION:
+                    self._send_err(
+                        f"protocol version mismatch: master={ver} "
+                        f"service={HTTP_PROTOCOL_VERSION}", 400)
+                    return
+                info = self.state.prepare_phase(req["config"])
+                self._send_json(info)
+            elif url.path == "/preparefile":
+                name = urllib.parse.parse_qs(url.query).get("name", ["upload"])[0]
+                dest = os.path.join("/tmp", f"elbencho_amd_svc_{os.getpid()}_{name}")
+                with open(dest, "wb") as f:
+                    f.write(body)
+                self._send_json({"ok": True, "path": dest})
+            else:
+                self._send_err(f"unknown endpoint: {url.path}", 404)
+        except Exception as e:  # noqa: BLE001
+            self._send_err(str(e))
+
+
+def check_port_available(port: int) -> None:
+    s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    s.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+    try:
+        s.bind(("", port))
+    except OSError as e:
+        raise RuntimeError(f"service port {port} is not available: {e}") from e
+    finally:
+        s.close()
+
+
+def daemonize(logfile: str) -> None:
+    """Classic double fork; stdout/stderr to logfile."""
+    if os.fork() > 0:
+        os._exit(0)
+    os.setsid()
+    if os.fork() > 0:
+        os._exit(0)
+    sys.stdout.flush()
+    sys.stderr.flush()
+    log = open(logfile, "a")
+    devnull = open(os.devnull)
+    os.dup2(devnull.fileno(), 0)
+    os.dup2(log.fileno(), 1)
+    os.dup2(log.fileno(), 2)
+
+
+def run_service(cfg: BenchConfig) -> int:
+    check_port_available(cfg.service_port)
+
+    if not cfg.foreground:
+        logfile = os.path.join(
+            os.environ.get("TMPDIR", "/tmp"),
+            f"elbencho_amd_service.{os.getuid()}.{cfg.service_port}.log")
+        print(f"Starting service as daemon on port {cfg.service_port}. Log: {logfile}")
+        daemonize(logfile)
+
+    state = ServiceState(cfg)
+    Handler.state = state
+    server = ThreadingHTTPServer(("", cfg.service_port), Handler)
+    server.daemon_threads = True
+
+    print(f"elbencho-amd service v{VERSION} listening on port {cfg.service_port}")
+
+    t = threading.Thread(target=server.serve_forever, daemon=True)
+    t.start()
+    try:
+        while not state.quit_requested.is_set():
+            time.sleep(0.2)
+    except KeyboardInterrupt:
+        pass
+    server.shutdown()
+    return 0

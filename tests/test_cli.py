@@ -1,0 +1,130 @@
+"""CLI end-to-end: elbencho-compatible invocations, CSV/JSON output."""
+
+import csv
+import json
+import os
+
+import pytest
+
+from elbencho_amd.cli import main
+from elbencho_amd.stats import CSV_COLUMNS
+
+
+def test_multifile_create_read_delete(tmp_path, capsys):
+    # mirrors reference tools/test-examples.sh multifile cases (:226-274)
+    base = str(tmp_path)
+    rc = main(["-t", "2", "-d", "-n", "3", "-w", "-N", "4", "-s", "1m", "-b", "1m",
+               "--lat", "--verify", "1", "--no0usecerr", "--nolive", base])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "OPERATION" in out and "MKDIRS" in out and "WRITE" in out
+    assert "FIRST DONE" in out and "LAST DONE" in out
+
+    rc = main(["-t", "2", "-n", "3", "-r", "-N", "4", "-s", "1m", "-b", "128k",
+               "--verify", "1", "--nolive", base])
+    assert rc == 0
+
+    rc = main(["-t", "2", "-n", "3", "-N", "4", "-F", "-D", "--nolive", base])
+    assert rc == 0
+    assert list(tmp_path.iterdir()) == []
+
+
+def test_file_mode_with_csv_json(tmp_path):
+    f = tmp_path / "file1"
+    csvf = tmp_path / "res.csv"
+    jsonf = tmp_path / "res.json"
+    rc = main(["-w", "-r", "-t", "2", "-b", "256k", "-s", "2m", "--nolive",
+               "--csvfile", str(csvf), "--jsonfile", str(jsonf),
+               "--label", "testrun", str(f)])
+    assert rc == 0
+
+    with open(csvf, newline="") as fh:
+        rows = list(csv.reader(fh))
+    assert rows[0] == CSV_COLUMNS
+    assert len(rows) == 3  # header + WRITE + READ
+    op_idx = CSV_COLUMNS.index("operation")
+    assert rows[1][op_idx] == "WRITE"
+    assert rows[2][op_idx] == "READ"
+    lbl_idx = CSV_COLUMNS.index("label")
+    assert rows[1][lbl_idx] == "testrun"
+
+    with open(jsonf) as fh:
+        docs = [json.loads(ln) for ln in fh]
+    assert [d["phase_type"] for d in docs] == ["WRITE", "READ"]
+    assert docs[0]["last_done"]["bytes"] == 2 * 1024 * 1024
+    assert docs[0]["label"] == "testrun"
+
+
+def test_csv_schema_guard(tmp_path):
+    f = tmp_path / "file1"
+    csvf = tmp_path / "res.csv"
+    csvf.write_text("foo,bar\n1,2\n")
+    rc = main(["-w", "-t", "1", "-b", "64k", "-s", "64k", "--nolive",
+               "--csvfile", str(csvf), str(f)])
+    assert rc == 1  # refused: incompatible CSV
+
+
+def test_path_bracket_expansion(tmp_path):
+    rc = main(["-w", "-t", "1", "-b", "64k", "-s", "128k", "--nolive",
+               str(tmp_path / "f[1-3]")])
+    assert rc == 0
+    for i in (1, 2, 3):
+        assert (tmp_path / f"f{i}").stat().st_size == 128 * 1024
+
+
+def test_dryrun(tmp_path, capsys):
+    rc = main(["-w", "-t", "2", "-b", "64k", "-s", "1m", "--dryrun", "--nolive",
+               str(tmp_path / "f")])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "DRY RUN" in out
+    assert "WRITE" in out
+    assert str(1024 * 1024) in out
+
+
+def test_version():
+    with pytest.raises(SystemExit) as e:
+        main(["--version"])
+    assert e.value.code == 0
+
+
+def test_no_phase_selected(tmp_path):
+    rc = main(["--nolive", str(tmp_path / "f")])
+    assert rc == 1
+
+
+def test_resfile(tmp_path):
+    f = tmp_path / "file1"
+    res = tmp_path / "results.txt"
+    rc = main(["-w", "-t", "1", "-b", "64k", "-s", "256k", "--nolive",
+               "--resfile", str(res), str(f)])
+    assert rc == 0
+    assert "WRITE" in res.read_text()
+
+
+def test_iterations(tmp_path, capsys):
+    f = tmp_path / "file1"
+    rc = main(["-w", "-t", "1", "-b", "64k", "-s", "64k", "-i", "2", "--nolive", str(f)])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert out.count("WRITE") == 2
+    assert "iteration 2 of 2" in out
+
+
+def test_timelimit_interrupts(tmp_path, capsys):
+    f = tmp_path / "file1"
+    rc = main(["-w", "-t", "1", "-b", "4k", "-s", "1g", "--timelimit", "1",
+               "--limitwrite", "4m", "--liveint", "100", "--nolive", str(f)])
+    # interrupted phase reports worker errors -> nonzero rc
+    assert rc == 1
+    err_out = capsys.readouterr()
+    assert "time limit" in (err_out.err + err_out.out).lower()
+
+
+def test_config_file(tmp_path):
+    f = tmp_path / "file1"
+    conf = tmp_path / "bench.conf"
+    conf.write_text("write=true\nsize=128k\nblock=64k\nthreads=1\n")
+    rc = main(["--nolive", "-c", str(conf), str(f)])
+    assert rc == 0
+    assert f.stat().st_size == 128 * 1024

@@ -1,0 +1,204 @@
+"""Native engine end-to-end behavior on tmp filesystems (CPU paths)."""
+
+import os
+
+import pytest
+
+
+def run_phase(core, eng, name):
+    eng.start_phase(core.PHASES[name])
+    assert eng.wait_phase_done(60_000)
+    res = eng.finish_phase()
+    errs = [r["error"] for r in res if r["error"]]
+    assert not errs, errs
+    return res
+
+
+def test_file_write_read_verify(core, tmp_path):
+    paths = [str(tmp_path / "a"), str(tmp_path / "b")]
+    size = 3 * 1024 * 1024 + 1234  # odd tail exercises short blocks
+    cfg = dict(paths=paths, path_type="file", threads=4, num_dataset_threads=4,
+               file_size=size, block_size=256 * 1024, verify_salt=7, lat=True)
+    eng = core.Engine(cfg)
+    eng.prepare()
+
+    res = run_phase(core, eng, "WRITE")
+    assert sum(r["bytes"] for r in res) == size * 2
+    assert os.path.getsize(paths[0]) == size
+    assert os.path.getsize(paths[1]) == size
+
+    # whole-file checksum pattern is valid
+    with open(paths[0], "rb") as f:
+        data = f.read()
+    assert core.verify_checksum(data, 0, 7) == 2**64 - 1
+
+    res = run_phase(core, eng, "READ")
+    assert sum(r["bytes"] for r in res) == size * 2
+
+    # stonewall fields are sane
+    for r in res:
+        assert r["stonewall_bytes"] <= r["bytes"]
+        assert r["stonewall_elapsed_usec"] <= max(x["elapsed_usec"] for x in res)
+
+    # latency histograms carry values
+    assert sum(r["io_lat"][0] for r in res) == sum(r["iops"] for r in res)
+
+
+def test_verify_detects_corruption(core, tmp_path):
+    p = str(tmp_path / "f")
+    cfg = dict(paths=[p], path_type="file", threads=1, num_dataset_threads=1,
+               file_size=64 * 1024, block_size=16 * 1024, verify_salt=5)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    run_phase(core, eng, "WRITE")
+
+    with open(p, "r+b") as f:
+        f.seek(20000)
+        f.write(b"\xde\xad")
+
+    eng.start_phase(core.PHASES["READ"])
+    eng.wait_phase_done(60_000)
+    res = eng.finish_phase()
+    errs = [r["error"] for r in res if r["error"]]
+    assert errs and "verification failed" in errs[0].lower()
+
+
+def test_uring_iodepth(core, tmp_path):
+    p = str(tmp_path / "f")
+    size = 8 * 1024 * 1024
+    cfg = dict(paths=[p], path_type="file", threads=2, num_dataset_threads=2,
+               file_size=size, block_size=64 * 1024, iodepth=8, verify_salt=3, lat=True)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    res = run_phase(core, eng, "WRITE")
+    assert sum(r["bytes"] for r in res) == size
+    res = run_phase(core, eng, "READ")
+    assert sum(r["bytes"] for r in res) == size
+    assert sum(r["iops"] for r in res) == size // (64 * 1024)
+
+
+def test_random_full_coverage_write(core, tmp_path):
+    p = str(tmp_path / "f")
+    size = 4 * 1024 * 1024
+    cfg = dict(paths=[p], path_type="file", threads=2, num_dataset_threads=2,
+               file_size=size, block_size=16 * 1024, random=True, verify_salt=9)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    run_phase(core, eng, "WRITE")
+    # full coverage: file is complete and every byte matches the pattern
+    assert os.path.getsize(p) == size
+    with open(p, "rb") as f:
+        assert core.verify_checksum(f.read(), 0, 9) == 2**64 - 1
+
+
+def test_strided_write(core, tmp_path):
+    p = str(tmp_path / "f")
+    size = 2 * 1024 * 1024
+    cfg = dict(paths=[p], path_type="file", threads=4, num_dataset_threads=4,
+               file_size=size, block_size=64 * 1024, strided=True, verify_salt=2)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    res = run_phase(core, eng, "WRITE")
+    assert sum(r["bytes"] for r in res) == size
+    with open(p, "rb") as f:
+        assert core.verify_checksum(f.read(), 0, 2) == 2**64 - 1
+
+
+def test_backward_read(core, tmp_path):
+    p = str(tmp_path / "f")
+    size = 1024 * 1024
+    cfg = dict(paths=[p], path_type="file", threads=1, num_dataset_threads=1,
+               file_size=size, block_size=64 * 1024, verify_salt=4)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    run_phase(core, eng, "WRITE")
+    cfg["backward"] = True
+    eng2 = core.Engine(cfg)
+    eng2.prepare()
+    res = run_phase(core, eng2, "READ")
+    assert sum(r["bytes"] for r in res) == size
+
+
+def test_dir_mode_lifecycle(core, tmp_path):
+    cfg = dict(paths=[str(tmp_path)], path_type="dir", threads=2, num_dataset_threads=2,
+               dirs=3, files=4, file_size=64 * 1024, block_size=64 * 1024,
+               verify_salt=1, lat=True)
+    eng = core.Engine(cfg)
+    eng.prepare()
+
+    res = run_phase(core, eng, "MKDIRS")
+    assert sum(r["entries"] for r in res) == 6  # 2 threads x 3 dirs
+
+    res = run_phase(core, eng, "WRITE")
+    assert sum(r["entries"] for r in res) == 24  # 2 x 3 x 4
+    assert sum(r["bytes"] for r in res) == 24 * 64 * 1024
+    # reference-compatible layout: r{rank}/d{dir}/r{rank}-f{file}
+    assert (tmp_path / "r0" / "d0" / "r0-f0").exists()
+    assert (tmp_path / "r1" / "d2" / "r1-f3").exists()
+
+    res = run_phase(core, eng, "STAT")
+    assert sum(r["entries"] for r in res) == 24
+
+    res = run_phase(core, eng, "READ")
+    assert sum(r["bytes"] for r in res) == 24 * 64 * 1024
+
+    res = run_phase(core, eng, "RMFILES")
+    assert sum(r["entries"] for r in res) == 24
+
+    res = run_phase(core, eng, "RMDIRS")
+    assert sum(r["entries"] for r in res) == 6
+    assert list(tmp_path.iterdir()) == []
+
+
+def test_dir_mode_no_subdirs(core, tmp_path):
+    cfg = dict(paths=[str(tmp_path)], path_type="dir", threads=2, num_dataset_threads=2,
+               dirs=0, files=3, file_size=4096, block_size=4096)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    res = run_phase(core, eng, "WRITE")
+    assert sum(r["entries"] for r in res) == 6
+    assert (tmp_path / "r0-f0").exists()
+    assert (tmp_path / "r1-f2").exists()
+    run_phase(core, eng, "RMFILES")
+
+
+def test_interrupt(core, tmp_path):
+    p = str(tmp_path / "big")
+    cfg = dict(paths=[p], path_type="file", threads=1, num_dataset_threads=1,
+               file_size=1 << 30, block_size=4096,
+               limit_write_bps=10 * 1024 * 1024)  # slow it down
+    eng = core.Engine(cfg)
+    eng.prepare()
+    eng.start_phase(core.PHASES["WRITE"])
+    assert not eng.wait_phase_done(200)
+    eng.interrupt()
+    assert eng.wait_phase_done(10_000)
+    res = eng.finish_phase()
+    assert res[0]["error"] == "interrupted"
+
+
+def test_planned_work_matches_actual(core, tmp_path):
+    paths = [str(tmp_path / "x"), str(tmp_path / "y")]
+    size = 1024 * 1024 + 777
+    cfg = dict(paths=paths, path_type="file", threads=3, num_dataset_threads=3,
+               file_size=size, block_size=64 * 1024)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    planned_entries, planned_bytes = eng.planned_work(core.PHASES["WRITE"])
+    res = run_phase(core, eng, "WRITE")
+    assert sum(r["bytes"] for r in res) == planned_bytes == size * 2
+
+
+def test_rank_offset_partitioning(core, tmp_path):
+    """Two single-thread instances with rank offsets cover the file exactly."""
+    p = str(tmp_path / "f")
+    size = 2 * 1024 * 1024
+    for rank in (0, 1):
+        cfg = dict(paths=[p], path_type="file", threads=1, num_dataset_threads=2,
+                   rank_offset=rank, file_size=size, block_size=64 * 1024, verify_salt=6)
+        eng = core.Engine(cfg)
+        eng.prepare()
+        res = run_phase(core, eng, "WRITE")
+        assert sum(r["bytes"] for r in res) == size // 2
+    with open(p, "rb") as f:
+        assert core.verify_checksum(f.read(), 0, 6) == 2**64 - 1

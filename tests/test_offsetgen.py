@@ -1,0 +1,92 @@
+"""Offset generator semantics (native implementations via test hook)."""
+
+
+
+def test_sequential(core):
+    offs, total = core.gen_offsets("seq", 4096, 0, 4096 * 4 + 100)
+    assert total == 4096 * 4 + 100
+    assert offs == [(0, 4096), (4096, 4096), (8192, 4096), (12288, 4096), (16384, 100)]
+
+
+def test_reverse(core):
+    offs, total = core.gen_offsets("reverse", 4096, 0, 4096 * 2 + 100)
+    assert total == 4096 * 2 + 100
+    # tail first, then full blocks walking backwards
+    assert offs == [(8192, 100), (4096, 4096), (0, 4096)]
+
+
+def test_random_aligned_bounds(core):
+    bs = 4096
+    n = 64
+    offs, total = core.gen_offsets("random_aligned", bs, bs * 10, bs * n, seed=42)
+    assert total == bs * n
+    assert len(offs) == n
+    for off, ln in offs:
+        assert ln == bs
+        assert off >= bs * 10
+        assert (off - bs * 10) % bs == 0
+        assert off + ln <= bs * 10 + bs * n
+
+
+def test_random_unaligned_amount(core):
+    bs = 4096
+    offs, total = core.gen_offsets("random", bs, 0, 1 << 20, seed=7, amount=10 * bs)
+    assert total == 10 * bs
+    assert sum(ln for _, ln in offs) == 10 * bs
+    for off, ln in offs:
+        assert off + ln <= 1 << 20
+
+
+def test_full_coverage_visits_every_block_once(core):
+    bs = 4096
+    for nblocks in (1, 2, 7, 16, 33, 1000):
+        offs, total = core.gen_offsets("full_coverage", bs, 0, bs * nblocks, seed=3)
+        assert total == bs * nblocks
+        seen = sorted(off for off, _ in offs)
+        assert seen == [i * bs for i in range(nblocks)]
+
+
+def test_full_coverage_tail(core):
+    bs = 4096
+    offs, total = core.gen_offsets("full_coverage", bs, 0, bs * 5 + 17, seed=3)
+    assert total == bs * 5 + 17
+    lens = sorted(ln for _, ln in offs)
+    assert lens == [17, bs, bs, bs, bs, bs]
+
+
+def test_full_coverage_is_permuted(core):
+    bs = 4096
+    offs, _ = core.gen_offsets("full_coverage", bs, 0, bs * 256, seed=5)
+    seq = [off for off, _ in offs]
+    assert seq != sorted(seq), "full coverage order should not be sequential"
+
+
+def test_strided_partitions_range(core):
+    bs = 4096
+    nblocks = 10
+    nranks = 3
+    all_offs = []
+    for rank in range(nranks):
+        offs, _ = core.gen_offsets("strided", bs, 0, bs * nblocks, 1, rank, nranks)
+        for off, ln in offs:
+            assert (off // bs) % nranks == rank
+        all_offs.extend(off for off, _ in offs)
+    assert sorted(all_offs) == [i * bs for i in range(nblocks)]
+
+
+def test_checksum_fill_verify_roundtrip(core):
+    data = core.fill_checksum(4096, 8192, 1234)
+    assert core.verify_checksum(data, 8192, 1234) == 2**64 - 1  # UINT64_MAX = ok
+    bad = bytearray(data)
+    bad[100] ^= 0xFF
+    assert core.verify_checksum(bytes(bad), 8192, 1234) == 8192 + 100
+
+
+def test_checksum_unaligned(core):
+    # unaligned offset and odd length
+    data = core.fill_checksum(1000, 12345, 77)
+    assert core.verify_checksum(data, 12345, 77) == 2**64 - 1
+    # concatenation property: two adjacent fills equal one big fill
+    a = core.fill_checksum(500, 12345, 77)
+    b = core.fill_checksum(500, 12845, 77)
+    assert a + b == data

@@ -1,0 +1,117 @@
+"""Distributed service/master mode over localhost HTTP (no real cluster),
+mirroring reference tools/test-examples.sh:293-342."""
+
+import json
+import os
+import socket
+import subprocess
+import sys
+import time
+import urllib.request
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def free_port() -> int:
+    s = socket.socket()
+    s.bind(("", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+@pytest.fixture
+def services():
+    ports = [free_port(), free_port()]
+    procs = []
+    env = dict(os.environ, PYTHONPATH=REPO)
+    for p in ports:
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "elbencho_amd", "--service", "--foreground",
+             "--port", str(p)],
+            env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+    # wait for readiness
+    deadline = time.monotonic() + 15
+    for p in ports:
+        while True:
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{p}/protocolversion", timeout=1) as r:
+                    r.read()
+                break
+            except OSError:
+                if time.monotonic() > deadline:
+                    for pr in procs:
+                        pr.kill()
+                    raise RuntimeError("service did not become ready")
+                time.sleep(0.1)
+    yield ports
+    for pr in procs:
+        pr.terminate()
+        try:
+            pr.wait(5)
+        except subprocess.TimeoutExpired:
+            pr.kill()
+
+
+def run_master(args):
+    env = dict(os.environ, PYTHONPATH=REPO)
+    return subprocess.run([sys.executable, "-m", "elbencho_amd", "--nolive"] + args,
+                          env=env, capture_output=True, text=True, timeout=120)
+
+
+def test_service_endpoints(services):
+    port = services[0]
+    with urllib.request.urlopen(f"http://127.0.0.1:{port}/info", timeout=5) as r:
+        assert b"elbencho-amd service" in r.read()
+    with urllib.request.urlopen(f"http://127.0.0.1:{port}/status", timeout=5) as r:
+        st = json.loads(r.read())
+    assert st["idle"] is True
+
+
+def test_distributed_dir_mode(services, tmp_path):
+    hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+    res = run_master(["--hosts", hosts, "-t", "2", "-d", "-n", "2", "-w", "-r",
+                      "-N", "3", "-s", "16k", "-F", "-D", "--verify", "1",
+                      str(tmp_path)])
+    assert res.returncode == 0, res.stdout + res.stderr
+    # 2 services x 2 threads x 2 dirs x 3 files
+    assert "Files total" in res.stdout
+    assert list(tmp_path.iterdir()) == []
+    # master output must aggregate both services:
+    # 2 services x 2 threads x 2 dirs x 3 files = 24
+    for line in res.stdout.splitlines():
+        if "Files total" in line:
+            assert line.split()[-1] == "24"
+            break
+
+
+def test_distributed_shared_file(services, tmp_path):
+    hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+    f = tmp_path / "shared"
+    res = run_master(["--hosts", hosts, "-t", "2", "-w", "-r", "-s", "1m",
+                      "-b", "64k", "--verify", "2", str(f)])
+    assert res.returncode == 0, res.stdout + res.stderr
+    assert f.stat().st_size == 1024 * 1024
+    # whole file written exactly once across 2 services x 2 threads
+    from elbencho_amd import load_core
+    core = load_core()
+    assert core.verify_checksum(f.read_bytes(), 0, 2) == 2**64 - 1
+
+
+def test_quit_services(services):
+    hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+    res = run_master(["--hosts", hosts, "--quit"])
+    assert res.returncode == 0
+    # services exit shortly after
+    deadline = time.monotonic() + 10
+    while time.monotonic() < deadline:
+        try:
+            urllib.request.urlopen(
+                f"http://127.0.0.1:{services[0]}/status", timeout=1).read()
+            time.sleep(0.2)
+        except OSError:
+            return
+    raise AssertionError("service did not quit")

@@ -1,0 +1,101 @@
+"""Unit parsing, config validation, path expansion."""
+
+import pytest
+
+from elbencho_amd.config import BenchConfig, ConfigError, expand_path_brackets
+from elbencho_amd.units import elapsed_ms_to_human, parse_size
+
+
+def test_parse_size_base2():
+    assert parse_size("4k") == 4096
+    assert parse_size("4K") == 4096
+    assert parse_size("1m") == 1 << 20
+    assert parse_size("1M") == 1 << 20
+    assert parse_size("2g") == 2 << 30
+    assert parse_size("1t") == 1 << 40
+    assert parse_size("512") == 512
+    assert parse_size("128K") == 128 * 1024
+    assert parse_size("4KiB") == 4096
+    assert parse_size("4KB", base10=False) == 4096
+    assert parse_size("0") == 0
+    assert parse_size(None) == 0
+    assert parse_size(1234) == 1234
+    assert parse_size("1.5k") == 1536
+
+
+def test_elapsed_human():
+    assert elapsed_ms_to_human(1) == "1ms"
+    assert elapsed_ms_to_human(1001) == "1.001s"
+    assert elapsed_ms_to_human(123456) == "2m3.456s"
+    assert elapsed_ms_to_human((3 * 3600 + 25 * 60 + 45) * 1000) == "3h25m45s"
+
+
+def test_bracket_expansion():
+    assert expand_path_brackets("/mnt/f[1-3]") == ["/mnt/f1", "/mnt/f2", "/mnt/f3"]
+    assert expand_path_brackets("/mnt/f[1,5]") == ["/mnt/f1", "/mnt/f5"]
+    assert expand_path_brackets("/mnt/f") == ["/mnt/f"]
+    assert expand_path_brackets("h:[1711-1712]") == ["h:1711", "h:1712"]
+    # nested/multiple bracket pairs expand recursively
+    assert expand_path_brackets("/a[1-2]/b[1-2]") == [
+        "/a1/b1", "/a1/b2", "/a2/b1", "/a2/b2"]
+
+
+def test_phase_order():
+    cfg = BenchConfig()
+    cfg.run_write = cfg.run_read = cfg.run_stat = True
+    cfg.run_mkdirs = cfg.run_deldirs = cfg.run_delfiles = True
+    assert cfg.phase_list() == ["MKDIRS", "WRITE", "STAT", "READ", "RMFILES", "RMDIRS"]
+
+
+def test_validation_errors(tmp_path):
+    cfg = BenchConfig()
+    cfg.run_write = True
+    with pytest.raises(ConfigError):
+        cfg.finalize()  # no paths
+
+    cfg = BenchConfig()
+    cfg.paths = [str(tmp_path)]
+    cfg.run_write = True
+    with pytest.raises(ConfigError):  # dir mode needs -N
+        cfg.finalize()
+
+    cfg = BenchConfig()
+    cfg.paths = [str(tmp_path / "f1")]
+    cfg.run_write = True
+    cfg.file_size = 1024
+    cfg.direct = True
+    cfg.block_size = 100  # not 512-aligned
+    with pytest.raises(ConfigError):
+        cfg.finalize()
+
+
+def test_wire_roundtrip(tmp_path):
+    cfg = BenchConfig()
+    cfg.paths = [str(tmp_path / "x")]
+    cfg.run_write = True
+    cfg.file_size = 4096
+    cfg.block_size = 1024
+    cfg.threads = 3
+    cfg.gpu_ids = [0, 1]
+    cfg.finalize()
+    wire = cfg.to_wire()
+    # master-only keys stripped
+    assert "hosts" not in wire and "csv_file" not in wire
+    cfg2 = BenchConfig.from_wire(wire)
+    assert cfg2.threads == 3
+    assert cfg2.gpu_ids == [0, 1]
+    assert cfg2.file_size == 4096
+    assert cfg2.path_type == cfg.path_type
+
+
+def test_engine_dict(tmp_path):
+    cfg = BenchConfig()
+    cfg.paths = [str(tmp_path / "x")]
+    cfg.run_write = True
+    cfg.file_size = 4096
+    cfg.threads = 2
+    cfg.finalize()
+    d = cfg.engine_dict()
+    assert d["threads"] == 2
+    assert d["num_dataset_threads"] == 2
+    assert d["path_type"] == "file"
