@@ -185,6 +185,31 @@ PYBIND11_MODULE(_core, m)
         return py::bytes(ctx.hostBuf(0), len);
     }, py::arg("len"), py::arg("refill_len"), py::arg("seed"), py::arg("dev") = 0);
 
+    // Kernel micro-benchmark: bandwidth of the gfx950 fill/verify kernels and
+    // the staging copies on one device buffer (GB/s, stream-synchronized).
+    m.def("gpu_kernel_bench", [](uint64_t len, int iters, int dev) {
+        GpuCtx ctx(dev, 1, len, true);
+        auto timeIt = [&](auto&& fn) {
+            fn(); // warmup
+            ctx.syncStream();
+            auto t0 = std::chrono::steady_clock::now();
+            for (int i = 0; i < iters; i++) fn();
+            ctx.syncStream();
+            auto dt = std::chrono::duration<double>(std::chrono::steady_clock::now() - t0)
+                          .count();
+            return (double)len * iters / dt / 1e9; // GB/s
+        };
+        py::dict d;
+        d["fill_rand_gbps"] = timeIt([&] { ctx.fillRandDev(0, len, 42); });
+        d["fill_checksum_gbps"] = timeIt([&] { ctx.fillChecksumDev(0, len, 0, 7); });
+        d["blockvar_gbps"] = timeIt([&] { ctx.blockVarRefillDev(0, len, len / 2, 9); });
+        ctx.fillChecksumDev(0, len, 0, 7);
+        d["verify_gbps"] = timeIt([&] { (void)ctx.verifyChecksumDev(0, len, 0, 7); });
+        d["d2h_gbps"] = timeIt([&] { ctx.copyD2HAsync(0, len); ctx.syncStream(); });
+        d["h2d_gbps"] = timeIt([&] { ctx.copyH2DAsync(0, len); ctx.syncStream(); });
+        return d;
+    }, py::arg("len") = (uint64_t)256 << 20, py::arg("iters") = 10, py::arg("dev") = 0);
+
     py::class_<Engine>(m, "Engine")
         .def(py::init([](const py::dict& cfg) {
             return std::make_unique<Engine>(configFromDict(cfg));
