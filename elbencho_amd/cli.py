@@ -93,6 +93,19 @@ def build_parser() -> argparse.ArgumentParser:
                    help="Verify each block by read-back immediately after writing.")
     g.add_argument("--readinline", action="store_true",
                    help="Read files immediately after writing in the same phase.")
+    g.add_argument("--statinline", action="store_true",
+                   help="Stat (fstat) each file right after open in dir mode.")
+    g.add_argument("--mmap", action="store_true",
+                   help="Use memory mapping (mmap + memcpy) instead of read/write calls.")
+    g.add_argument("--fadv", default="", metavar="LIST",
+                   help="Comma-separated fadvise flags applied after open: "
+                        "seq,rand,willneed,dontneed,noreuse.")
+    g.add_argument("--madv", default="", metavar="LIST",
+                   help="Comma-separated madvise flags for --mmap: "
+                        "seq,rand,willneed,dontneed,hugepage,nohugepage.")
+    g.add_argument("--flock", default="", metavar="MODE", choices=["", "range", "full"],
+                   help="Lock file ranges ('range') or whole files ('full') around "
+                        "each I/O via fcntl.")
     g.add_argument("--blockvarpct", type=int, default=100, metavar="PCT",
                    help="Percentage of each block refilled with random data between writes. "
                         "(Default: 100)")
@@ -325,6 +338,11 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.verify = args.verify
     cfg.verify_direct = args.verifydirect
     cfg.read_inline = args.readinline
+    cfg.stat_inline = args.statinline
+    cfg.mmap = args.mmap
+    cfg.fadv = args.fadv
+    cfg.madv = args.madv
+    cfg.flock_mode = args.flock
     cfg.blockvar_pct = args.blockvarpct
     cfg.blockvar_algo = args.blockvaralgo
 

@@ -104,6 +104,11 @@ class BenchConfig:
     verify: int = -1               # --verify SALT (-1 off)
     verify_direct: bool = False    # --verifydirect
     read_inline: bool = False      # --readinline
+    stat_inline: bool = False      # --statinline
+    mmap: bool = False             # --mmap
+    fadv: str = ""                 # --fadv (csv: seq,rand,willneed,dontneed,noreuse)
+    madv: str = ""                 # --madv (csv: seq,rand,willneed,dontneed,hugepage)
+    flock_mode: str = ""           # --flock ("range"|"full")
     blockvar_pct: int = 100        # --blockvarpct
     blockvar_algo: str = "fast"    # --blockvaralgo
 
@@ -305,6 +310,13 @@ class BenchConfig:
         if self.verify >= 0 and self.blockvar_pct and False:
             pass  # verify overrides block variance; no error
 
+    def _rwmix_threads_effective(self) -> int:
+        if self.rwmix_threads:
+            return self.rwmix_threads
+        if self.rwmix_thread_pct:
+            return max(1, self.threads * self.rwmix_thread_pct // 100)
+        return 0
+
     # ------------------------------------------------------------------
     def phase_list(self) -> list[str]:
         """Ordered phase names for one iteration (reference order,
@@ -368,6 +380,18 @@ class BenchConfig:
             gpu_ids=self.gpu_ids,
             gpu_pinned=self.gpu_pinned,
             lat=self.lat,
+            rwmix_pct=self.rwmix_pct,
+            rwmix_threads=self._rwmix_threads_effective(),
+            mmap=self.mmap,
+            fadv_flags=_fadv_to_bits(self.fadv),
+            madv_flags=_madv_to_flags(self.madv),
+            flock_mode={"": 0, "range": 1, "full": 2}[self.flock_mode],
+            read_inline=self.read_inline,
+            stat_inline=self.stat_inline,
+            ops_log=self.ops_log_path,
+            ops_log_lock=self.ops_log_lock,
+            cores=_parse_int_list(self.cpu_cores),
+            zones=_parse_int_list(self.numa_zones),
             limit_read_bps=self.limit_read,
             limit_write_bps=self.limit_write,
             ignore_del_errors=self.ignore_del_errors,
@@ -375,6 +399,46 @@ class BenchConfig:
             inf_loop=self.inf_loop,
             bench_seed=self.bench_seed or 0x243F6A8885A308D3,
         )
+
+
+# --fadv bit values must match csrc/engine.cpp FadvBits
+_FADV_BITS = {"seq": 1, "sequential": 1, "rand": 2, "random": 2, "willneed": 4,
+              "dontneed": 8, "noreuse": 16}
+
+# madvise flag values from <sys/mman.h>
+_MADV_FLAGS = {"seq": 2, "sequential": 2, "rand": 1, "random": 1, "willneed": 3,
+               "dontneed": 4, "hugepage": 14, "nohugepage": 15}
+
+
+def _fadv_to_bits(csv_str: str) -> int:
+    bits = 0
+    for tok in csv_str.split(","):
+        tok = tok.strip().lower()
+        if not tok:
+            continue
+        if tok not in _FADV_BITS:
+            raise ConfigError(f"unknown fadvise flag: {tok}")
+        bits |= _FADV_BITS[tok]
+    return bits
+
+
+def _madv_to_flags(csv_str: str) -> int:
+    # madvise advices are not bitmask-combinable; take the last one given
+    flags = 0
+    for tok in csv_str.split(","):
+        tok = tok.strip().lower()
+        if not tok:
+            continue
+        if tok not in _MADV_FLAGS:
+            raise ConfigError(f"unknown madvise flag: {tok}")
+        flags = _MADV_FLAGS[tok]
+    return flags
+
+
+def _parse_int_list(s: str) -> list[int]:
+    if not s:
+        return []
+    return [int(x) for x in s.replace(",", " ").split()]
 
 
 def parse_gpu_ids(s: str) -> list[int]:

@@ -11,7 +11,10 @@
 
 #include <fcntl.h>
 #include <linux/fs.h>
+#include <pthread.h>
+#include <sched.h>
 #include <sys/ioctl.h>
+#include <sys/mman.h>
 #include <sys/stat.h>
 #include <sys/types.h>
 #include <unistd.h>
@@ -125,6 +128,7 @@ Worker::~Worker()
 {
     if (ownHostBufs)
         for (auto p : hostBufs) free(p);
+    free(scratchBuf);
 
     if (gpu) { // return the GPU context for reuse by the next phase
         std::lock_guard<std::mutex> lk(eng.gpuCacheMtx);
@@ -132,6 +136,14 @@ Worker::~Worker()
             eng.gpuCtxCache.resize(localRank + 1);
         eng.gpuCtxCache[localRank] = std::move(gpu);
     }
+}
+
+bool Worker::rwMixDecideRead()
+{
+    bool doRead = (rwMixReads * 100) < (rwMixOps * (uint64_t)eng.cfg.rwMixPct);
+    rwMixOps++;
+    if (doRead) rwMixReads++;
+    return doRead;
 }
 
 void Worker::checkInterrupt()
@@ -167,6 +179,49 @@ void Worker::allocBuffers()
         for (int i = 0; i < slots; i++) gpu->copyH2DAsync(i, cfg.blockSize);
         gpu->syncStream();
     }
+
+    if (cfg.verifyDirect && posix_memalign((void**)&scratchBuf, 4096, cfg.blockSize))
+        throw WorkerError("scratch buffer allocation failed");
+}
+
+// CPU core / NUMA zone binding (reference analogue: Worker.cpp:102-146 +
+// NumaTk). Zones bind to all CPUs listed in /sys/devices/system/node.
+void Worker::applyBinding()
+{
+    const auto& cfg = eng.cfg;
+
+    cpu_set_t set;
+    CPU_ZERO(&set);
+    bool haveSet = false;
+
+    if (!cfg.cpuCores.empty()) {
+        int core = cfg.cpuCores[globalRank % cfg.cpuCores.size()];
+        CPU_SET(core, &set);
+        haveSet = true;
+    } else if (!cfg.numaZones.empty()) {
+        int zone = cfg.numaZones[globalRank % cfg.numaZones.size()];
+        std::string path = "/sys/devices/system/node/node" + std::to_string(zone) + "/cpulist";
+        FILE* f = fopen(path.c_str(), "r");
+        if (!f) throw WorkerError("unknown NUMA zone " + std::to_string(zone));
+        char buf[512] = {0};
+        if (fgets(buf, sizeof(buf), f)) {
+            // parse "0-15,32-47" style lists
+            char* save = nullptr;
+            for (char* tok = strtok_r(buf, ",\n", &save); tok;
+                 tok = strtok_r(nullptr, ",\n", &save)) {
+                int lo, hi;
+                if (sscanf(tok, "%d-%d", &lo, &hi) == 2)
+                    for (int i = lo; i <= hi; i++) CPU_SET(i, &set);
+                else if (sscanf(tok, "%d", &lo) == 1)
+                    CPU_SET(lo, &set);
+            }
+            haveSet = true;
+        }
+        fclose(f);
+    }
+
+    if (haveSet && pthread_setaffinity_np(pthread_self(), sizeof(set), &set))
+        throw WorkerError("CPU affinity binding failed");
 }
 
 void Worker::setupGpu()
@@ -243,11 +298,43 @@ std::unique_ptr<OffsetGen> Worker::makeOffsetGen(uint64_t rangeStart, uint64_t r
     return gen;
 }
 
-// single synchronous block I/O incl. GPU staging + verify hooks
-ssize_t Worker::blockIO(bool isWrite, int fd, int slot, uint64_t len, uint64_t fileOff)
+namespace {
+
+// fcntl advisory lock guard (reference analogue: FileTk::flock range/full)
+struct FlockGuard {
+    int fd = -1;
+    struct flock fl{};
+
+    FlockGuard(int fd_, int mode, bool isWrite, uint64_t off, uint64_t len) : fd(fd_)
+    {
+        if (!mode) { fd = -1; return; }
+        fl.l_type = isWrite ? F_WRLCK : F_RDLCK;
+        fl.l_whence = SEEK_SET;
+        fl.l_start = (mode == 2) ? 0 : (off_t)off;
+        fl.l_len = (mode == 2) ? 0 : (off_t)len;
+        if (fcntl(fd, F_SETLKW, &fl)) throw WorkerError("flock (fcntl F_SETLKW) failed");
+    }
+
+    ~FlockGuard()
+    {
+        if (fd < 0) return;
+        fl.l_type = F_UNLCK;
+        fcntl(fd, F_SETLK, &fl);
+    }
+};
+
+} // namespace
+
+// single synchronous block I/O incl. GPU staging, verify hooks, mmap path,
+// flock and ops logging
+ssize_t Worker::blockIO(bool isWrite, int fd, int slot, uint64_t len, uint64_t fileOff,
+                        char* mmapBase, const std::string* path)
 {
     const auto& cfg = eng.cfg;
     char* buf = hostBufs[slot];
+    const bool logOps = eng.opsLog.isEnabled();
+    static const std::string emptyPath;
+    const std::string& target = path ? *path : emptyPath;
 
     if (isWrite) {
         rateLimiter.wait(len);
@@ -258,12 +345,38 @@ ssize_t Worker::blockIO(bool isWrite, int fd, int slot, uint64_t len, uint64_t f
             gpu->syncStream();
         }
 
-        ssize_t res = pwrite(fd, buf, len, fileOff);
-        if (res >= 0 && cfg.fsyncPerFile) { /* fsync handled at file close */ }
+        FlockGuard lock(fd, cfg.flockMode, true, fileOff, len);
+
+        ssize_t res;
+        if (mmapBase) {
+            if (logOps) eng.opsLog.log(globalRank, "memcpy_w", target, fileOff, len, true, false);
+            std::memcpy(mmapBase + fileOff, buf, len);
+            res = (ssize_t)len;
+            if (logOps) eng.opsLog.log(globalRank, "memcpy_w", target, fileOff, len, false, false);
+        } else {
+            if (logOps) eng.opsLog.log(globalRank, "pwrite", target, fileOff, len, true, false);
+            res = pwrite(fd, buf, len, fileOff);
+            if (logOps) eng.opsLog.log(globalRank, "pwrite", target, fileOff, len, false, res < 0);
+        }
+
+        if (res == (ssize_t)len && cfg.verifyDirect) verifyDirectReadback(fd, slot, len, fileOff);
         return res;
     } else {
         rateLimiter.wait(len);
-        ssize_t res = pread(fd, buf, len, fileOff);
+
+        FlockGuard lock(fd, cfg.flockMode, false, fileOff, len);
+
+        ssize_t res;
+        if (mmapBase) {
+            if (logOps) eng.opsLog.log(globalRank, "memcpy_r", target, fileOff, len, true, false);
+            std::memcpy(buf, mmapBase + fileOff, len);
+            res = (ssize_t)len;
+            if (logOps) eng.opsLog.log(globalRank, "memcpy_r", target, fileOff, len, false, false);
+        } else {
+            if (logOps) eng.opsLog.log(globalRank, "pread", target, fileOff, len, true, false);
+            res = pread(fd, buf, len, fileOff);
+            if (logOps) eng.opsLog.log(globalRank, "pread", target, fileOff, len, false, res < 0);
+        }
         if (res < 0) return res;
 
         if (gpu) { // stage host->GPU (HBM3E resident buffers)
@@ -274,6 +387,18 @@ ssize_t Worker::blockIO(bool isWrite, int fd, int slot, uint64_t len, uint64_t f
         postReadCheck(slot, (uint64_t)res, fileOff);
         return res;
     }
+}
+
+// --verifydirect: read each block back right after writing and compare
+void Worker::verifyDirectReadback(int fd, int slot, uint64_t len, uint64_t fileOff)
+{
+    ssize_t res = pread(fd, scratchBuf, len, fileOff);
+    if (res != (ssize_t)len)
+        throw WorkerError("verify-direct readback failed at offset " +
+                          std::to_string(fileOff));
+    if (std::memcmp(scratchBuf, hostBufs[slot], len))
+        throw WorkerError("verify-direct mismatch: block at offset " +
+                          std::to_string(fileOff) + " differs after readback");
 }
 
 void Worker::preWriteFill(int slot, uint64_t len, uint64_t fileOff)
@@ -356,6 +481,54 @@ struct FdGuard {
     }
 };
 
+// per-file mmap guard (reference analogue: FileTk::mmapAndMadvise)
+struct MmapGuard {
+    std::vector<std::pair<char*, uint64_t>> maps;
+
+    void map(int fd, uint64_t len, bool writable, int madvFlags, const std::string& path)
+    {
+        int prot = writable ? (PROT_READ | PROT_WRITE) : PROT_READ;
+        void* p = mmap(nullptr, len, prot, MAP_SHARED, fd, 0);
+        if (p == MAP_FAILED)
+            throw WorkerError("mmap failed. Path: " + path + "; SysErr: " + strerror(errno));
+        if (madvFlags && madvise(p, len, madvFlags))
+            throw WorkerError("madvise failed. Path: " + path);
+        maps.emplace_back((char*)p, len);
+    }
+
+    char* base(uint64_t idx) const
+    {
+        return idx < maps.size() ? maps[idx].first : nullptr;
+    }
+
+    ~MmapGuard()
+    {
+        for (auto& [p, len] : maps) munmap(p, len);
+    }
+};
+
+// --fadv: bitmask encodes the POSIX_FADV_* advices to apply in order
+enum FadvBits {
+    FADV_BIT_SEQ = 1,
+    FADV_BIT_RAND = 2,
+    FADV_BIT_WILLNEED = 4,
+    FADV_BIT_DONTNEED = 8,
+    FADV_BIT_NOREUSE = 16,
+};
+
+inline void applyFadvise(int fd, int bits, const std::string& path)
+{
+    if (!bits) return;
+    struct { int bit; int advice; } table[] = {
+        {FADV_BIT_SEQ, POSIX_FADV_SEQUENTIAL},   {FADV_BIT_RAND, POSIX_FADV_RANDOM},
+        {FADV_BIT_WILLNEED, POSIX_FADV_WILLNEED}, {FADV_BIT_DONTNEED, POSIX_FADV_DONTNEED},
+        {FADV_BIT_NOREUSE, POSIX_FADV_NOREUSE},
+    };
+    for (auto& e : table)
+        if ((bits & e.bit) && posix_fadvise(fd, 0, 0, e.advice))
+            throw WorkerError("posix_fadvise failed. Path: " + path);
+}
+
 } // namespace
 
 void Worker::fileModeBlocks(bool isWrite)
@@ -367,10 +540,15 @@ void Worker::fileModeBlocks(bool isWrite)
     const uint64_t bs = cfg.blockSize;
     const size_t numFiles = cfg.paths.size();
     const uint64_t numBlocksPerFile = (fileSize + bs - 1) / bs;
+    const bool rwMixActive = isWrite && (cfg.rwMixPct > 0 || isDedicatedReader);
 
     // open all files
     FdGuard fg;
-    int openFlags = (isWrite ? O_WRONLY : O_RDONLY) | (cfg.directIO ? O_DIRECT : 0);
+    MmapGuard mg;
+    int rwFlags = isWrite
+                      ? ((cfg.verifyDirect || rwMixActive || cfg.useMmap) ? O_RDWR : O_WRONLY)
+                      : O_RDONLY;
+    int openFlags = rwFlags | (cfg.directIO ? O_DIRECT : 0);
     if (isWrite && cfg.pathType == PathType::FILE) openFlags |= O_CREAT;
     for (const auto& p : cfg.paths) {
         int fd = open(p.c_str(), openFlags, 0644);
@@ -378,10 +556,12 @@ void Worker::fileModeBlocks(bool isWrite)
         fg.fds.push_back(fd);
         if (isWrite && cfg.pathType == PathType::FILE) {
             if (cfg.truncate && ftruncate(fd, 0)) throwErrno("truncate", p);
-            if (cfg.truncToSize != UINT64_MAX && ftruncate(fd, fileSize))
+            if ((cfg.truncToSize != UINT64_MAX || cfg.useMmap) && ftruncate(fd, fileSize))
                 throwErrno("truncate-to-size", p);
             if (cfg.preallocFile && fallocate(fd, 0, 0, fileSize)) throwErrno("fallocate", p);
         }
+        applyFadvise(fd, cfg.fadviseFlags, p);
+        if (cfg.useMmap) mg.map(fd, fileSize, isWrite, cfg.madviseFlags, p);
     }
 
     // Virtual concatenated range: sequential uses ceil blocks per file (the
@@ -431,24 +611,36 @@ void Worker::fileModeBlocks(bool isWrite)
         uint64_t fileIdx, inFileOff, ioLen;
         if (!mapBlock(spec, fileIdx, inFileOff, ioLen)) continue;
 
+        // rwmix: dedicated readers always read; otherwise hold reads/total
+        // at rwMixPct (reference --rwmixpct / --rwmixthr semantics)
+        bool mixRead = rwMixActive && (isDedicatedReader || rwMixDecideRead());
+        bool blockWrite = isWrite && !mixRead;
+
         auto t0 = lat ? Clock::now() : Clock::time_point();
 
-        ssize_t res = blockIO(isWrite, fg.fds[fileIdx], 0, ioLen, inFileOff);
+        ssize_t res = blockIO(blockWrite, fg.fds[fileIdx], 0, ioLen, inFileOff,
+                              mg.base(fileIdx), &cfg.paths[fileIdx]);
         if (res < 0)
-            throwErrno(isWrite ? "write" : "read", cfg.paths[fileIdx]);
+            throwErrno(blockWrite ? "write" : "read", cfg.paths[fileIdx]);
         if ((uint64_t)res != ioLen)
-            throw WorkerError(std::string("unexpected short ") + (isWrite ? "write" : "read") +
+            throw WorkerError(std::string("unexpected short ") + (blockWrite ? "write" : "read") +
                               ". Path: " + cfg.paths[fileIdx] +
                               "; expected: " + std::to_string(ioLen) +
-                              "; got: " + std::to_string(res));
+                              "; got: " + std::to_string(res) +
+                              (mixRead ? "; Hint: rwmix reads need a pre-written file." : ""));
 
         if (lat)
             addIoLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
-                Clock::now() - t0).count());
+                Clock::now() - t0).count(), mixRead);
 
-        liveOps.bytes.fetch_add(ioLen, std::memory_order_relaxed);
-        liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+        AtomicLiveOps& ops = mixRead ? liveOpsReadMix : liveOps;
+        ops.bytes.fetch_add(ioLen, std::memory_order_relaxed);
+        ops.iops.fetch_add(1, std::memory_order_relaxed);
     }
+
+    if (isWrite && cfg.fsyncPerFile)
+        for (size_t i = 0; i < fg.fds.size(); i++)
+            if (fsync(fg.fds[i])) throwErrno("fsync", cfg.paths[i]);
 }
 
 void Worker::fileModeBlocksUring(bool isWrite)
@@ -461,8 +653,11 @@ void Worker::fileModeBlocksUring(bool isWrite)
     const int depth = cfg.ioDepth;
     const bool lat = cfg.measureLat;
 
+    const bool rwMixActive = isWrite && (cfg.rwMixPct > 0 || isDedicatedReader);
+
     FdGuard fg;
-    int openFlags = (isWrite ? O_WRONLY : O_RDONLY) | (cfg.directIO ? O_DIRECT : 0);
+    int rwFlags = isWrite ? ((cfg.verifyDirect || rwMixActive) ? O_RDWR : O_WRONLY) : O_RDONLY;
+    int openFlags = rwFlags | (cfg.directIO ? O_DIRECT : 0);
     if (isWrite && cfg.pathType == PathType::FILE) openFlags |= O_CREAT;
     for (const auto& p : cfg.paths) {
         int fd = open(p.c_str(), openFlags, 0644);
@@ -474,6 +669,7 @@ void Worker::fileModeBlocksUring(bool isWrite)
                 throwErrno("truncate-to-size", p);
             if (cfg.preallocFile && fallocate(fd, 0, 0, fileSize)) throwErrno("fallocate", p);
         }
+        applyFadvise(fd, cfg.fadviseFlags, p);
     }
 
     const uint64_t mapBPF = (cfg.random || cfg.strided) ? (fileSize / bs) : numBlocksPerFile;
@@ -516,6 +712,7 @@ void Worker::fileModeBlocksUring(bool isWrite)
         uint64_t len = 0;
         Clock::time_point start;
         int fileIdx = 0;
+        bool isWriteOp = true;
     };
     std::vector<SlotState> slots(depth);
 
@@ -529,20 +726,21 @@ void Worker::fileModeBlocksUring(bool isWrite)
         uint64_t fileIdx, inFileOff, ioLen;
         if (!mapBlock(spec, fileIdx, inFileOff, ioLen)) return true; // skip this block
 
-        if (isWrite) {
-            rateLimiter.wait(ioLen);
+        bool mixRead = rwMixActive && (isDedicatedReader || rwMixDecideRead());
+        bool blockWrite = isWrite && !mixRead;
+
+        rateLimiter.wait(ioLen);
+        if (blockWrite) {
             preWriteFill(slot, ioLen, inFileOff);
             if (gpu) {
                 gpu->copyD2HAsync(slot, ioLen);
                 gpu->syncStream();
             }
-        } else {
-            rateLimiter.wait(ioLen);
         }
 
         slots[slot] = {inFileOff, ioLen, lat ? Clock::now() : Clock::time_point(),
-                       (int)fileIdx};
-        if (!ring.prep(isWrite, fg.fds[fileIdx], hostBufs[slot], ioLen, inFileOff,
+                       (int)fileIdx, blockWrite};
+        if (!ring.prep(blockWrite, fg.fds[fileIdx], hostBufs[slot], ioLen, inFileOff,
                        (uint64_t)slot))
             throw WorkerError("io_uring SQ unexpectedly full");
         inFlight++;
@@ -564,29 +762,33 @@ void Worker::fileModeBlocksUring(bool isWrite)
             SlotState& st = slots[slot];
             inFlight--;
 
+            const bool wasWrite = st.isWriteOp;
             if (comps[i].res < 0)
-                throw WorkerError(std::string("async ") + (isWrite ? "write" : "read") +
+                throw WorkerError(std::string("async ") + (wasWrite ? "write" : "read") +
                                   " failed. Path: " + cfg.paths[st.fileIdx] +
                                   "; SysErr: " + strerror(-comps[i].res));
             if ((uint64_t)comps[i].res != st.len)
                 throw WorkerError(std::string("unexpected short async ") +
-                                  (isWrite ? "write" : "read") +
+                                  (wasWrite ? "write" : "read") +
                                   ". Path: " + cfg.paths[st.fileIdx]);
 
-            if (!isWrite) {
+            if (!wasWrite) {
                 if (gpu) {
                     gpu->copyH2DAsync(slot, st.len);
                     gpu->syncStream();
                 }
-                postReadCheck(slot, st.len, st.inFileOff);
+                if (!isWrite) // rwmix reads in a write phase skip pattern checks
+                    postReadCheck(slot, st.len, st.inFileOff);
             }
 
+            const bool mixRead = isWrite && !wasWrite;
             if (lat)
                 addIoLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
-                    Clock::now() - st.start).count());
+                    Clock::now() - st.start).count(), mixRead);
 
-            liveOps.bytes.fetch_add(st.len, std::memory_order_relaxed);
-            liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+            AtomicLiveOps& ops = mixRead ? liveOpsReadMix : liveOps;
+            ops.bytes.fetch_add(st.len, std::memory_order_relaxed);
+            ops.iops.fetch_add(1, std::memory_order_relaxed);
 
             if ((opCount++ % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
 
@@ -628,16 +830,17 @@ void Worker::dirModeMkdirs()
 {
     const auto& cfg = eng.cfg;
     const bool lat = cfg.measureLat;
+    const int dirRank = cfg.dirSharing ? 0 : globalRank; // --dirsharing
     char rel[64];
 
     // rank root dir under every bench path that this worker will use
-    snprintf(rel, sizeof(rel), "r%d", globalRank);
+    snprintf(rel, sizeof(rel), "r%d", dirRank);
     for (const auto& p : cfg.paths) mkdirIgnoreExists(p + "/" + rel);
 
     for (uint64_t d = 0; d < cfg.numDirs; d++) {
         if ((d % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
-        size_t pathIdx = (globalRank + d) % cfg.paths.size();
-        snprintf(rel, sizeof(rel), "r%d/d%lu", globalRank, (unsigned long)d);
+        size_t pathIdx = (dirRank + d) % cfg.paths.size();
+        snprintf(rel, sizeof(rel), "r%d/d%lu", dirRank, (unsigned long)d);
         std::string full = cfg.paths[pathIdx] + "/" + rel;
 
         auto t0 = lat ? Clock::now() : Clock::time_point();
@@ -653,12 +856,16 @@ void Worker::dirModeRmdirs()
 {
     const auto& cfg = eng.cfg;
     const bool lat = cfg.measureLat;
+    const int dirRank = cfg.dirSharing ? 0 : globalRank;
     char rel[64];
+
+    // with --dirsharing only one worker removes the shared dirs
+    if (cfg.dirSharing && globalRank != cfg.rankOffset) return;
 
     for (uint64_t d = 0; d < cfg.numDirs; d++) {
         if ((d % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
-        size_t pathIdx = (globalRank + d) % cfg.paths.size();
-        snprintf(rel, sizeof(rel), "r%d/d%lu", globalRank, (unsigned long)d);
+        size_t pathIdx = (dirRank + d) % cfg.paths.size();
+        snprintf(rel, sizeof(rel), "r%d/d%lu", dirRank, (unsigned long)d);
         std::string full = cfg.paths[pathIdx] + "/" + rel;
 
         auto t0 = lat ? Clock::now() : Clock::time_point();
@@ -672,7 +879,7 @@ void Worker::dirModeRmdirs()
     }
 
     // rank root dirs (not counted as entries)
-    snprintf(rel, sizeof(rel), "r%d", globalRank);
+    snprintf(rel, sizeof(rel), "r%d", dirRank);
     for (const auto& p : cfg.paths) {
         std::string full = p + "/" + rel;
         if (rmdir(full.c_str()) && errno != ENOENT && errno != ENOTEMPTY) {
@@ -692,10 +899,18 @@ void Worker::dirModeFiles(Phase phase)
     const uint64_t fileSize = cfg.fileSize;
     char rel[128];
 
+    const bool rwMixActive = isWrite && (cfg.rwMixPct > 0 || isDedicatedReader);
+
     int openFlags = 0;
-    if (isWrite) openFlags = O_CREAT | O_WRONLY;
+    if (isWrite)
+        openFlags = O_CREAT | ((cfg.verifyDirect || cfg.readInline || rwMixActive ||
+                                cfg.useMmap) ? O_RDWR : O_WRONLY);
     if (isRead) openFlags = O_RDONLY;
     if (cfg.directIO) openFlags |= O_DIRECT;
+
+    // --dirsharing: all threads work in the dirs of rank 0 (file names keep
+    // the per-rank prefix, so files stay unique; reference workerDirRank)
+    const int dirRank = cfg.dirSharing ? 0 : globalRank;
 
     // offsets within one file
     std::unique_ptr<OffsetGen> gen;
@@ -704,9 +919,9 @@ void Worker::dirModeFiles(Phase phase)
         for (uint64_t f = 0; f < cfg.numFiles; f++) {
             if ((f % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
 
-            size_t pathIdx = (globalRank + d) % cfg.paths.size();
+            size_t pathIdx = (dirRank + d) % cfg.paths.size();
             if (haveSubdirs)
-                snprintf(rel, sizeof(rel), "r%d/d%lu/r%d-f%lu", globalRank, (unsigned long)d,
+                snprintf(rel, sizeof(rel), "r%d/d%lu/r%d-f%lu", dirRank, (unsigned long)d,
                          globalRank, (unsigned long)f);
             else
                 snprintf(rel, sizeof(rel), "r%d-f%lu", globalRank, (unsigned long)f);
@@ -718,33 +933,73 @@ void Worker::dirModeFiles(Phase phase)
             switch (phase) {
                 case Phase::WRITE:
                 case Phase::READ: {
+                    if (eng.opsLog.isEnabled())
+                        eng.opsLog.log(globalRank, "open", full, 0, 0, true, false);
                     int fd = open(full.c_str(), openFlags, 0644);
                     if (fd < 0) throwErrno("open", full);
+                    if (eng.opsLog.isEnabled())
+                        eng.opsLog.log(globalRank, "open", full, 0, 0, false, false);
 
+                    MmapGuard mg;
                     try {
+                        if (cfg.statInline) { // --statinline: fstat right after open
+                            struct stat st;
+                            if (fstat(fd, &st)) throwErrno("fstat", full);
+                        }
+
                         if (isWrite) {
                             if (cfg.truncate && ftruncate(fd, 0)) throwErrno("truncate", full);
-                            if (cfg.truncToSize != UINT64_MAX && ftruncate(fd, fileSize))
+                            if ((cfg.truncToSize != UINT64_MAX || cfg.useMmap) &&
+                                ftruncate(fd, fileSize))
                                 throwErrno("truncate-to-size", full);
                             if (cfg.preallocFile && fallocate(fd, 0, 0, fileSize))
                                 throwErrno("fallocate", full);
                         }
+
+                        applyFadvise(fd, cfg.fadviseFlags, full);
+                        if (cfg.useMmap) mg.map(fd, fileSize, isWrite, cfg.madviseFlags, full);
 
                         if (!gen) gen = makeOffsetGen(0, fileSize);
                         else gen->reset(0, fileSize);
 
                         BlockSpec spec;
                         while (gen->next(spec)) {
+                            bool mixRead =
+                                rwMixActive && (isDedicatedReader || rwMixDecideRead());
+                            bool blockWrite = isWrite && !mixRead;
                             auto t0 = lat ? Clock::now() : Clock::time_point();
-                            ssize_t res = blockIO(isWrite, fd, 0, spec.len, spec.offset);
-                            if (res < 0) throwErrno(isWrite ? "write" : "read", full);
+                            ssize_t res = blockIO(blockWrite, fd, 0, spec.len, spec.offset,
+                                                  mg.base(0), &full);
+                            if (res < 0) throwErrno(blockWrite ? "write" : "read", full);
                             if ((uint64_t)res != spec.len)
                                 throw WorkerError("unexpected short I/O on " + full);
                             if (lat)
                                 addIoLat((uint64_t)std::chrono::duration_cast<
-                                    std::chrono::microseconds>(Clock::now() - t0).count());
-                            liveOps.bytes.fetch_add(spec.len, std::memory_order_relaxed);
-                            liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+                                    std::chrono::microseconds>(Clock::now() - t0).count(),
+                                    mixRead);
+                            AtomicLiveOps& ops = mixRead ? liveOpsReadMix : liveOps;
+                            ops.bytes.fetch_add(spec.len, std::memory_order_relaxed);
+                            ops.iops.fetch_add(1, std::memory_order_relaxed);
+                        }
+
+                        if (isWrite && cfg.readInline) {
+                            // --readinline: read the file back within the write
+                            // phase; bytes accounted as rwmix reads
+                            gen->reset(0, fileSize);
+                            while (gen->next(spec)) {
+                                auto t0 = lat ? Clock::now() : Clock::time_point();
+                                ssize_t res = blockIO(false, fd, 0, spec.len, spec.offset,
+                                                      mg.base(0), &full);
+                                if (res != (ssize_t)spec.len)
+                                    throwErrno("inline readback", full);
+                                if (lat)
+                                    addIoLat((uint64_t)std::chrono::duration_cast<
+                                        std::chrono::microseconds>(Clock::now() - t0).count(),
+                                        true);
+                                liveOpsReadMix.bytes.fetch_add(spec.len,
+                                                               std::memory_order_relaxed);
+                                liveOpsReadMix.iops.fetch_add(1, std::memory_order_relaxed);
+                            }
                         }
 
                         if (isWrite && cfg.fsyncPerFile && fsync(fd)) throwErrno("fsync", full);
@@ -858,24 +1113,37 @@ void Worker::threadMain()
     try {
         const auto& cfg = eng.cfg;
 
+        applyBinding();
+
         rng.reset(makeRandAlgo(cfg.randAlgo,
                                cfg.benchSeed ^ (0xBF58476D1CE4E5B9ULL * (globalRank + 1))));
         fillRng.reset(makeRandAlgo(cfg.blockVarAlgo,
                                    cfg.benchSeed ^ (0x94D049BB133111EBULL * (globalRank + 1))));
 
+        // --rwmixthr: first N threads of a write phase only read
+        isDedicatedReader =
+            (eng.currentPhase == Phase::WRITE) && (localRank < cfg.rwMixThreads);
+
         // buffers only needed for data phases
         if (eng.currentPhase == Phase::WRITE || eng.currentPhase == Phase::READ)
             allocBuffers();
 
-        rateLimiter.init(eng.currentPhase == Phase::WRITE ? eng.cfg.limitWriteBps
-                                                          : eng.cfg.limitReadBps);
+        rateLimiter.init((eng.currentPhase == Phase::WRITE && !isDedicatedReader)
+                             ? eng.cfg.limitWriteBps
+                             : eng.cfg.limitReadBps);
 
         { // start gate: all workers begin simultaneously
             std::unique_lock<std::mutex> lk(eng.gateMtx);
             eng.gateCv.wait(lk, [&] { return eng.gateOpen; });
         }
 
-        runPhase();
+        const bool loopingPhase =
+            cfg.infiniteLoop && (eng.currentPhase == Phase::WRITE ||
+                                 eng.currentPhase == Phase::READ ||
+                                 eng.currentPhase == Phase::STAT);
+        do {
+            runPhase();
+        } while (loopingPhase && !eng.interruptFlag.load(std::memory_order_relaxed));
     } catch (const InterruptedError&) {
         error = "interrupted";
         hadError = true;
@@ -945,6 +1213,8 @@ void Engine::startPhase(Phase phase)
     if (phaseRunning) throw std::runtime_error("phase already running");
 
     currentPhase = phase;
+    if (!cfg.opsLogPath.empty() && !opsLog.isEnabled())
+        opsLog.open(cfg.opsLogPath, cfg.opsLogLock);
     interruptFlag.store(false);
     workersDone.store(0);
     workersWithError.store(0);
@@ -978,11 +1248,14 @@ void Engine::onWorkerDone(Worker& w, bool hadError)
         // worker's live counters ("first done" result)
         bool didWork = w.liveOps.entries.load(std::memory_order_relaxed) ||
                        w.liveOps.bytes.load(std::memory_order_relaxed) ||
-                       w.liveOps.iops.load(std::memory_order_relaxed);
+                       w.liveOps.iops.load(std::memory_order_relaxed) ||
+                       w.liveOpsReadMix.bytes.load(std::memory_order_relaxed) ||
+                       w.liveOpsReadMix.iops.load(std::memory_order_relaxed);
         if (didWork) {
             uint64_t elapsed = nowUSecSince(phaseStart);
             for (auto& peer : workers) {
                 peer->stonewallOps.takeFrom(peer->liveOps);
+                peer->stonewallOpsReadMix.takeFrom(peer->liveOpsReadMix);
                 peer->stonewallElapsedUSec = elapsed;
             }
             stonewallTriggered.store(true);
@@ -1017,8 +1290,10 @@ Engine::LivePoll Engine::poll()
 
     for (auto& w : workers) {
         lp.entries += w->liveOps.entries.load(std::memory_order_relaxed);
-        lp.bytes += w->liveOps.bytes.load(std::memory_order_relaxed);
-        lp.iops += w->liveOps.iops.load(std::memory_order_relaxed);
+        lp.bytes += w->liveOps.bytes.load(std::memory_order_relaxed) +
+                    w->liveOpsReadMix.bytes.load(std::memory_order_relaxed);
+        lp.iops += w->liveOps.iops.load(std::memory_order_relaxed) +
+                   w->liveOpsReadMix.iops.load(std::memory_order_relaxed);
         lp.latNumIOs += w->liveIoLatNum.load(std::memory_order_relaxed);
         lp.latSumIOs += w->liveIoLatSum.load(std::memory_order_relaxed);
         lp.latNumEntries += w->liveEntryLatNum.load(std::memory_order_relaxed);
@@ -1040,11 +1315,16 @@ std::vector<WorkerResult> Engine::finishPhase()
         r.rank = w->globalRank;
         r.elapsedUSec = w->elapsedUSec;
         r.total.takeFrom(w->liveOps);
+        r.totalReadMix.takeFrom(w->liveOpsReadMix);
         r.stonewall = stonewallTriggered.load() ? w->stonewallOps : r.total;
+        r.stonewallReadMix =
+            stonewallTriggered.load() ? w->stonewallOpsReadMix : r.totalReadMix;
         r.stonewallElapsedUSec =
             stonewallTriggered.load() ? w->stonewallElapsedUSec : w->elapsedUSec;
         r.ioLatVec = w->ioLat.toVec();
         r.entryLatVec = w->entryLat.toVec();
+        r.ioLatReadMixVec = w->ioLatReadMix.toVec();
+        r.entryLatReadMixVec = w->entryLatReadMix.toVec();
         r.error = w->error;
         results.push_back(std::move(r));
     }

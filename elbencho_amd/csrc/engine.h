@@ -24,6 +24,7 @@
 #include "gpu.h"
 #include "histogram.h"
 #include "offsetgen.h"
+#include "opslog.h"
 #include "rand.h"
 #include "rate.h"
 
@@ -57,9 +58,25 @@ struct EngineConfig {
 
     int64_t verifySalt = -1; // -1 = off
     bool verifyDirect = false;
+    bool readInline = false;  // read each file back right after writing it
+    bool statInline = false;  // fstat right after open in dir mode
     int blockVarPct = 100;
     std::string blockVarAlgo = "fast";
     std::string randAlgo = "fast";
+
+    int rwMixPct = 0;     // % of blocks read instead of written in a write phase
+    int rwMixThreads = 0; // first N threads of a write phase only read
+
+    bool useMmap = false;
+    int fadviseFlags = 0;  // POSIX_FADV_* bitmask-ish (applied in order)
+    int madviseFlags = 0;  // MADV_* combined
+    int flockMode = 0;     // 0 none, 1 range, 2 full file
+
+    std::string opsLogPath;
+    bool opsLogLock = false;
+
+    std::vector<int> cpuCores;  // round-robin thread->core binding
+    std::vector<int> numaZones; // round-robin thread->NUMA-zone binding
 
     std::vector<int> gpuIDs; // empty = CPU buffers only
     bool gpuPinnedHostBufs = true;
@@ -80,9 +97,13 @@ struct WorkerResult {
     uint64_t elapsedUSec = 0;
     LiveOpsSnapshot total;
     LiveOpsSnapshot stonewall;
+    LiveOpsSnapshot totalReadMix;     // reads done within a write phase
+    LiveOpsSnapshot stonewallReadMix;
     uint64_t stonewallElapsedUSec = 0;
     std::vector<uint64_t> ioLatVec;
     std::vector<uint64_t> entryLatVec;
+    std::vector<uint64_t> ioLatReadMixVec;
+    std::vector<uint64_t> entryLatReadMixVec;
     std::string error;
 };
 
@@ -97,7 +118,9 @@ public:
     void threadMain();
 
     AtomicLiveOps liveOps;
+    AtomicLiveOps liveOpsReadMix; // rwmix reads within a write phase
     LiveOpsSnapshot stonewallOps;
+    LiveOpsSnapshot stonewallOpsReadMix;
     uint64_t stonewallElapsedUSec = 0;
 
     // histograms are worker-private until the thread is joined; the live
@@ -107,6 +130,8 @@ public:
 
     LatencyHistogram ioLat;
     LatencyHistogram entryLat;
+    LatencyHistogram ioLatReadMix;
+    LatencyHistogram entryLatReadMix;
 
     uint64_t elapsedUSec = 0;
     std::string error;
@@ -131,12 +156,15 @@ private:
     void anyModeDropCaches();
 
     // per-block helpers (sync path)
-    void addIoLat(uint64_t us)
+    void addIoLat(uint64_t us, bool readMix = false)
     {
-        ioLat.add(us);
+        (readMix ? ioLatReadMix : ioLat).add(us);
         liveIoLatNum.fetch_add(1, std::memory_order_relaxed);
         liveIoLatSum.fetch_add(us, std::memory_order_relaxed);
     }
+    // rwmix decision: keep reads/total at rwMixPct (write phase only)
+    bool rwMixDecideRead(); // defined in engine.cpp (needs Engine)
+    void applyBinding();
     void addEntryLat(uint64_t us)
     {
         entryLat.add(us);
@@ -145,7 +173,9 @@ private:
     }
     void preWriteFill(int slot, uint64_t len, uint64_t fileOff);
     void postReadCheck(int slot, uint64_t len, uint64_t fileOff);
-    ssize_t blockIO(bool isWrite, int fd, int slot, uint64_t len, uint64_t fileOff);
+    ssize_t blockIO(bool isWrite, int fd, int slot, uint64_t len, uint64_t fileOff,
+                    char* mmapBase = nullptr, const std::string* path = nullptr);
+    void verifyDirectReadback(int fd, int slot, uint64_t len, uint64_t fileOff);
 
     // setup
     void allocBuffers();
@@ -156,6 +186,9 @@ private:
     Engine& eng;
     std::vector<char*> hostBufs; // CPU-owned unless GPU mode (then GpuCtx owns)
     bool ownHostBufs = false;
+    char* scratchBuf = nullptr;  // verify-direct readback buffer
+    uint64_t rwMixOps = 0, rwMixReads = 0;
+    bool isDedicatedReader = false; // rwMixThreads role
     std::unique_ptr<GpuCtx> gpu;
     std::unique_ptr<RandAlgo> rng;        // offsets
     std::unique_ptr<RandAlgo> fillRng;    // block variance fill
@@ -217,6 +250,8 @@ public:
     // expensive; reusing them keeps phase setup off the measured path)
     std::mutex gpuCacheMtx;
     std::vector<std::unique_ptr<GpuCtx>> gpuCtxCache;
+
+    OpsLogger opsLog;
 
     void onWorkerDone(Worker& w, bool hadError);
 

@@ -56,6 +56,18 @@ EngineConfig configFromDict(const py::dict& d)
     c.fsyncPerFile = getB("fsync", false);
     c.verifySalt = getI("verify_salt", -1);
     c.verifyDirect = getB("verify_direct", false);
+    c.readInline = getB("read_inline", false);
+    c.statInline = getB("stat_inline", false);
+    c.rwMixPct = (int)getI("rwmix_pct", 0);
+    c.rwMixThreads = (int)getI("rwmix_threads", 0);
+    c.useMmap = getB("mmap", false);
+    c.fadviseFlags = (int)getI("fadv_flags", 0);
+    c.madviseFlags = (int)getI("madv_flags", 0);
+    c.flockMode = (int)getI("flock_mode", 0);
+    c.opsLogPath = getS("ops_log", "");
+    c.opsLogLock = getB("ops_log_lock", false);
+    if (d.contains("cores")) c.cpuCores = d["cores"].cast<std::vector<int>>();
+    if (d.contains("zones")) c.numaZones = d["zones"].cast<std::vector<int>>();
     c.blockVarPct = (int)getI("blockvar_pct", 100);
     c.blockVarAlgo = getS("blockvar_algo", "fast");
     c.randAlgo = getS("rand_algo", "balanced_single");
@@ -84,8 +96,16 @@ py::dict resultToDict(const WorkerResult& r)
     d["stonewall_bytes"] = r.stonewall.bytes;
     d["stonewall_iops"] = r.stonewall.iops;
     d["stonewall_elapsed_usec"] = r.stonewallElapsedUSec;
+    d["rm_entries"] = r.totalReadMix.entries;
+    d["rm_bytes"] = r.totalReadMix.bytes;
+    d["rm_iops"] = r.totalReadMix.iops;
+    d["rm_stonewall_entries"] = r.stonewallReadMix.entries;
+    d["rm_stonewall_bytes"] = r.stonewallReadMix.bytes;
+    d["rm_stonewall_iops"] = r.stonewallReadMix.iops;
     d["io_lat"] = r.ioLatVec;
     d["entry_lat"] = r.entryLatVec;
+    d["io_lat_rm"] = r.ioLatReadMixVec;
+    d["entry_lat_rm"] = r.entryLatReadMixVec;
     d["error"] = r.error;
     return d;
 }

@@ -73,8 +73,16 @@ class WorkerStats:
     stonewall_entries: int = 0
     stonewall_bytes: int = 0
     stonewall_iops: int = 0
+    rm_entries: int = 0
+    rm_bytes: int = 0
+    rm_iops: int = 0
+    rm_stonewall_entries: int = 0
+    rm_stonewall_bytes: int = 0
+    rm_stonewall_iops: int = 0
     io_lat: list[int] = field(default_factory=list)
     entry_lat: list[int] = field(default_factory=list)
+    io_lat_rm: list[int] = field(default_factory=list)
+    entry_lat_rm: list[int] = field(default_factory=list)
     error: str = ""
     num_workers: int = 1  # >1 when this row aggregates a remote service
 
@@ -90,8 +98,16 @@ class WorkerStats:
             stonewall_entries=d["stonewall_entries"],
             stonewall_bytes=d["stonewall_bytes"],
             stonewall_iops=d["stonewall_iops"],
+            rm_entries=d.get("rm_entries", 0),
+            rm_bytes=d.get("rm_bytes", 0),
+            rm_iops=d.get("rm_iops", 0),
+            rm_stonewall_entries=d.get("rm_stonewall_entries", 0),
+            rm_stonewall_bytes=d.get("rm_stonewall_bytes", 0),
+            rm_stonewall_iops=d.get("rm_stonewall_iops", 0),
             io_lat=list(d["io_lat"]),
             entry_lat=list(d["entry_lat"]),
+            io_lat_rm=list(d.get("io_lat_rm", [])),
+            entry_lat_rm=list(d.get("entry_lat_rm", [])),
             error=d["error"],
         )
 
@@ -111,8 +127,17 @@ class PhaseResults:
     sw_entries: int = 0
     sw_bytes: int = 0
     sw_iops: int = 0
+    # rwmix read totals (reads within a write phase)
+    rm_entries: int = 0
+    rm_bytes: int = 0
+    rm_iops: int = 0
+    rm_sw_entries: int = 0
+    rm_sw_bytes: int = 0
+    rm_sw_iops: int = 0
     io_lat: Histogram = field(default_factory=Histogram)
     entry_lat: Histogram = field(default_factory=Histogram)
+    io_lat_rm: Histogram = field(default_factory=Histogram)
+    entry_lat_rm: Histogram = field(default_factory=Histogram)
     cpu_first: int = 0
     cpu_last: int = 0
     worker_elapsed_usec: list[int] = field(default_factory=list)
@@ -146,10 +171,20 @@ def aggregate_phase(phase_name: str, phase_id: str, start_time: float,
         r.sw_entries += w.stonewall_entries
         r.sw_bytes += w.stonewall_bytes
         r.sw_iops += w.stonewall_iops
+        r.rm_entries += w.rm_entries
+        r.rm_bytes += w.rm_bytes
+        r.rm_iops += w.rm_iops
+        r.rm_sw_entries += w.rm_stonewall_entries
+        r.rm_sw_bytes += w.rm_stonewall_bytes
+        r.rm_sw_iops += w.rm_stonewall_iops
         if w.io_lat:
             r.io_lat.merge(w.io_lat)
         if w.entry_lat:
             r.entry_lat.merge(w.entry_lat)
+        if w.io_lat_rm:
+            r.io_lat_rm.merge(w.io_lat_rm)
+        if w.entry_lat_rm:
+            r.entry_lat_rm.merge(w.entry_lat_rm)
         r.worker_elapsed_usec.append(w.elapsed_usec)
         if w.error:
             r.errors.append(f"Rank {w.rank}: {w.error}")
@@ -190,16 +225,33 @@ def print_phase_results(cfg: BenchConfig, r: PhaseResults, out=None) -> None:
             rows.append(("Dirs/s", r.per_sec_first(r.sw_entries) // cfg.files,
                          r.per_sec_last(r.entries) // cfg.files))
 
+    is_rwmix = bool(r.rm_bytes or r.rm_iops or r.rm_entries)
+    mib = 1024 * 1024
+
     if r.iops:
         # suppress IOPS when it would equal files/s (dir mode, 1 block per file)
         if (cfg.path_type != PATH_DIR) or (cfg.block_size != cfg.file_size) or not r.entries:
-            rows.append(("IOPS", r.per_sec_first(r.sw_iops), r.per_sec_last(r.iops)))
+            rows.append(("IOPS write" if is_rwmix else "IOPS",
+                         r.per_sec_first(r.sw_iops), r.per_sec_last(r.iops)))
+    if is_rwmix and r.rm_iops:
+        rows.append(("IOPS read", r.per_sec_first(r.rm_sw_iops), r.per_sec_last(r.rm_iops)))
+        rows.append(("IOPS total", r.per_sec_first(r.sw_iops + r.rm_sw_iops),
+                     r.per_sec_last(r.iops + r.rm_iops)))
 
     if r.bytes:
-        mib = 1024 * 1024
-        rows.append(("Throughput MiB/s", r.per_sec_first(r.sw_bytes) // mib,
-                     r.per_sec_last(r.bytes) // mib))
-        rows.append(("Total MiB", r.sw_bytes // mib, r.bytes // mib))
+        rows.append(("MiB/s write" if is_rwmix else "Throughput MiB/s",
+                     r.per_sec_first(r.sw_bytes) // mib, r.per_sec_last(r.bytes) // mib))
+    if is_rwmix and r.rm_bytes:
+        rows.append(("MiB/s read", r.per_sec_first(r.rm_sw_bytes) // mib,
+                     r.per_sec_last(r.rm_bytes) // mib))
+        rows.append(("MiB/s total", r.per_sec_first(r.sw_bytes + r.rm_sw_bytes) // mib,
+                     r.per_sec_last(r.bytes + r.rm_bytes) // mib))
+
+    if r.bytes:
+        rows.append(("MiB write" if is_rwmix else "Total MiB",
+                     r.sw_bytes // mib, r.bytes // mib))
+    if is_rwmix and r.rm_bytes:
+        rows.append(("MiB read", r.rm_sw_bytes // mib, r.rm_bytes // mib))
 
     if r.entries:
         rows.append((f"{entry_type} total", r.sw_entries, r.entries))
@@ -216,6 +268,8 @@ def print_phase_results(cfg: BenchConfig, r: PhaseResults, out=None) -> None:
         _print_latency(cfg, "Ent lat us", r.entry_lat, out)
     if cfg.lat and r.io_lat.num_values:
         _print_latency(cfg, "IO lat us", r.io_lat, out)
+    if cfg.lat and r.io_lat_rm.num_values:
+        _print_latency(cfg, "IO lat rd us", r.io_lat_rm, out)
 
     if cfg.all_elapsed and r.worker_elapsed_usec:
         vals = " ".join(elapsed_ms_to_human(us // 1000) for us in r.worker_elapsed_usec)
@@ -328,6 +382,25 @@ def append_csv_result(cfg: BenchConfig, r: PhaseResults, path: str) -> None:
         "version": VERSION,
         "command": " ".join(sys.argv),
     }
+    if r.rm_bytes or r.rm_iops or r.rm_entries:
+        row.update({
+            "rwmix read entries/s [first]": r.per_sec_first(r.rm_sw_entries),
+            "rwmix read entries/s [last]": r.per_sec_last(r.rm_entries),
+            "rwmix read IOPS [first]": r.per_sec_first(r.rm_sw_iops),
+            "rwmix read IOPS [last]": r.per_sec_last(r.rm_iops),
+            "rwmix read MiB/s [first]": r.per_sec_first(r.rm_sw_bytes) // mib,
+            "rwmix read MiB/s [last]": r.per_sec_last(r.rm_bytes) // mib,
+            "rwmix read entries [first]": r.rm_sw_entries,
+            "rwmix read entries [last]": r.rm_entries,
+            "rwmix read MiB [first]": r.rm_sw_bytes // mib,
+            "rwmix read MiB [last]": r.rm_bytes // mib,
+            "rwmix read Ent lat us [min]": r.entry_lat_rm.min_us,
+            "rwmix read Ent lat us [avg]": int(r.entry_lat_rm.avg_us),
+            "rwmix read Ent lat us [max]": r.entry_lat_rm.max_us,
+            "rwmix read IO lat us [min]": r.io_lat_rm.min_us,
+            "rwmix read IO lat us [avg]": int(r.io_lat_rm.avg_us),
+            "rwmix read IO lat us [max]": r.io_lat_rm.max_us,
+        })
     for c in CSV_COLUMNS:
         row.setdefault(c, "")
     with open(path, "a", newline="") as f:
@@ -385,6 +458,17 @@ def phase_results_json(cfg: BenchConfig, r: PhaseResults) -> dict[str, Any]:
     }
     if cfg.label:
         doc["label"] = cfg.label
+    if r.rm_bytes or r.rm_iops:
+        doc["first_done_rwmix_read"] = {
+            "iops": r.per_sec_first(r.rm_sw_iops),
+            "mib_per_sec": r.per_sec_first(r.rm_sw_bytes) // mib,
+            "bytes": r.rm_sw_bytes,
+        }
+        doc["last_done_rwmix_read"] = {
+            "iops": r.per_sec_last(r.rm_iops),
+            "mib_per_sec": r.per_sec_last(r.rm_bytes) // mib,
+            "bytes": r.rm_bytes,
+        }
     if r.entry_lat.num_values:
         doc["entries_latency"] = _lat_json(r.entry_lat)
     if r.io_lat.num_values:

@@ -1,0 +1,171 @@
+"""Engine feature coverage: rwmix, mmap, flock, fadvise, verify-direct,
+readinline, statinline, opslog, core binding, infloop."""
+
+import json
+import os
+
+import pytest
+
+from tests.test_engine import run_phase
+
+
+def test_rwmix_pct_ratio(core, tmp_path):
+    p = str(tmp_path / "f")
+    size = 16 * 1024 * 1024
+    base = dict(paths=[p], path_type="file", threads=2, num_dataset_threads=2,
+                file_size=size, block_size=64 * 1024)
+    eng = core.Engine(base)
+    eng.prepare()
+    run_phase(core, eng, "WRITE")  # prefill so rwmix reads succeed
+
+    cfg = dict(base, rwmix_pct=30)
+    eng2 = core.Engine(cfg)
+    eng2.prepare()
+    res = run_phase(core, eng2, "WRITE")
+    wbytes = sum(r["bytes"] for r in res)
+    rbytes = sum(r["rm_bytes"] for r in res)
+    assert wbytes + rbytes == size
+    ratio = 100 * rbytes / (wbytes + rbytes)
+    assert 25 <= ratio <= 35  # 30% +- rounding
+
+
+def test_rwmix_dedicated_readers(core, tmp_path):
+    p = str(tmp_path / "f")
+    size = 8 * 1024 * 1024
+    base = dict(paths=[p], path_type="file", threads=4, num_dataset_threads=4,
+                file_size=size, block_size=64 * 1024)
+    eng = core.Engine(base)
+    eng.prepare()
+    run_phase(core, eng, "WRITE")
+
+    cfg = dict(base, rwmix_threads=2)
+    eng2 = core.Engine(cfg)
+    eng2.prepare()
+    res = run_phase(core, eng2, "WRITE")
+    # first 2 ranks only read, last 2 only write
+    for r in res:
+        if r["rank"] < 2:
+            assert r["rm_bytes"] > 0 and r["bytes"] == 0
+        else:
+            assert r["bytes"] > 0 and r["rm_bytes"] == 0
+
+
+def test_mmap_write_read_verify(core, tmp_path):
+    p = str(tmp_path / "f")
+    size = 4 * 1024 * 1024
+    cfg = dict(paths=[p], path_type="file", threads=2, num_dataset_threads=2,
+               file_size=size, block_size=256 * 1024, mmap=True, verify_salt=3)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    run_phase(core, eng, "WRITE")
+    assert os.path.getsize(p) == size
+    with open(p, "rb") as f:
+        assert core.verify_checksum(f.read(), 0, 3) == 2**64 - 1
+    run_phase(core, eng, "READ")
+
+
+def test_verify_direct_catches_nothing_on_good_fs(core, tmp_path):
+    p = str(tmp_path / "f")
+    cfg = dict(paths=[p], path_type="file", threads=1, num_dataset_threads=1,
+               file_size=1 << 20, block_size=64 * 1024, verify_direct=True)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    res = run_phase(core, eng, "WRITE")
+    assert sum(r["bytes"] for r in res) == 1 << 20
+
+
+def test_flock_and_fadvise(core, tmp_path):
+    p = str(tmp_path / "f")
+    cfg = dict(paths=[p], path_type="file", threads=2, num_dataset_threads=2,
+               file_size=1 << 20, block_size=64 * 1024,
+               flock_mode=1, fadv_flags=1 | 4)  # range lock + seq/willneed
+    eng = core.Engine(cfg)
+    eng.prepare()
+    run_phase(core, eng, "WRITE")
+    run_phase(core, eng, "READ")
+
+
+def test_dir_mode_readinline_statinline(core, tmp_path):
+    cfg = dict(paths=[str(tmp_path)], path_type="dir", threads=2, num_dataset_threads=2,
+               dirs=1, files=3, file_size=128 * 1024, block_size=64 * 1024,
+               read_inline=True, stat_inline=True, verify_salt=2)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    run_phase(core, eng, "MKDIRS")
+    res = run_phase(core, eng, "WRITE")
+    total = 2 * 1 * 3 * 128 * 1024
+    assert sum(r["bytes"] for r in res) == total
+    # inline readback accounted as rwmix reads
+    assert sum(r["rm_bytes"] for r in res) == total
+
+
+def test_dir_sharing(core, tmp_path):
+    cfg = dict(paths=[str(tmp_path)], path_type="dir", threads=2, num_dataset_threads=2,
+               dirs=2, files=2, file_size=4096, block_size=4096, dir_sharing=True)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    run_phase(core, eng, "MKDIRS")
+    run_phase(core, eng, "WRITE")
+    # all files live under rank 0's dirs; names keep per-rank uniqueness
+    assert (tmp_path / "r0" / "d0" / "r0-f0").exists()
+    assert (tmp_path / "r0" / "d0" / "r1-f0").exists()
+    assert not (tmp_path / "r1").exists() or not any((tmp_path / "r1").iterdir())
+
+
+def test_opslog(core, tmp_path):
+    p = str(tmp_path / "f")
+    log = str(tmp_path / "ops.jsonl")
+    cfg = dict(paths=[p], path_type="file", threads=1, num_dataset_threads=1,
+               file_size=256 * 1024, block_size=64 * 1024, ops_log=log)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    run_phase(core, eng, "WRITE")
+    lines = [json.loads(ln) for ln in open(log)]
+    assert len(lines) == 2 * (256 // 64)  # pre+post per block
+    assert lines[0]["op"] == "pwrite" and lines[0]["type"] == "pre"
+    assert lines[1]["type"] == "post"
+    assert lines[0]["entry"] == p
+
+
+def test_core_binding(core, tmp_path):
+    p = str(tmp_path / "f")
+    cfg = dict(paths=[p], path_type="file", threads=2, num_dataset_threads=2,
+               file_size=256 * 1024, block_size=64 * 1024, cores=[0])
+    eng = core.Engine(cfg)
+    eng.prepare()
+    res = run_phase(core, eng, "WRITE")
+    assert sum(r["bytes"] for r in res) == 256 * 1024
+
+
+def test_infloop_runs_until_interrupt(core, tmp_path):
+    p = str(tmp_path / "f")
+    size = 256 * 1024
+    cfg = dict(paths=[p], path_type="file", threads=1, num_dataset_threads=1,
+               file_size=size, block_size=64 * 1024, inf_loop=True)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    eng.start_phase(core.PHASES["WRITE"])
+    assert not eng.wait_phase_done(300)  # still looping
+    eng.interrupt()
+    assert eng.wait_phase_done(10_000)
+    res = eng.finish_phase()
+    # multiple passes of the file happened
+    assert res[0]["bytes"] > size
+
+
+def test_rwmix_csv_and_console(tmp_path, capsys):
+    from elbencho_amd.cli import main
+    from elbencho_amd.stats import CSV_COLUMNS
+    import csv as csvmod
+
+    f = tmp_path / "f"
+    csvf = tmp_path / "res.csv"
+    assert main(["-w", "-t", "1", "-b", "64k", "-s", "4m", "--nolive", str(f)]) == 0
+    rc = main(["-w", "-t", "1", "-b", "64k", "-s", "4m", "--rwmixpct", "50",
+               "--nolive", "--csvfile", str(csvf), str(f)])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "MiB/s read" in out and "MiB/s write" in out
+    with open(csvf, newline="") as fh:
+        rows = list(csvmod.DictReader(fh))
+    assert int(rows[0]["rwmix read MiB [last]"]) >= 1
