@@ -217,6 +217,8 @@ class BenchConfig:
 
     # internal derived values (not user options)
     num_dataset_threads: int = 0
+    tree_dirs_resolved: list = field(default_factory=list)
+    tree_files_resolved: list = field(default_factory=list)  # [(relpath, size)]
 
     # ------------------------------------------------------------------
     def finalize(self) -> None:
@@ -237,6 +239,16 @@ class BenchConfig:
 
         if self.bench_mode == "posix" and self.paths:
             self.path_type = self._infer_path_type()
+
+        if self.treefile and not self.tree_scan:
+            from elbencho_amd.pathstore import parse_treefile
+            tree = parse_treefile(self.treefile)
+            if self.tree_round_up:
+                tree.round_up(self.tree_round_up)
+            self.tree_dirs_resolved = tree.dirs
+            self.tree_files_resolved = tree.files
+            if self.path_type != PATH_DIR and self.paths:
+                raise ConfigError("--treefile requires directory bench paths")
 
         # derived: total dataset threads across hosts sharing the dataset
         num_hosts = len(self.hosts) if self.hosts else 1
@@ -392,6 +404,9 @@ class BenchConfig:
             ops_log_lock=self.ops_log_lock,
             cores=_parse_int_list(self.cpu_cores),
             zones=_parse_int_list(self.numa_zones),
+            tree_dirs=self.tree_dirs_resolved,
+            tree_files=self.tree_files_resolved,
+            sharesize=self.sharesize,
             limit_read_bps=self.limit_read,
             limit_write_bps=self.limit_write,
             ignore_del_errors=self.ignore_del_errors,

@@ -83,6 +83,18 @@ class Coordinator:
             from elbencho_amd.service import run_service
             return run_service(cfg)
 
+        if cfg.tree_scan:
+            from elbencho_amd.pathstore import scan_path, write_treefile
+            if not cfg.treefile:
+                print("ERROR: --treescan requires --treefile as output path",
+                      file=sys.stderr)
+                return 1
+            tree = scan_path(cfg.tree_scan)
+            write_treefile(tree, cfg.treefile)
+            print(f"Scanned {cfg.tree_scan}: {len(tree.dirs)} dirs, "
+                  f"{len(tree.files)} files -> {cfg.treefile}", file=self.out)
+            return 0
+
         try:
             signal.signal(signal.SIGINT, self._on_sigint)
         except ValueError:

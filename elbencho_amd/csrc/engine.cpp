@@ -1034,6 +1034,166 @@ void Worker::dirModeFiles(Phase phase)
 }
 
 // ---------------------------------------------------------------------------
+// custom tree mode (reference analogue: PathStore::getWorkerSublist* +
+// LocalWorker custom-tree iterate, LocalWorker.cpp:2960/:3294)
+// ---------------------------------------------------------------------------
+
+// mkdir -p: create every missing parent of a relative path under base.
+static void mkdirBottomUp(const std::string& base, const std::string& rel)
+{
+    std::string cur = base;
+    size_t pos = 0;
+    while (pos != std::string::npos) {
+        size_t next = rel.find('/', pos);
+        std::string part = rel.substr(pos, next == std::string::npos ? next : next - pos);
+        if (!part.empty()) {
+            cur += "/" + part;
+            if (mkdir(cur.c_str(), 0777) && errno != EEXIST) throwErrno("mkdir", cur);
+        }
+        pos = (next == std::string::npos) ? next : next + 1;
+    }
+}
+
+void Worker::customTreeDirs(Phase phase)
+{
+    const auto& cfg = eng.cfg;
+    const bool lat = cfg.measureLat;
+    const std::string& base = cfg.paths[globalRank % cfg.paths.size()];
+    const uint64_t numRanks = cfg.numDataSetThreads;
+
+    if (phase == Phase::MKDIRS) {
+        for (size_t i = 0; i < cfg.treeDirs.size(); i++) {
+            if ((int)(i % numRanks) != globalRank) continue;
+            if ((i % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+            auto t0 = lat ? Clock::now() : Clock::time_point();
+            mkdirBottomUp(base, cfg.treeDirs[i]);
+            if (lat)
+                addEntryLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
+                    Clock::now() - t0).count());
+            liveOps.entries.fetch_add(1, std::memory_order_relaxed);
+        }
+    } else { // RMDIRS: deepest dirs first; peers may still hold parents, so
+             // tolerate ENOTEMPTY/ENOENT (reference behavior for shared trees)
+        std::vector<size_t> mine;
+        for (size_t i = 0; i < cfg.treeDirs.size(); i++)
+            if ((int)(i % numRanks) == globalRank) mine.push_back(i);
+        std::sort(mine.begin(), mine.end(), [&](size_t a, size_t b) {
+            return std::count(cfg.treeDirs[a].begin(), cfg.treeDirs[a].end(), '/') >
+                   std::count(cfg.treeDirs[b].begin(), cfg.treeDirs[b].end(), '/');
+        });
+        for (size_t i : mine) {
+            checkInterrupt();
+            std::string full = base + "/" + cfg.treeDirs[i];
+            auto t0 = lat ? Clock::now() : Clock::time_point();
+            if (rmdir(full.c_str()) && errno != ENOENT && errno != ENOTEMPTY) {
+                if (!cfg.ignoreDelErrors) throwErrno("rmdir", full);
+            }
+            if (lat)
+                addEntryLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
+                    Clock::now() - t0).count());
+            liveOps.entries.fetch_add(1, std::memory_order_relaxed);
+        }
+    }
+}
+
+void Worker::customTreeFiles(Phase phase)
+{
+    const auto& cfg = eng.cfg;
+    const bool lat = cfg.measureLat;
+    const bool isWrite = (phase == Phase::WRITE);
+    const bool isRead = (phase == Phase::READ);
+    const std::string& base = cfg.paths[globalRank % cfg.paths.size()];
+    const uint64_t numRanks = cfg.numDataSetThreads;
+
+    int openFlags = 0;
+    if (isWrite) openFlags = O_CREAT | O_WRONLY;
+    if (isRead) openFlags = O_RDONLY;
+    if (cfg.directIO) openFlags |= O_DIRECT;
+
+    std::unique_ptr<OffsetGen> gen;
+    size_t nonSharedIdx = 0; // running index over the non-shared sublist
+
+    for (size_t i = 0; i < cfg.treeFiles.size(); i++) {
+        const auto& [rel, size] = cfg.treeFiles[i];
+        const bool shared = cfg.shareSize && size >= cfg.shareSize;
+
+        uint64_t rangeStart = 0, rangeLen = size;
+        if (!shared) {
+            // whole files round-robin across ranks
+            bool mine = (int)(nonSharedIdx % numRanks) == globalRank;
+            nonSharedIdx++;
+            if (!mine) continue;
+        } else if (phase == Phase::WRITE || phase == Phase::READ) {
+            // blockwise range slice of each shared file per rank
+            fairShareSlice(size, rangeStart, rangeLen);
+            if (!rangeLen) continue;
+        } else {
+            // stat/unlink of shared files: one rank per file
+            if ((int)(i % numRanks) != globalRank) continue;
+        }
+
+        if ((i % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+        std::string full = base + "/" + rel;
+        auto tEntry0 = lat ? Clock::now() : Clock::time_point();
+
+        switch (phase) {
+            case Phase::WRITE:
+            case Phase::READ: {
+                int fd = open(full.c_str(), openFlags, 0644);
+                if (fd < 0 && isWrite && errno == ENOENT) {
+                    // parent dirs may be missing when no MKDIRS phase ran
+                    size_t slash = rel.rfind('/');
+                    if (slash != std::string::npos) mkdirBottomUp(base, rel.substr(0, slash));
+                    fd = open(full.c_str(), openFlags, 0644);
+                }
+                if (fd < 0) throwErrno("open", full);
+                try {
+                    if (!gen) gen = makeOffsetGen(rangeStart, rangeLen);
+                    else gen->reset(rangeStart, rangeLen);
+                    BlockSpec spec;
+                    while (gen->next(spec)) {
+                        uint64_t ioLen = std::min(spec.len, size - spec.offset);
+                        auto t0 = lat ? Clock::now() : Clock::time_point();
+                        ssize_t res = blockIO(isWrite, fd, 0, ioLen, spec.offset,
+                                              nullptr, &full);
+                        if (res != (ssize_t)ioLen)
+                            throwErrno(isWrite ? "write" : "read", full);
+                        if (lat)
+                            addIoLat((uint64_t)std::chrono::duration_cast<
+                                std::chrono::microseconds>(Clock::now() - t0).count());
+                        liveOps.bytes.fetch_add(ioLen, std::memory_order_relaxed);
+                        liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+                    }
+                } catch (...) {
+                    close(fd);
+                    throw;
+                }
+                close(fd);
+                break;
+            }
+            case Phase::STAT: {
+                struct stat st;
+                if (stat(full.c_str(), &st)) throwErrno("stat", full);
+                break;
+            }
+            case Phase::RMFILES: {
+                if (unlink(full.c_str()) && errno != ENOENT) {
+                    if (!cfg.ignoreDelErrors) throwErrno("unlink", full);
+                }
+                break;
+            }
+            default:
+                throw WorkerError("bad custom tree phase");
+        }
+
+        if (lat)
+            addEntryLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
+                Clock::now() - tEntry0).count());
+        liveOps.entries.fetch_add(1, std::memory_order_relaxed);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // sync / dropcaches
 // ---------------------------------------------------------------------------
 
@@ -1063,33 +1223,49 @@ void Worker::runPhase()
 {
     const auto& cfg = eng.cfg;
 
+    const bool customTree = !cfg.treeFiles.empty() || !cfg.treeDirs.empty();
+
     switch (eng.currentPhase) {
         case Phase::MKDIRS:
-            dirModeMkdirs();
+            if (customTree)
+                customTreeDirs(Phase::MKDIRS);
+            else
+                dirModeMkdirs();
             break;
         case Phase::RMDIRS:
-            dirModeRmdirs();
+            if (customTree)
+                customTreeDirs(Phase::RMDIRS);
+            else
+                dirModeRmdirs();
             break;
         case Phase::WRITE:
-            if (cfg.pathType == PathType::DIR)
+            if (customTree)
+                customTreeFiles(Phase::WRITE);
+            else if (cfg.pathType == PathType::DIR)
                 dirModeFiles(Phase::WRITE);
             else
                 fileModeBlocks(true);
             break;
         case Phase::READ:
-            if (cfg.pathType == PathType::DIR)
+            if (customTree)
+                customTreeFiles(Phase::READ);
+            else if (cfg.pathType == PathType::DIR)
                 dirModeFiles(Phase::READ);
             else
                 fileModeBlocks(false);
             break;
         case Phase::STAT:
-            if (cfg.pathType == PathType::DIR)
+            if (customTree)
+                customTreeFiles(Phase::STAT);
+            else if (cfg.pathType == PathType::DIR)
                 dirModeFiles(Phase::STAT);
             else
                 fileModeStat();
             break;
         case Phase::RMFILES:
-            if (cfg.pathType == PathType::DIR)
+            if (customTree)
+                customTreeFiles(Phase::RMFILES);
+            else if (cfg.pathType == PathType::DIR)
                 dirModeFiles(Phase::RMFILES);
             else
                 fileModeDelete();

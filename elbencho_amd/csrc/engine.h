@@ -78,6 +78,13 @@ struct EngineConfig {
     std::vector<int> cpuCores;  // round-robin thread->core binding
     std::vector<int> numaZones; // round-robin thread->NUMA-zone binding
 
+    // custom tree mode (reference PathStore / --treefile):
+    // dirs + (relpath, size) files under paths[0]; files >= shareSize are
+    // range-sliced across ranks, smaller ones distributed round-robin.
+    std::vector<std::string> treeDirs;
+    std::vector<std::pair<std::string, uint64_t>> treeFiles;
+    uint64_t shareSize = 0;
+
     std::vector<int> gpuIDs; // empty = CPU buffers only
     bool gpuPinnedHostBufs = true;
 
@@ -152,6 +159,8 @@ private:
     void dirModeMkdirs();
     void dirModeRmdirs();
     void dirModeFiles(Phase phase);
+    void customTreeDirs(Phase phase);
+    void customTreeFiles(Phase phase);
     void anyModeSync();
     void anyModeDropCaches();
 

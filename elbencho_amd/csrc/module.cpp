@@ -68,6 +68,11 @@ EngineConfig configFromDict(const py::dict& d)
     c.opsLogLock = getB("ops_log_lock", false);
     if (d.contains("cores")) c.cpuCores = d["cores"].cast<std::vector<int>>();
     if (d.contains("zones")) c.numaZones = d["zones"].cast<std::vector<int>>();
+    if (d.contains("tree_dirs"))
+        c.treeDirs = d["tree_dirs"].cast<std::vector<std::string>>();
+    if (d.contains("tree_files"))
+        c.treeFiles = d["tree_files"].cast<std::vector<std::pair<std::string, uint64_t>>>();
+    c.shareSize = getU64("sharesize", 0);
     c.blockVarPct = (int)getI("blockvar_pct", 100);
     c.blockVarAlgo = getS("blockvar_algo", "fast");
     c.randAlgo = getS("rand_algo", "balanced_single");

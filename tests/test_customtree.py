@@ -1,0 +1,98 @@
+"""Custom tree mode: treefile parse/scan, shared/non-shared partitioning."""
+
+import os
+
+from elbencho_amd.cli import main
+from elbencho_amd.pathstore import CustomTree, parse_treefile, scan_path, write_treefile
+
+from tests.test_engine import run_phase
+
+
+def test_treefile_roundtrip(tmp_path):
+    tree = CustomTree(
+        dirs=["a", "a/b", "c"],
+        files=[("a/f1", 1024), ("a/b/f2", 4096), ("top", 0)],
+    )
+    tf = tmp_path / "tree.txt"
+    write_treefile(tree, str(tf))
+    tree2 = parse_treefile(str(tf))
+    assert tree2.dirs == tree.dirs
+    assert tree2.files == tree.files
+
+
+def test_treefile_roundup():
+    tree = CustomTree(files=[("f", 1000), ("g", 4096), ("h", 0)])
+    tree.round_up(4096)
+    assert tree.files == [("f", 4096), ("g", 4096), ("h", 0)]
+
+
+def test_split_share():
+    tree = CustomTree(files=[("small", 100), ("big", 10_000)])
+    ns, sh = tree.split_share(1000)
+    assert ns == [("small", 100)]
+    assert sh == [("big", 10_000)]
+    ns, sh = tree.split_share(0)
+    assert len(ns) == 2 and sh == []
+
+
+def test_custom_tree_engine_lifecycle(core, tmp_path):
+    base = tmp_path / "bench"
+    base.mkdir()
+    dirs = ["d1", "d1/sub", "d2"]
+    files = [("d1/f1", 64 * 1024), ("d1/sub/f2", 128 * 1024), ("d2/f3", 64 * 1024),
+             ("big", 1024 * 1024)]
+    cfg = dict(paths=[str(base)], path_type="dir", threads=2, num_dataset_threads=2,
+               block_size=64 * 1024, tree_dirs=dirs, tree_files=files,
+               sharesize=512 * 1024, verify_salt=4, lat=True)
+    eng = core.Engine(cfg)
+    eng.prepare()
+
+    res = run_phase(core, eng, "MKDIRS")
+    assert sum(r["entries"] for r in res) == 3
+    for d in dirs:
+        assert (base / d).is_dir()
+
+    res = run_phase(core, eng, "WRITE")
+    total = sum(s for _, s in files)
+    assert sum(r["bytes"] for r in res) == total
+    for p, s in files:
+        assert os.path.getsize(base / p) == s
+        with open(base / p, "rb") as f:
+            assert core.verify_checksum(f.read(), 0, 4) == 2**64 - 1
+    # shared big file was written by both ranks (range-sliced)
+    by_rank = {r["rank"]: r["bytes"] for r in res}
+    assert all(b > 0 for b in by_rank.values())
+
+    run_phase(core, eng, "STAT")
+    run_phase(core, eng, "READ")
+
+    res = run_phase(core, eng, "RMFILES")
+    assert sum(r["entries"] for r in res) == len(files)
+    res = run_phase(core, eng, "RMDIRS")
+    assert not any((base / d).exists() for d in dirs)
+
+
+def test_treescan_cli(tmp_path, capsys):
+    src = tmp_path / "src"
+    (src / "x" / "y").mkdir(parents=True)
+    (src / "x" / "a.bin").write_bytes(b"\0" * 500)
+    (src / "x" / "y" / "b.bin").write_bytes(b"\0" * 100)
+    tf = tmp_path / "out.tree"
+    rc = main(["--treescan", str(src), "--treefile", str(tf), "--nolive"])
+    assert rc == 0
+    tree = parse_treefile(str(tf))
+    assert tree.dirs == ["x", "x/y"]
+    assert ("x/a.bin", 500) in tree.files
+    assert ("x/y/b.bin", 100) in tree.files
+
+
+def test_treefile_cli_run(tmp_path):
+    base = tmp_path / "bench"
+    base.mkdir()
+    tf = tmp_path / "tree.txt"
+    tree = CustomTree(dirs=["d"], files=[("d/f1", 65536), ("d/f2", 65536)])
+    write_treefile(tree, str(tf))
+    rc = main(["-d", "-w", "-r", "-F", "-D", "-t", "2", "-b", "64k", "--nolive",
+               "--treefile", str(tf), str(base)])
+    assert rc == 0
+    assert not (base / "d").exists()
