@@ -217,6 +217,8 @@ class BenchConfig:
 
     # internal derived values (not user options)
     num_dataset_threads: int = 0
+    netbench_is_server: bool = False   # set per host by the master
+    netbench_num_conns: int = 0        # server: expected client connections
     tree_dirs_resolved: list = field(default_factory=list)
     tree_files_resolved: list = field(default_factory=list)  # [(relpath, size)]
 
@@ -291,6 +293,11 @@ class BenchConfig:
                            self.run_delfiles, self.run_deldirs])
         if needs_paths and not self.paths and self.bench_mode != "netbench":
             raise ConfigError("benchmark paths are required")
+        if self.bench_mode == "netbench":
+            if not self.hosts:
+                raise ConfigError("--netbench requires service mode (--hosts)")
+            if not self.servers:
+                raise ConfigError("--netbench requires --servers")
 
         if self.threads < 1:
             raise ConfigError("number of threads must be >= 1")
@@ -333,6 +340,8 @@ class BenchConfig:
     def phase_list(self) -> list[str]:
         """Ordered phase names for one iteration (reference order,
         Coordinator.cpp:311-334)."""
+        if self.bench_mode == "netbench":
+            return ["NETBENCH"] if self.run_write else []
         order = [
             ("MKDIRS", self.run_mkdirs),
             ("WRITE", self.run_write),
@@ -404,6 +413,13 @@ class BenchConfig:
             ops_log_lock=self.ops_log_lock,
             cores=_parse_int_list(self.cpu_cores),
             zones=_parse_int_list(self.numa_zones),
+            netbench_is_server=self.netbench_is_server,
+            netbench_servers=self.servers,
+            netbench_port=self.service_port + 1000,
+            netbench_num_conns=self.netbench_num_conns,
+            resp_size=self.resp_size,
+            send_buf=self.send_buf,
+            recv_buf=self.recv_buf,
             tree_dirs=self.tree_dirs_resolved,
             tree_files=self.tree_files_resolved,
             sharesize=self.sharesize,

@@ -24,6 +24,7 @@
 #include <cstdio>
 #include <cstring>
 
+#include "netbench.h"
 #include "uring.h"
 
 namespace eb {
@@ -1072,8 +1073,8 @@ void Worker::customTreeDirs(Phase phase)
                     Clock::now() - t0).count());
             liveOps.entries.fetch_add(1, std::memory_order_relaxed);
         }
-    } else { // RMDIRS: deepest dirs first; peers may still hold parents, so
-             // tolerate ENOTEMPTY/ENOENT (reference behavior for shared trees)
+    } else { // RMDIRS: deepest dirs first; a dir whose child belongs to a
+             // peer rank may still be non-empty, so retry until it drains
         std::vector<size_t> mine;
         for (size_t i = 0; i < cfg.treeDirs.size(); i++)
             if ((int)(i % numRanks) == globalRank) mine.push_back(i);
@@ -1081,18 +1082,32 @@ void Worker::customTreeDirs(Phase phase)
             return std::count(cfg.treeDirs[a].begin(), cfg.treeDirs[a].end(), '/') >
                    std::count(cfg.treeDirs[b].begin(), cfg.treeDirs[b].end(), '/');
         });
-        for (size_t i : mine) {
+
+        std::vector<size_t> pending = mine;
+        for (int attempt = 0; !pending.empty() && attempt < 5000; attempt++) {
             checkInterrupt();
-            std::string full = base + "/" + cfg.treeDirs[i];
-            auto t0 = lat ? Clock::now() : Clock::time_point();
-            if (rmdir(full.c_str()) && errno != ENOENT && errno != ENOTEMPTY) {
-                if (!cfg.ignoreDelErrors) throwErrno("rmdir", full);
+            std::vector<size_t> still;
+            for (size_t i : pending) {
+                std::string full = base + "/" + cfg.treeDirs[i];
+                auto t0 = lat ? Clock::now() : Clock::time_point();
+                int rc = rmdir(full.c_str());
+                if (rc && errno == ENOTEMPTY) { // peer's child not gone yet
+                    still.push_back(i);
+                    continue;
+                }
+                if (rc && errno != ENOENT && !cfg.ignoreDelErrors) throwErrno("rmdir", full);
+                if (lat)
+                    addEntryLat((uint64_t)std::chrono::duration_cast<
+                        std::chrono::microseconds>(Clock::now() - t0).count());
+                liveOps.entries.fetch_add(1, std::memory_order_relaxed);
             }
-            if (lat)
-                addEntryLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
-                    Clock::now() - t0).count());
-            liveOps.entries.fetch_add(1, std::memory_order_relaxed);
+            if (still.size() == pending.size())
+                std::this_thread::sleep_for(std::chrono::milliseconds(1));
+            pending.swap(still);
         }
+        if (!pending.empty() && !cfg.ignoreDelErrors)
+            throw WorkerError("rmdir: directories stayed non-empty: " +
+                              cfg.treeDirs[pending[0]]);
     }
 }
 
@@ -1194,6 +1209,140 @@ void Worker::customTreeFiles(Phase phase)
 }
 
 // ---------------------------------------------------------------------------
+// netbench (reference LocalWorker.cpp:7789-8064)
+// ---------------------------------------------------------------------------
+
+void Worker::netbenchServer()
+{
+    const auto& cfg = eng.cfg;
+
+    // worker 0 accepts all expected client connections, then distributes
+    // them round-robin (reference: first worker of each server host accepts,
+    // LocalWorker.cpp:646-728)
+    if (localRank == 0) {
+        int listenFd = netListen(cfg.netbenchPort, 128);
+        try {
+            int expected = cfg.netbenchNumConns > 0 ? cfg.netbenchNumConns : 1;
+            std::vector<int> conns;
+            for (int i = 0; i < expected; i++) {
+                struct pollfd pfd = {listenFd, POLLIN, 0};
+                for (;;) {
+                    checkInterrupt();
+                    int pr = poll(&pfd, 1, 250);
+                    if (pr > 0) break;
+                }
+                int cfd = accept(listenFd, nullptr, nullptr);
+                if (cfd < 0) throwErrno("accept", "netbench");
+                setSockBufs(cfd, cfg.sendBufSize, cfg.recvBufSize);
+                conns.push_back(cfd);
+            }
+            {
+                std::lock_guard<std::mutex> lk(eng.nbMtx);
+                eng.nbConns = std::move(conns);
+                eng.nbAcceptDone = true;
+            }
+            eng.nbCv.notify_all();
+        } catch (...) {
+            close(listenFd);
+            {
+                std::lock_guard<std::mutex> lk(eng.nbMtx);
+                eng.nbAcceptDone = true; // release waiting peers
+            }
+            eng.nbCv.notify_all();
+            throw;
+        }
+        close(listenFd);
+    } else {
+        std::unique_lock<std::mutex> lk(eng.nbMtx);
+        eng.nbCv.wait(lk, [&] { return eng.nbAcceptDone; });
+    }
+
+    // take my round-robin subset of the connections
+    std::vector<int> mine;
+    {
+        std::lock_guard<std::mutex> lk(eng.nbMtx);
+        for (size_t i = 0; i < eng.nbConns.size(); i++)
+            if ((int)(i % cfg.numThreads) == localRank) mine.push_back(eng.nbConns[i]);
+    }
+    if (mine.empty()) return;
+
+    const uint64_t bs = cfg.blockSize;
+    std::vector<char> block(bs);
+    std::vector<char> resp(cfg.respSize, 'R');
+    std::vector<struct pollfd> pfds;
+    for (int fd : mine) pfds.push_back({fd, POLLIN, 0});
+
+    size_t openConns = mine.size();
+    while (openConns) {
+        checkInterrupt();
+        int pr = poll(pfds.data(), pfds.size(), 250);
+        if (pr <= 0) continue;
+        for (auto& pfd : pfds) {
+            if (!(pfd.revents & (POLLIN | POLLHUP | POLLERR)) || pfd.fd < 0) continue;
+            if (!recvExact(pfd.fd, block.data(), bs)) { // EOF: client finished
+                close(pfd.fd);
+                pfd.fd = -1;
+                openConns--;
+                continue;
+            }
+            if (!sendExact(pfd.fd, resp.data(), cfg.respSize))
+                throw WorkerError("netbench: response send failed");
+            // transfer stats are accounted on the CLIENT side only, so the
+            // master's aggregate equals the payload bytes sent once
+        }
+    }
+}
+
+void Worker::netbenchClient()
+{
+    const auto& cfg = eng.cfg;
+    if (cfg.netbenchServers.empty())
+        throw WorkerError("netbench client without --servers");
+
+    const bool lat = cfg.measureLat;
+    const uint64_t bs = cfg.blockSize;
+    const uint64_t total = cfg.fileSize; // bytes per client thread
+
+    // round-robin client->server assignment (reference LocalWorker.cpp:735)
+    const std::string& srv = cfg.netbenchServers[globalRank % cfg.netbenchServers.size()];
+    std::string host = srv;
+    int port = cfg.netbenchPort;
+    if (auto pos = srv.rfind(':'); pos != std::string::npos) {
+        host = srv.substr(0, pos);
+        port = std::stoi(srv.substr(pos + 1)) + 1000; // service port + 1000
+    }
+
+    int fd = netConnect(host, port, 0, 30, eng.interruptFlag);
+    setSockBufs(fd, cfg.sendBufSize, cfg.recvBufSize);
+
+    std::vector<char> resp(cfg.respSize);
+    uint64_t sent = 0;
+    uint64_t opCount = 0;
+
+    try {
+        while (sent < total) {
+            if ((opCount++ % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+            rateLimiter.wait(bs);
+            auto t0 = lat ? Clock::now() : Clock::time_point();
+            if (!sendExact(fd, hostBufs[0], bs))
+                throw WorkerError("netbench: block send failed");
+            if (!recvExact(fd, resp.data(), cfg.respSize))
+                throw WorkerError("netbench: response recv failed");
+            if (lat)
+                addIoLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(
+                    Clock::now() - t0).count());
+            sent += bs;
+            liveOps.bytes.fetch_add(bs, std::memory_order_relaxed);
+            liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+        }
+    } catch (...) {
+        close(fd);
+        throw;
+    }
+    close(fd);
+}
+
+// ---------------------------------------------------------------------------
 // sync / dropcaches
 // ---------------------------------------------------------------------------
 
@@ -1270,6 +1419,12 @@ void Worker::runPhase()
             else
                 fileModeDelete();
             break;
+        case Phase::NETBENCH:
+            if (cfg.netbenchIsServer)
+                netbenchServer();
+            else
+                netbenchClient();
+            break;
         case Phase::SYNC:
             anyModeSync();
             break;
@@ -1301,7 +1456,8 @@ void Worker::threadMain()
             (eng.currentPhase == Phase::WRITE) && (localRank < cfg.rwMixThreads);
 
         // buffers only needed for data phases
-        if (eng.currentPhase == Phase::WRITE || eng.currentPhase == Phase::READ)
+        if (eng.currentPhase == Phase::WRITE || eng.currentPhase == Phase::READ ||
+            eng.currentPhase == Phase::NETBENCH)
             allocBuffers();
 
         rateLimiter.init((eng.currentPhase == Phase::WRITE && !isDedicatedReader)
@@ -1389,6 +1545,8 @@ void Engine::startPhase(Phase phase)
     if (phaseRunning) throw std::runtime_error("phase already running");
 
     currentPhase = phase;
+    nbAcceptDone = false;
+    nbConns.clear();
     if (!cfg.opsLogPath.empty() && !opsLog.isEnabled())
         opsLog.open(cfg.opsLogPath, cfg.opsLogLock);
     interruptFlag.store(false);

@@ -78,6 +78,14 @@ struct EngineConfig {
     std::vector<int> cpuCores;  // round-robin thread->core binding
     std::vector<int> numaZones; // round-robin thread->NUMA-zone binding
 
+    // netbench (service-only TCP request/response benchmark)
+    bool netbenchIsServer = false;
+    std::vector<std::string> netbenchServers; // "host" or "host:port"
+    int netbenchPort = 2611;   // service port + 1000 convention
+    int netbenchNumConns = 0;  // server: total client connections to expect
+    uint64_t respSize = 1;
+    int sendBufSize = 0, recvBufSize = 0;
+
     // custom tree mode (reference PathStore / --treefile):
     // dirs + (relpath, size) files under paths[0]; files >= shareSize are
     // range-sliced across ranks, smaller ones distributed round-robin.
@@ -161,6 +169,8 @@ private:
     void dirModeFiles(Phase phase);
     void customTreeDirs(Phase phase);
     void customTreeFiles(Phase phase);
+    void netbenchServer();
+    void netbenchClient();
     void anyModeSync();
     void anyModeDropCaches();
 
@@ -261,6 +271,12 @@ public:
     std::vector<std::unique_ptr<GpuCtx>> gpuCtxCache;
 
     OpsLogger opsLog;
+
+    // netbench server: worker 0 accepts all connections, peers take a subset
+    std::mutex nbMtx;
+    std::condition_variable nbCv;
+    std::vector<int> nbConns;
+    bool nbAcceptDone = false;
 
     void onWorkerDone(Worker& w, bool hadError);
 

@@ -73,6 +73,14 @@ EngineConfig configFromDict(const py::dict& d)
     if (d.contains("tree_files"))
         c.treeFiles = d["tree_files"].cast<std::vector<std::pair<std::string, uint64_t>>>();
     c.shareSize = getU64("sharesize", 0);
+    c.netbenchIsServer = getB("netbench_is_server", false);
+    if (d.contains("netbench_servers"))
+        c.netbenchServers = d["netbench_servers"].cast<std::vector<std::string>>();
+    c.netbenchPort = (int)getI("netbench_port", 2611);
+    c.netbenchNumConns = (int)getI("netbench_num_conns", 0);
+    c.respSize = getU64("resp_size", 1);
+    c.sendBufSize = (int)getI("send_buf", 0);
+    c.recvBufSize = (int)getI("recv_buf", 0);
     c.blockVarPct = (int)getI("blockvar_pct", 100);
     c.blockVarAlgo = getS("blockvar_algo", "fast");
     c.randAlgo = getS("rand_algo", "balanced_single");

@@ -163,11 +163,54 @@ class RemoteRunner:
         shared = not cfg.no_svc_share and cfg.path_type != "dir"
         num_hosts = len(self.hosts)
 
+        # netbench role assignment: hosts named in --servers act as servers,
+        # the rest are clients (reference scale-out semantics)
+        server_hosts = set()
+        if cfg.bench_mode == "netbench":
+            # exact host:port match when the server spec names a port;
+            # hostname-only specs match any port of that host
+            with_port = {s for s in cfg.servers if ":" in s}
+            name_only = {s for s in cfg.servers if ":" not in s}
+            for hs in self.hosts:
+                hp = hs.client.hostport
+                if hp in with_port or hp.rsplit(":", 1)[0] in name_only:
+                    server_hosts.add(hp)
+            if not server_hosts:
+                raise RuntimeError("--netbench: none of --hosts matches --servers")
+            n_clients = len(self.hosts) - len(server_hosts)
+            if n_clients < 1:
+                raise RuntimeError("--netbench needs at least one non-server host")
+            self._nb_total_client_threads = n_clients * cfg.threads
+            self._nb_num_servers = len(server_hosts)
+
+        client_rank = [0]  # contiguous rank offsets over client hosts only
+
         def prep(i: int, hs: HostState):
             wire = cfg.to_wire()
             wire["threads"] = cfg.threads
-            wire["rank_offset"] = i * cfg.threads
             wire["num_dataset_threads"] = (cfg.threads * num_hosts) if shared else cfg.threads
+            if cfg.bench_mode == "netbench":
+                is_server = hs.client.hostport in server_hosts
+                wire["netbench_is_server"] = is_server
+                svc_port = hs.client.port
+                wire["service_port"] = svc_port  # server listens on port+1000
+                if is_server:
+                    # this server's share of client connections (clients pick
+                    # servers round-robin by global rank)
+                    srv_list = sorted(server_hosts)
+                    j = srv_list.index(hs.client.hostport)
+                    n = self._nb_total_client_threads
+                    k = self._nb_num_servers
+                    wire["netbench_num_conns"] = n // k + (1 if (n % k) > j else 0)
+                    wire["rank_offset"] = 0
+                else:
+                    wire["rank_offset"] = client_rank[0]
+                    client_rank[0] += cfg.threads
+                # rewrite servers list to hostport form for the engine
+                srv_list = sorted(server_hosts)
+                wire["servers"] = srv_list
+            else:
+                wire["rank_offset"] = i * cfg.threads
             hs.done = False
             hs.error = ""
             hs.last_status = {}
@@ -250,6 +293,10 @@ class RemoteRunner:
             if phase_name in ("STAT", "RMFILES"):
                 return dirs * cfg.files * cfg.threads * n_hosts, 0
             return 0, 0
+        if phase_name == "NETBENCH":
+            n_clients = getattr(self, "_nb_total_client_threads",
+                                len(self.hosts) * cfg.threads)
+            return 0, cfg.file_size * n_clients
         if phase_name in ("WRITE", "READ"):
             total = cfg.file_size * len(cfg.paths)
             if cfg.no_svc_share:

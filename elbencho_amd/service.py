@@ -63,6 +63,8 @@ class ServiceState:
                 cfg.gpu_ids = list(self.base_cfg.gpu_ids)
             cfg.service_mode = False
             cfg.hosts = []
+            if wire_cfg.get("service_port"):
+                cfg.service_port = wire_cfg["service_port"]
             self.cfg = cfg
             self.runner = LocalRunner(cfg)
             self.results = None

@@ -115,3 +115,22 @@ def test_quit_services(services):
         except OSError:
             return
     raise AssertionError("service did not quit")
+
+
+def test_netbench(services, tmp_path):
+    """Netbench: first host is server, second is client (reference
+    tools/test-examples.sh has no netbench case; this mirrors docs usage)."""
+    hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+    server = f"127.0.0.1:{services[0]}"
+    res = run_master(["--hosts", hosts, "--netbench", "-w", "-t", "2", "-s", "4m",
+                      "-b", "64k", "--servers", server, "--respsize", "1k",
+                      "--lat", str(tmp_path)])
+    assert res.returncode == 0, res.stdout + res.stderr
+    assert "NETBENCH" in res.stdout
+    # 2 client threads x 4 MiB = 8 MiB transferred
+    for line in res.stdout.splitlines():
+        if "Total MiB" in line:
+            assert line.split()[-1] == "8"
+            break
+    else:
+        raise AssertionError("no Total MiB row in:\n" + res.stdout)
