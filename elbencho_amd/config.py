@@ -298,6 +298,11 @@ class BenchConfig:
                 raise ConfigError("--netbench requires service mode (--hosts)")
             if not self.servers:
                 raise ConfigError("--netbench requires --servers")
+        if self.bench_mode == "s3":
+            if not self.s3_endpoints:
+                raise ConfigError("S3 mode requires --s3endpoints")
+            if (self.run_write or self.run_read) and self.files < 1:
+                raise ConfigError("S3 object read/write requires -N/--files >= 1")
 
         if self.threads < 1:
             raise ConfigError("number of threads must be >= 1")
@@ -342,6 +347,17 @@ class BenchConfig:
         Coordinator.cpp:311-334)."""
         if self.bench_mode == "netbench":
             return ["NETBENCH"] if self.run_write else []
+        if self.bench_mode == "s3":
+            order = [
+                ("MKDIRS", self.run_mkdirs),       # MKBUCKETS
+                ("WRITE", self.run_write),         # PUT objects
+                ("STAT", self.run_stat),           # HEAD objects
+                ("LISTOBJ", bool(self.s3_list_obj)),
+                ("READ", self.run_read),           # GET objects
+                ("RMFILES", self.run_delfiles),    # delete objects
+                ("RMDIRS", self.run_deldirs),      # RMBUCKETS
+            ]
+            return [name for name, enabled in order if enabled]
         order = [
             ("MKDIRS", self.run_mkdirs),
             ("WRITE", self.run_write),

@@ -103,6 +103,9 @@ class Coordinator:
         if cfg.hosts:
             from elbencho_amd.remote import RemoteRunner
             self.runner = RemoteRunner(cfg)
+        elif cfg.bench_mode == "s3":
+            from elbencho_amd.s3 import S3Runner
+            self.runner = S3Runner(cfg)
         else:
             import os
             if "RANK" in os.environ and "WORLD_SIZE" in os.environ:
@@ -197,16 +200,22 @@ class Coordinator:
             self.run_phase("DROPCACHES", quiet=False)
 
     # ------------------------------------------------------------------
+    S3_DISPLAY = {"MKDIRS": "MKBUCKETS", "RMDIRS": "RMBUCKETS", "WRITE": "WRITE",
+                  "READ": "READ", "STAT": "HEADOBJ", "RMFILES": "RMOBJECTS",
+                  "LISTOBJ": "LISTOBJ"}
+
     def run_phase(self, name: str, quiet: bool = False) -> bool:
         cfg = self.cfg
         phase_id = str(uuid.uuid4())
         start_time = time.time()
+        display = (self.S3_DISPLAY.get(name, name)
+                   if cfg.bench_mode == "s3" else name)
 
         planned_entries, planned_bytes = self.runner.planned_work(name)
 
         cpu_first_meter = CpuUtil()  # phase start -> stonewall
         cpu_last_meter = CpuUtil()   # phase start -> phase end
-        live = LiveStatsPrinter(cfg, name, planned_entries, planned_bytes)
+        live = LiveStatsPrinter(cfg, display, planned_entries, planned_bytes)
         live_csv = LiveCsvWriter(cfg.live_csv, cfg, name) if cfg.live_csv else None
 
         if self.dist:
@@ -240,7 +249,7 @@ class Coordinator:
             cpu_first = cpu_last
         workers = self.runner.finish()
 
-        results = aggregate_phase(name, phase_id, start_time, workers,
+        results = aggregate_phase(display, phase_id, start_time, workers,
                                   cpu_first, cpu_last)
 
         if self.dist:

@@ -1,0 +1,625 @@
+"""S3 object storage engine.
+
+Reference analogue: the LocalWorker S3 engine (+ toolkits/S3Tk,
+S3UploadStore) — /root/reference/source/workers/LocalWorker.cpp:3822-7290:
+bucket create/delete, single-part and multipart upload, (ranged) download,
+HEAD, listing (+verify), multi-delete, random-object reads, object prefix,
+integrity verify. Independent implementation: a minimal SigV4 REST client on
+http.client with one persistent connection per worker thread (the reference
+links the AWS SDK; this environment has no SDK and S3 throughput is
+network-bound, so a native-C++ client is not the bottleneck).
+
+Endpoint round-robin across workers matches S3Tk.cpp:167.
+"""
+
+from __future__ import annotations
+
+import datetime
+import hashlib
+import hmac
+import http.client
+import threading
+import time
+import urllib.parse
+import xml.etree.ElementTree as ET
+from dataclasses import dataclass
+from typing import Any, Optional
+
+from elbencho_amd import load_core
+from elbencho_amd.config import BenchConfig
+from elbencho_amd.histogram import Histogram
+from elbencho_amd.stats import WorkerStats
+
+EMPTY_SHA256 = hashlib.sha256(b"").hexdigest()
+
+
+class S3Error(RuntimeError):
+    pass
+
+
+# ---------------------------------------------------------------------------
+# SigV4 signing + REST client
+# ---------------------------------------------------------------------------
+
+class S3Client:
+    """One S3 endpoint connection with AWS Signature V4 (path-style URLs)."""
+
+    def __init__(self, endpoint: str, access_key: str, secret_key: str,
+                 region: str = "us-east-1", timeout: float = 60.0):
+        u = urllib.parse.urlparse(endpoint if "//" in endpoint else "http://" + endpoint)
+        self.host = u.hostname or "localhost"
+        self.port = u.port or (443 if u.scheme == "https" else 80)
+        self.secure = u.scheme == "https"
+        self.access_key = access_key
+        self.secret_key = secret_key
+        self.region = region or "us-east-1"
+        self.timeout = timeout
+        self._conn: Optional[http.client.HTTPConnection] = None
+
+    # --- low level ---
+    def _connect(self):
+        cls = http.client.HTTPSConnection if self.secure else http.client.HTTPConnection
+        self._conn = cls(self.host, self.port, timeout=self.timeout)
+
+    def close(self):
+        if self._conn:
+            self._conn.close()
+            self._conn = None
+
+    def _sign(self, method: str, path: str, query: dict[str, str],
+              headers: dict[str, str], payload_hash: str) -> dict[str, str]:
+        now = datetime.datetime.now(datetime.timezone.utc)
+        amz_date = now.strftime("%Y%m%dT%H%M%SZ")
+        datestamp = now.strftime("%Y%m%d")
+
+        headers = dict(headers)
+        headers["host"] = f"{self.host}:{self.port}"
+        headers["x-amz-date"] = amz_date
+        headers["x-amz-content-sha256"] = payload_hash
+
+        canonical_query = "&".join(
+            f"{urllib.parse.quote(k, safe='')}={urllib.parse.quote(v, safe='')}"
+            for k, v in sorted(query.items()))
+        signed_headers = ";".join(sorted(h.lower() for h in headers))
+        canonical_headers = "".join(
+            f"{k.lower()}:{headers[k].strip()}\n" for k in sorted(headers, key=str.lower))
+        canonical_request = "\n".join([
+            method, urllib.parse.quote(path), canonical_query, canonical_headers,
+            signed_headers, payload_hash])
+
+        scope = f"{datestamp}/{self.region}/s3/aws4_request"
+        string_to_sign = "\n".join([
+            "AWS4-HMAC-SHA256", amz_date, scope,
+            hashlib.sha256(canonical_request.encode()).hexdigest()])
+
+        def hmac_sha256(key: bytes, msg: str) -> bytes:
+            return hmac.new(key, msg.encode(), hashlib.sha256).digest()
+
+        k = hmac_sha256(("AWS4" + self.secret_key).encode(), datestamp)
+        k = hmac_sha256(k, self.region)
+        k = hmac_sha256(k, "s3")
+        k = hmac_sha256(k, "aws4_request")
+        sig = hmac.new(k, string_to_sign.encode(), hashlib.sha256).hexdigest()
+
+        headers["Authorization"] = (
+            f"AWS4-HMAC-SHA256 Credential={self.access_key}/{scope}, "
+            f"SignedHeaders={signed_headers}, Signature={sig}")
+        return headers
+
+    def request(self, method: str, path: str, query: dict[str, str] | None = None,
+                body: bytes = b"", headers: dict[str, str] | None = None,
+                want_body: bool = True) -> tuple[int, bytes, dict[str, str]]:
+        query = query or {}
+        headers = headers or {}
+        payload_hash = hashlib.sha256(body).hexdigest() if body else EMPTY_SHA256
+        headers = self._sign(method, path, query, headers, payload_hash)
+
+        qs = urllib.parse.urlencode(query)
+        url = path + ("?" + qs if qs else "")
+
+        for attempt in (0, 1):  # one reconnect retry on stale connections
+            if self._conn is None:
+                self._connect()
+            try:
+                self._conn.request(method, url, body=body or None, headers=headers)
+                resp = self._conn.getresponse()
+                data = resp.read() if want_body else resp.read()
+                return resp.status, data, dict(resp.getheaders())
+            except (ConnectionError, http.client.HTTPException, OSError):
+                self.close()
+                if attempt:
+                    raise
+        raise S3Error("unreachable")
+
+    def _check(self, status: int, data: bytes, what: str):
+        if status >= 300:
+            raise S3Error(f"{what} failed: HTTP {status}: {data[:300].decode(errors='replace')}")
+
+    # --- bucket ops ---
+    def create_bucket(self, bucket: str):
+        status, data, _ = self.request("PUT", f"/{bucket}")
+        if status == 409:  # BucketAlreadyOwnedByYou — idempotent like the reference
+            return
+        self._check(status, data, f"create bucket {bucket}")
+
+    def delete_bucket(self, bucket: str):
+        status, data, _ = self.request("DELETE", f"/{bucket}")
+        self._check(status, data, f"delete bucket {bucket}")
+
+    def head_bucket(self, bucket: str) -> bool:
+        status, _, _ = self.request("HEAD", f"/{bucket}", want_body=False)
+        return status < 300
+
+    # --- object ops ---
+    def put_object(self, bucket: str, key: str, body: bytes):
+        status, data, _ = self.request("PUT", f"/{bucket}/{key}", body=body)
+        self._check(status, data, f"put {bucket}/{key}")
+
+    def get_object(self, bucket: str, key: str,
+                   byte_range: tuple[int, int] | None = None) -> bytes:
+        headers = {}
+        if byte_range:
+            headers["Range"] = f"bytes={byte_range[0]}-{byte_range[1]}"
+        status, data, _ = self.request("GET", f"/{bucket}/{key}", headers=headers)
+        self._check(status, data, f"get {bucket}/{key}")
+        return data
+
+    def head_object(self, bucket: str, key: str) -> dict[str, str]:
+        status, data, headers = self.request("HEAD", f"/{bucket}/{key}", want_body=False)
+        self._check(status, b"", f"head {bucket}/{key}")
+        return headers
+
+    def delete_object(self, bucket: str, key: str):
+        status, data, _ = self.request("DELETE", f"/{bucket}/{key}")
+        self._check(status, data, f"delete {bucket}/{key}")
+
+    def multi_delete(self, bucket: str, keys: list[str]):
+        objs = "".join(f"<Object><Key>{k}</Key></Object>" for k in keys)
+        body = (f"<Delete><Quiet>true</Quiet>{objs}</Delete>").encode()
+        md5 = __import__("base64").b64encode(hashlib.md5(body).digest()).decode()
+        status, data, _ = self.request("POST", f"/{bucket}", query={"delete": ""},
+                                       body=body, headers={"Content-MD5": md5})
+        self._check(status, data, f"multi-delete in {bucket}")
+
+    def list_objects(self, bucket: str, prefix: str = "", max_keys: int = 1000,
+                     continuation: str = "") -> tuple[list[tuple[str, int]], str]:
+        q = {"list-type": "2", "max-keys": str(max_keys)}
+        if prefix:
+            q["prefix"] = prefix
+        if continuation:
+            q["continuation-token"] = continuation
+        status, data, _ = self.request("GET", f"/{bucket}", query=q)
+        self._check(status, data, f"list {bucket}")
+        root = ET.fromstring(data)
+        ns = root.tag.split("}")[0] + "}" if "}" in root.tag else ""
+        out = []
+        for c in root.findall(f"{ns}Contents"):
+            key = c.find(f"{ns}Key").text
+            size = int(c.find(f"{ns}Size").text)
+            out.append((key, size))
+        token_el = root.find(f"{ns}NextContinuationToken")
+        return out, (token_el.text if token_el is not None else "")
+
+    # --- multipart ---
+    def create_multipart(self, bucket: str, key: str) -> str:
+        status, data, _ = self.request("POST", f"/{bucket}/{key}", query={"uploads": ""})
+        self._check(status, data, f"initiate multipart {bucket}/{key}")
+        root = ET.fromstring(data)
+        ns = root.tag.split("}")[0] + "}" if "}" in root.tag else ""
+        return root.find(f"{ns}UploadId").text
+
+    def upload_part(self, bucket: str, key: str, upload_id: str, part_num: int,
+                    body: bytes) -> str:
+        status, data, headers = self.request(
+            "PUT", f"/{bucket}/{key}",
+            query={"partNumber": str(part_num), "uploadId": upload_id}, body=body)
+        self._check(status, data, f"upload part {part_num} of {bucket}/{key}")
+        return headers.get("ETag", headers.get("etag", f'"{part_num}"'))
+
+    def complete_multipart(self, bucket: str, key: str, upload_id: str,
+                           parts: list[tuple[int, str]]):
+        parts_xml = "".join(
+            f"<Part><PartNumber>{n}</PartNumber><ETag>{etag}</ETag></Part>"
+            for n, etag in sorted(parts))
+        body = f"<CompleteMultipartUpload>{parts_xml}</CompleteMultipartUpload>".encode()
+        status, data, _ = self.request("POST", f"/{bucket}/{key}",
+                                       query={"uploadId": upload_id}, body=body)
+        self._check(status, data, f"complete multipart {bucket}/{key}")
+
+    def abort_multipart(self, bucket: str, key: str, upload_id: str):
+        status, data, _ = self.request("DELETE", f"/{bucket}/{key}",
+                                       query={"uploadId": upload_id})
+        if status not in (204, 404):
+            self._check(status, data, f"abort multipart {bucket}/{key}")
+
+
+# ---------------------------------------------------------------------------
+# shared multipart upload registry (reference S3UploadStore.{h,cpp})
+# ---------------------------------------------------------------------------
+
+class SharedUploadStore:
+    """First-writer-wins uploadId registry + completed-part collection, so
+    multiple workers can upload disjoint part ranges of one object."""
+
+    def __init__(self):
+        self.lock = threading.Lock()
+        self.uploads: dict[tuple[str, str], dict] = {}
+
+    def get_or_create(self, client: S3Client, bucket: str, key: str,
+                      num_parts_total: int) -> str:
+        with self.lock:
+            ent = self.uploads.get((bucket, key))
+            if ent is None:
+                upload_id = client.create_multipart(bucket, key)
+                ent = {"id": upload_id, "parts": [], "total": num_parts_total}
+                self.uploads[(bucket, key)] = ent
+            return ent["id"]
+
+    def add_part(self, bucket: str, key: str, part_num: int, etag: str) -> bool:
+        """Record a completed part; True when all parts are done (the caller
+        that gets True completes the upload — 'whoever finishes last')."""
+        with self.lock:
+            ent = self.uploads[(bucket, key)]
+            ent["parts"].append((part_num, etag))
+            return len(ent["parts"]) >= ent["total"]
+
+    def get_parts(self, bucket: str, key: str) -> list[tuple[int, str]]:
+        with self.lock:
+            return list(self.uploads[(bucket, key)]["parts"])
+
+    def abort_unfinished(self, client: S3Client):
+        with self.lock:
+            for (bucket, key), ent in self.uploads.items():
+                if len(ent["parts"]) < ent["total"]:
+                    try:
+                        client.abort_multipart(bucket, key, ent["id"])
+                    except (S3Error, OSError):
+                        pass
+
+
+# ---------------------------------------------------------------------------
+# the S3 benchmark runner (Coordinator backend)
+# ---------------------------------------------------------------------------
+
+@dataclass
+class _Counters:
+    entries: int = 0
+    bytes: int = 0
+    iops: int = 0
+
+
+class S3Worker(threading.Thread):
+    def __init__(self, runner: "S3Runner", local_rank: int, phase: str):
+        super().__init__(daemon=True)
+        self.r = runner
+        self.local_rank = local_rank
+        self.rank = runner.cfg.rank_offset + local_rank
+        self.phase = phase
+        self.ops = _Counters()
+        self.sw: Optional[_Counters] = None
+        self.sw_elapsed_us = 0
+        self.io_lat = Histogram()
+        self.entry_lat = Histogram()
+        self.error = ""
+        self.elapsed_us = 0
+        cfg = runner.cfg
+        ep = cfg.s3_endpoints[self.rank % len(cfg.s3_endpoints)]
+        self.client = S3Client(ep, cfg.s3_key, cfg.s3_secret, cfg.s3_region)
+        self.core = load_core()
+
+    # --- object name layout mirrors dir mode: r{rank}/d{dir}/r{rank}-f{file} ---
+    def _object_names(self):
+        cfg = self.r.cfg
+        dirs = max(cfg.dirs, 1)
+        for d in range(dirs):
+            for f in range(cfg.files):
+                if cfg.dirs > 0:
+                    yield f"{cfg.s3_obj_prefix}r{self.rank}/d{d}/r{self.rank}-f{f}"
+                else:
+                    yield f"{cfg.s3_obj_prefix}r{self.rank}-f{f}"
+
+    def _bucket(self, idx: int = 0) -> str:
+        buckets = self.r.buckets
+        return buckets[(self.rank + idx) % len(buckets)]
+
+    def _check_interrupt(self):
+        if self.r.interrupt_flag.is_set():
+            raise KeyboardInterrupt
+
+    def run(self):
+        try:
+            self.r.start_gate.wait()
+            t0 = time.monotonic()
+            self._run_phase()
+            self.elapsed_us = int((time.monotonic() - t0) * 1e6)
+        except KeyboardInterrupt:
+            self.error = "interrupted"
+        except Exception as e:  # noqa: BLE001
+            self.error = str(e)
+            self.r.interrupt_flag.set()
+        finally:
+            self.elapsed_us = self.elapsed_us or int(
+                (time.monotonic() - self.r.phase_start) * 1e6)
+            self.r.on_worker_done(self)
+
+    # ------------------------------------------------------------------
+    def _run_phase(self):
+        cfg = self.r.cfg
+        ph = self.phase
+        if ph == "MKDIRS":  # MKBUCKETS
+            for i, b in enumerate(self.r.buckets):
+                if i % cfg.num_dataset_threads == self.rank:
+                    self.client.create_bucket(b)
+                    self.ops.entries += 1
+        elif ph == "RMDIRS":  # RMBUCKETS
+            for i, b in enumerate(self.r.buckets):
+                if i % cfg.num_dataset_threads == self.rank:
+                    self.client.delete_bucket(b)
+                    self.ops.entries += 1
+        elif ph == "WRITE":
+            self._put_objects()
+        elif ph == "READ":
+            self._get_objects()
+        elif ph == "STAT":  # HEADOBJ
+            for name in self._object_names():
+                self._check_interrupt()
+                t0 = time.monotonic()
+                self.client.head_object(self._bucket(), name)
+                self.entry_lat.vec = _add_lat(self.entry_lat, t0)
+                self.ops.entries += 1
+        elif ph == "RMFILES":  # RMOBJECTS / MULTIDEL
+            if cfg.s3_multi_del > 0:
+                batch: list[str] = []
+                for name in self._object_names():
+                    batch.append(name)
+                    if len(batch) >= cfg.s3_multi_del:
+                        self._check_interrupt()
+                        self.client.multi_delete(self._bucket(), batch)
+                        self.ops.entries += len(batch)
+                        batch = []
+                if batch:
+                    self.client.multi_delete(self._bucket(), batch)
+                    self.ops.entries += len(batch)
+            else:
+                for name in self._object_names():
+                    self._check_interrupt()
+                    t0 = time.monotonic()
+                    self.client.delete_object(self._bucket(), name)
+                    self.entry_lat.vec = _add_lat(self.entry_lat, t0)
+                    self.ops.entries += 1
+        elif ph == "LISTOBJ":
+            self._list_objects()
+        else:
+            raise S3Error(f"S3 phase not supported: {ph}")
+
+    # ------------------------------------------------------------------
+    def _make_block(self, length: int, obj_off: int) -> bytes:
+        cfg = self.r.cfg
+        if cfg.verify >= 0:
+            return self.core.fill_checksum(length, obj_off, cfg.verify)
+        return bytes(self.r.rand_block[:length])
+
+    def _put_objects(self):
+        cfg = self.r.cfg
+        size = cfg.file_size
+        bs = cfg.block_size
+        for name in self._object_names():
+            self._check_interrupt()
+            te = time.monotonic()
+            bucket = self._bucket()
+            if size <= bs:  # single part
+                t0 = time.monotonic()
+                self.client.put_object(bucket, name, self._make_block(size, 0))
+                self.io_lat.vec = _add_lat(self.io_lat, t0)
+                self.ops.bytes += size
+                self.ops.iops += 1
+            else:  # multipart: block size = part size (reference -b semantics)
+                upload_id = self.client.create_multipart(bucket, name)
+                parts = []
+                off = 0
+                part_num = 1
+                try:
+                    while off < size:
+                        self._check_interrupt()
+                        ln = min(bs, size - off)
+                        t0 = time.monotonic()
+                        etag = self.client.upload_part(bucket, name, upload_id,
+                                                       part_num, self._make_block(ln, off))
+                        self.io_lat.vec = _add_lat(self.io_lat, t0)
+                        parts.append((part_num, etag))
+                        self.ops.bytes += ln
+                        self.ops.iops += 1
+                        off += ln
+                        part_num += 1
+                    self.client.complete_multipart(bucket, name, upload_id, parts)
+                except BaseException:
+                    self.client.abort_multipart(bucket, name, upload_id)
+                    raise
+            self.entry_lat.vec = _add_lat(self.entry_lat, te)
+            self.ops.entries += 1
+
+    def _get_objects(self):
+        cfg = self.r.cfg
+        size = cfg.file_size
+        bs = cfg.block_size
+        for name in self._object_names():
+            self._check_interrupt()
+            te = time.monotonic()
+            bucket = self._bucket()
+            off = 0
+            while off < size:
+                ln = min(bs, size - off)
+                t0 = time.monotonic()
+                data = self.client.get_object(bucket, name, (off, off + ln - 1))
+                self.io_lat.vec = _add_lat(self.io_lat, t0)
+                if len(data) != ln:
+                    raise S3Error(f"short ranged read of {name}: {len(data)} != {ln}")
+                if cfg.verify >= 0 and not cfg.s3_fastget:
+                    bad = self.core.verify_checksum(data, off, cfg.verify)
+                    if bad != 2**64 - 1:
+                        raise S3Error(f"S3 data verification failed for {name} at "
+                                      f"object offset {bad}")
+                self.ops.bytes += ln
+                self.ops.iops += 1
+                off += ln
+            self.entry_lat.vec = _add_lat(self.entry_lat, te)
+            self.ops.entries += 1
+
+    def _list_objects(self):
+        cfg = self.r.cfg
+        if self.local_rank != 0:
+            return
+        for bucket in self.r.buckets:
+            token = ""
+            seen = 0
+            while True:
+                self._check_interrupt()
+                objs, token = self.client.list_objects(
+                    bucket, prefix=cfg.s3_obj_prefix,
+                    max_keys=min(1000, cfg.s3_list_obj or 1000),
+                    continuation=token)
+                seen += len(objs)
+                self.ops.entries += len(objs)
+                if not token or (cfg.s3_list_obj and seen >= cfg.s3_list_obj):
+                    break
+            if cfg.s3_list_verify:
+                expected = (cfg.num_dataset_threads * max(cfg.dirs, 1) * cfg.files)
+                if seen != expected:
+                    raise S3Error(f"listing verification failed for {bucket}: "
+                                  f"saw {seen} objects, expected {expected}")
+
+
+def _add_lat(h: Histogram, t0: float) -> list[int]:
+    us = int((time.monotonic() - t0) * 1e6)
+    idx = _bucket_index(us)
+    v = h.vec
+    v[0] += 1
+    v[1] += us
+    v[2] = min(v[2], us)
+    v[3] = max(v[3], us)
+    v[4 + idx] += 1
+    return v
+
+
+def _bucket_index(v: int) -> int:
+    if v < 4:
+        return v
+    log2v = v.bit_length() - 1
+    frac = (v >> (log2v - 2)) & 3
+    return log2v * 4 + frac - 4
+
+
+class S3Runner:
+    """Coordinator backend for S3 mode (bench_mode == "s3")."""
+
+    def __init__(self, cfg: BenchConfig):
+        if not cfg.s3_endpoints:
+            raise S3Error("S3 mode requires --s3endpoints")
+        self.cfg = cfg
+        self.buckets = [p[len("s3://"):] if p.startswith("s3://") else p
+                        for p in cfg.paths]
+        if not self.buckets:
+            raise S3Error("S3 mode requires s3://bucket paths")
+        self.workers: list[S3Worker] = []
+        self.interrupt_flag = threading.Event()
+        self.start_gate = threading.Event()
+        self.done_count = 0
+        self.done_lock = threading.Lock()
+        self.done_cv = threading.Condition(self.done_lock)
+        self.stonewalled = False
+        self.phase_start = 0.0
+        # pre-generated random payload block (like the engine's host buffers)
+        import os as _os
+        self.rand_block = bytearray(_os.urandom(min(cfg.block_size, 1 << 22)))
+        while len(self.rand_block) < cfg.block_size:
+            self.rand_block += self.rand_block
+        self.rand_block = self.rand_block[:cfg.block_size]
+        self.upload_store = SharedUploadStore()
+
+    # --- runner interface ---
+    def start(self, phase_name: str) -> None:
+        self.interrupt_flag.clear()
+        self.start_gate.clear()
+        self.done_count = 0
+        self.stonewalled = False
+        self.workers = [S3Worker(self, i, phase_name) for i in range(self.cfg.threads)]
+        for w in self.workers:
+            w.start()
+        self.phase_start = time.monotonic()
+        self.start_gate.set()
+
+    def on_worker_done(self, w: S3Worker) -> None:
+        with self.done_cv:
+            if not self.stonewalled and not w.error and (
+                    w.ops.bytes or w.ops.entries or w.ops.iops):
+                elapsed = int((time.monotonic() - self.phase_start) * 1e6)
+                for peer in self.workers:
+                    peer.sw = _Counters(peer.ops.entries, peer.ops.bytes, peer.ops.iops)
+                    peer.sw_elapsed_us = elapsed
+                self.stonewalled = True
+            self.done_count += 1
+            self.done_cv.notify_all()
+
+    def wait(self, timeout_ms: int) -> bool:
+        with self.done_cv:
+            return self.done_cv.wait_for(
+                lambda: self.done_count >= len(self.workers),
+                timeout=None if timeout_ms < 0 else timeout_ms / 1000.0)
+
+    def poll(self) -> dict[str, Any]:
+        agg = {"entries": 0, "bytes": 0, "iops": 0,
+               "workers_done": self.done_count, "workers_total": len(self.workers),
+               "workers_with_error": sum(1 for w in self.workers if w.error),
+               "elapsed_usec": int((time.monotonic() - self.phase_start) * 1e6),
+               "stonewall_triggered": self.stonewalled,
+               "lat_num_ios": 0, "lat_sum_ios": 0, "lat_num_entries": 0,
+               "lat_sum_entries": 0}
+        for w in self.workers:
+            agg["entries"] += w.ops.entries
+            agg["bytes"] += w.ops.bytes
+            agg["iops"] += w.ops.iops
+            agg["lat_num_ios"] += w.io_lat.vec[0]
+            agg["lat_sum_ios"] += w.io_lat.vec[1]
+        return agg
+
+    def interrupt(self) -> None:
+        self.interrupt_flag.set()
+
+    def finish(self) -> list[WorkerStats]:
+        for w in self.workers:
+            w.join()
+        out = []
+        for w in self.workers:
+            sw = w.sw or w.ops
+            out.append(WorkerStats(
+                rank=w.rank,
+                elapsed_usec=w.elapsed_us,
+                entries=w.ops.entries, bytes=w.ops.bytes, iops=w.ops.iops,
+                stonewall_elapsed_usec=w.sw_elapsed_us or w.elapsed_us,
+                stonewall_entries=sw.entries, stonewall_bytes=sw.bytes,
+                stonewall_iops=sw.iops,
+                io_lat=list(w.io_lat.vec), entry_lat=list(w.entry_lat.vec),
+                error=w.error))
+        return out
+
+    def planned_work(self, phase_name: str) -> tuple[int, int]:
+        cfg = self.cfg
+        nobj = max(cfg.dirs, 1) * cfg.files * cfg.threads
+        if phase_name in ("WRITE", "READ"):
+            return nobj, nobj * cfg.file_size
+        if phase_name in ("STAT", "RMFILES"):
+            return nobj, 0
+        if phase_name in ("MKDIRS", "RMDIRS"):
+            return len(self.buckets), 0
+        return 0, 0
+
+    def close(self) -> None:
+        for w in self.workers:
+            if w.is_alive():
+                self.interrupt_flag.set()
+        # abort unfinished shared multipart uploads (reference behavior)
+        if self.workers:
+            try:
+                self.upload_store.abort_unfinished(self.workers[0].client)
+            except (S3Error, OSError):
+                pass

@@ -211,7 +211,11 @@ def print_results_table_header(out=None) -> None:
 
 def print_phase_results(cfg: BenchConfig, r: PhaseResults, out=None) -> None:
     out = out or sys.stdout
-    entry_type = "Dirs" if r.phase_name in ("MKDIRS", "RMDIRS") else "Files"
+    if cfg.bench_mode == "s3":
+        entry_type = ("Buckets" if r.phase_name in ("MKBUCKETS", "RMBUCKETS")
+                      else "Objects")
+    else:
+        entry_type = "Dirs" if r.phase_name in ("MKDIRS", "RMDIRS") else "Files"
 
     rows: list[tuple[str, Any, Any]] = []
     rows.append(("Elapsed time",
