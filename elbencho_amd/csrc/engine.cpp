@@ -156,6 +156,7 @@ void Worker::allocBuffers()
 {
     const auto& cfg = eng.cfg;
     int slots = std::max(1, cfg.ioDepth);
+    if (!cfg.gpuIDs.empty()) slots = std::max(2, slots); // staging pipeline
 
     if (!cfg.gpuIDs.empty()) {
         setupGpu();
@@ -229,7 +230,7 @@ void Worker::setupGpu()
 {
     const auto& cfg = eng.cfg;
     int devId = cfg.gpuIDs[globalRank % cfg.gpuIDs.size()];
-    int slots = std::max(1, cfg.ioDepth);
+    int slots = std::max(2, cfg.ioDepth);
 
     if (gpuDeviceCount() <= 0)
         throw WorkerError("GPU requested (gpuids) but no HIP device is available — "
@@ -602,6 +603,123 @@ void Worker::fileModeBlocks(bool isWrite)
         gen = makeOffsetGen(myStart, myLen);
     }
 
+    // GPU staging pipeline: with two slots and per-slot events, storage I/O
+    // of block i overlaps the PCIe copy of block i-1 and GPU verify batches
+    // 64 blocks per stream sync. Plain per-block path when latency histograms
+    // (exact per-op timing), mmap, flock or rwmix are requested.
+    const bool gpuPipelined = gpu && cfg.ioDepth == 1 && !cfg.measureLat &&
+                              !cfg.useMmap && cfg.flockMode == 0 && !rwMixActive &&
+                              !cfg.verifyDirect && hostBufs.size() >= 2;
+    if (gpuPipelined) {
+        constexpr uint64_t VERIFY_FETCH_INTERVAL = 64;
+        const bool doVerify = cfg.verifySalt >= 0;
+        bool slotBusy[2] = {false, false};
+        uint64_t sinceFetch = 0;
+        BlockSpec spec;
+        uint64_t opCount = 0;
+
+        auto fetchVerify = [&]() {
+            GpuVerifyResult r = gpu->fetchVerifyResult();
+            if (r.numMismatches)
+                throw WorkerError(
+                    "Data verification failed (GPU). First bad file offset: " +
+                    std::to_string(r.firstBadFileOffset) + "; mismatching 8-byte words: " +
+                    std::to_string(r.numMismatches));
+        };
+
+        if (isWrite) {
+            // one-block lookahead: prepare (fill + D2H) next while pwriting cur
+            int slot = 0;
+            bool havePrev = false;
+            int prevSlot = 0;
+            uint64_t prevFileIdx = 0, prevInFileOff = 0, prevIoLen = 0;
+
+            for (;;) {
+                bool haveCur = false;
+                uint64_t fileIdx = 0, inFileOff = 0, ioLen = 0;
+                while (gen->next(spec)) {
+                    if (mapBlock(spec, fileIdx, inFileOff, ioLen)) {
+                        haveCur = true;
+                        break;
+                    }
+                }
+
+                if (haveCur) {
+                    if ((opCount++ % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+                    rateLimiter.wait(ioLen);
+                    if (slotBusy[slot]) gpu->waitSlotEvent(slot);
+                    preWriteFill(slot, ioLen, inFileOff);
+                    gpu->copyD2HAsync(slot, ioLen);
+                    gpu->recordSlotEvent(slot);
+                    slotBusy[slot] = true;
+                }
+
+                if (havePrev) { // write the previously prepared block
+                    gpu->waitSlotEvent(prevSlot);
+                    ssize_t res = pwrite(fg.fds[prevFileIdx], hostBufs[prevSlot],
+                                         prevIoLen, prevInFileOff);
+                    if (res != (ssize_t)prevIoLen) throwErrno("write", cfg.paths[prevFileIdx]);
+                    liveOps.bytes.fetch_add(prevIoLen, std::memory_order_relaxed);
+                    liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+                }
+
+                if (!haveCur) break;
+                havePrev = true;
+                prevSlot = slot;
+                prevFileIdx = fileIdx;
+                prevInFileOff = inFileOff;
+                prevIoLen = ioLen;
+                slot ^= 1;
+            }
+            if (cfg.fsyncPerFile)
+                for (size_t i = 0; i < fg.fds.size(); i++)
+                    if (fsync(fg.fds[i])) throwErrno("fsync", cfg.paths[i]);
+        } else {
+            int slot = 0;
+            while (gen->next(spec)) {
+                if ((opCount++ % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+
+                uint64_t fileIdx, inFileOff, ioLen;
+                if (!mapBlock(spec, fileIdx, inFileOff, ioLen)) continue;
+
+                rateLimiter.wait(ioLen);
+                if (slotBusy[slot]) gpu->waitSlotEvent(slot); // host buf reuse
+
+                ssize_t res = pread(fg.fds[fileIdx], hostBufs[slot], ioLen, inFileOff);
+                if (res != (ssize_t)ioLen) throwErrno("read", cfg.paths[fileIdx]);
+
+                gpu->copyH2DAsync(slot, ioLen);
+                if (doVerify) {
+                    if ((inFileOff % 8 == 0) && (ioLen % 16 == 0)) {
+                        gpu->verifyChecksumDevAsync(slot, ioLen, inFileOff,
+                                                    (uint64_t)cfg.verifySalt);
+                        if (++sinceFetch >= VERIFY_FETCH_INTERVAL) {
+                            fetchVerify();
+                            sinceFetch = 0;
+                        }
+                    } else { // odd tail: CPU check of the host copy
+                        uint64_t bad = verifyChecksumCPU(hostBufs[slot], ioLen, inFileOff,
+                                                         (uint64_t)cfg.verifySalt);
+                        if (bad != UINT64_MAX)
+                            throw WorkerError(
+                                "Data verification failed. First bad file offset: " +
+                                std::to_string(bad));
+                    }
+                }
+                gpu->recordSlotEvent(slot);
+                slotBusy[slot] = true;
+
+                liveOps.bytes.fetch_add(ioLen, std::memory_order_relaxed);
+                liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+                slot ^= 1;
+            }
+            if (doVerify) fetchVerify();
+        }
+
+        gpu->syncStream(); // drain outstanding staging copies before finishing
+        return;
+    }
+
     const bool lat = cfg.measureLat;
     BlockSpec spec;
     uint64_t opCount = 0;
@@ -716,6 +834,7 @@ void Worker::fileModeBlocksUring(bool isWrite)
         bool isWriteOp = true;
     };
     std::vector<SlotState> slots(depth);
+    std::vector<char> slotHasCopy(depth, 0);
 
     BlockSpec spec;
     int inFlight = 0;
@@ -731,6 +850,10 @@ void Worker::fileModeBlocksUring(bool isWrite)
         bool blockWrite = isWrite && !mixRead;
 
         rateLimiter.wait(ioLen);
+        if (gpu && slotHasCopy[slot]) { // prior async staging of this slot
+            gpu->waitSlotEvent(slot);
+            slotHasCopy[slot] = false;
+        }
         if (blockWrite) {
             preWriteFill(slot, ioLen, inFileOff);
             if (gpu) {
@@ -776,10 +899,16 @@ void Worker::fileModeBlocksUring(bool isWrite)
             if (!wasWrite) {
                 if (gpu) {
                     gpu->copyH2DAsync(slot, st.len);
-                    gpu->syncStream();
-                }
-                if (!isWrite) // rwmix reads in a write phase skip pattern checks
+                    if (cfg.verifySalt >= 0 && !isWrite) {
+                        gpu->syncStream();
+                        postReadCheck(slot, st.len, st.inFileOff);
+                    } else { // pipelined: wait only when the slot is reused
+                        gpu->recordSlotEvent(slot);
+                        slotHasCopy[slot] = 1;
+                    }
+                } else if (!isWrite) {
                     postReadCheck(slot, st.len, st.inFileOff);
+                }
             }
 
             const bool mixRead = isWrite && !wasWrite;
@@ -796,6 +925,8 @@ void Worker::fileModeBlocksUring(bool isWrite)
             if (!exhausted && !prepSlot(slot)) exhausted = true;
         }
     }
+
+    if (gpu) gpu->syncStream(); // drain async staging copies
 }
 
 void Worker::fileModeDelete()

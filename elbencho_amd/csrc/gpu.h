@@ -56,6 +56,17 @@ public:
     void copyD2HAsync(int slot, uint64_t len);
     void syncStream();
 
+    // per-slot completion events: record after the slot's async op, wait
+    // before reusing the slot's buffers (enables storage<->PCIe pipelining)
+    void recordSlotEvent(int slot);
+    void waitSlotEvent(int slot);
+
+    // async verify accumulating into persistent device counters; results
+    // fetched (and reset) by fetchVerifyResult() — lets the caller batch
+    // many blocks per stream synchronization
+    void verifyChecksumDevAsync(int slot, uint64_t len, uint64_t fileOff, uint64_t salt);
+    GpuVerifyResult fetchVerifyResult();
+
     // --- device-side buffer ops (hand-written gfx950 kernels) ---
 
     // Fill device buffer with xoshiro256++ random data (replaces curand).
