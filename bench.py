@@ -117,6 +117,7 @@ def main() -> int:
     if args.workload == "randread":
         mcfg["random"] = True
         mcfg["block_size"] = 4096
+        args.block = 4096
     meng = core.Engine(mcfg)
     meng.prepare()
 
@@ -154,8 +155,13 @@ def main() -> int:
     ms_per_step = elapsed * 1000.0 / args.steps
 
     if rank == 0:
+        metric_by_workload = {
+            "seqread": "seq-read GiB/s into GPU HBM",
+            "seqwrite": "seq-write GiB/s from GPU HBM",
+            "randread": "4K-random-read GiB/s into GPU HBM",
+        }
         doc = {
-            "metric": "seq-read GiB/s into GPU HBM",
+            "metric": metric_by_workload[args.workload],
             "value": round(value, 3),
             "unit": "GiB/s",
             "n_gpus": world if use_gpu else 0,
@@ -179,6 +185,8 @@ def main() -> int:
                 "parallelism": f"dp{world}" if world > 1 else "single",
             },
         }
+        if args.workload == "randread":
+            doc["config"]["iops_4k"] = int(total_bytes / 4096 / elapsed)
         print(json.dumps(doc), flush=True)
 
     # cleanup (tmpfs is shared RAM — do not leak multi-GiB files)

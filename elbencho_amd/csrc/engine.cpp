@@ -607,9 +607,13 @@ void Worker::fileModeBlocks(bool isWrite)
     // of block i overlaps the PCIe copy of block i-1 and GPU verify batches
     // 64 blocks per stream sync. Plain per-block path when latency histograms
     // (exact per-op timing), mmap, flock or rwmix are requested.
-    const bool gpuPipelined = gpu && cfg.ioDepth == 1 && !cfg.measureLat &&
-                              !cfg.useMmap && cfg.flockMode == 0 && !rwMixActive &&
-                              !cfg.verifyDirect && hostBufs.size() >= 2;
+    static const bool pipelineDisabled = [] {
+        const char* v = getenv("EB_GPU_PIPELINE");
+        return v && v[0] == '0';
+    }();
+    const bool gpuPipelined = gpu && !pipelineDisabled && cfg.ioDepth == 1 &&
+                              !cfg.measureLat && !cfg.useMmap && cfg.flockMode == 0 &&
+                              !rwMixActive && !cfg.verifyDirect && hostBufs.size() >= 2;
     if (gpuPipelined) {
         constexpr uint64_t VERIFY_FETCH_INTERVAL = 64;
         const bool doVerify = cfg.verifySalt >= 0;
