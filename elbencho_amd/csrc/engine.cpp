@@ -156,7 +156,7 @@ void Worker::allocBuffers()
 {
     const auto& cfg = eng.cfg;
     int slots = std::max(1, cfg.ioDepth);
-    if (!cfg.gpuIDs.empty()) slots = std::max(2, slots); // staging pipeline
+    if (!cfg.gpuIDs.empty()) slots = std::max(4, slots); // staging pipeline
 
     if (!cfg.gpuIDs.empty()) {
         setupGpu();
@@ -230,7 +230,7 @@ void Worker::setupGpu()
 {
     const auto& cfg = eng.cfg;
     int devId = cfg.gpuIDs[globalRank % cfg.gpuIDs.size()];
-    int slots = std::max(2, cfg.ioDepth);
+    int slots = std::max(4, cfg.ioDepth);
 
     if (gpuDeviceCount() <= 0)
         throw WorkerError("GPU requested (gpuids) but no HIP device is available — "
@@ -617,7 +617,8 @@ void Worker::fileModeBlocks(bool isWrite)
     if (gpuPipelined) {
         constexpr uint64_t VERIFY_FETCH_INTERVAL = 64;
         const bool doVerify = cfg.verifySalt >= 0;
-        bool slotBusy[2] = {false, false};
+        const int nSlots = (int)hostBufs.size();
+        std::vector<char> slotBusy(nSlots, 0);
         uint64_t sinceFetch = 0;
         BlockSpec spec;
         uint64_t opCount = 0;
@@ -655,7 +656,7 @@ void Worker::fileModeBlocks(bool isWrite)
                     preWriteFill(slot, ioLen, inFileOff);
                     gpu->copyD2HAsync(slot, ioLen);
                     gpu->recordSlotEvent(slot);
-                    slotBusy[slot] = true;
+                    slotBusy[slot] = 1;
                 }
 
                 if (havePrev) { // write the previously prepared block
@@ -673,7 +674,7 @@ void Worker::fileModeBlocks(bool isWrite)
                 prevFileIdx = fileIdx;
                 prevInFileOff = inFileOff;
                 prevIoLen = ioLen;
-                slot ^= 1;
+                slot = (slot + 1) % nSlots;
             }
             if (cfg.fsyncPerFile)
                 for (size_t i = 0; i < fg.fds.size(); i++)
@@ -711,11 +712,11 @@ void Worker::fileModeBlocks(bool isWrite)
                     }
                 }
                 gpu->recordSlotEvent(slot);
-                slotBusy[slot] = true;
+                slotBusy[slot] = 1;
 
                 liveOps.bytes.fetch_add(ioLen, std::memory_order_relaxed);
                 liveOps.iops.fetch_add(1, std::memory_order_relaxed);
-                slot ^= 1;
+                slot = (slot + 1) % nSlots;
             }
             if (doVerify) fetchVerify();
         }
