@@ -116,6 +116,17 @@ static void mkdirIgnoreExists(const std::string& path)
     if (mkdir(path.c_str(), 0777) && errno != EEXIST) throwErrno("mkdir", path);
 }
 
+// staging ring depth for GPU mode (tunable for experiments)
+static int gpuSlotCount(int ioDepth)
+{
+    static const int def = [] {
+        const char* v = getenv("EB_GPU_SLOTS");
+        int n = v ? atoi(v) : 2;
+        return (n >= 1 && n <= 64) ? n : 2;
+    }();
+    return std::max(def, ioDepth);
+}
+
 // ---------------------------------------------------------------------------
 // Worker
 // ---------------------------------------------------------------------------
@@ -156,7 +167,7 @@ void Worker::allocBuffers()
 {
     const auto& cfg = eng.cfg;
     int slots = std::max(1, cfg.ioDepth);
-    if (!cfg.gpuIDs.empty()) slots = std::max(4, slots); // staging pipeline
+    if (!cfg.gpuIDs.empty()) slots = gpuSlotCount(cfg.ioDepth); // staging pipeline
 
     if (!cfg.gpuIDs.empty()) {
         setupGpu();
@@ -230,7 +241,7 @@ void Worker::setupGpu()
 {
     const auto& cfg = eng.cfg;
     int devId = cfg.gpuIDs[globalRank % cfg.gpuIDs.size()];
-    int slots = std::max(4, cfg.ioDepth);
+    int slots = gpuSlotCount(cfg.ioDepth);
 
     if (gpuDeviceCount() <= 0)
         throw WorkerError("GPU requested (gpuids) but no HIP device is available — "
