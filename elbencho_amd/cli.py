@@ -367,6 +367,7 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.live_int_ms = args.liveint
     cfg.no_live = args.nolive
     cfg.live1 = args.live1
+    cfg.live1n = args.live1n
     cfg.live_csv = args.livecsv
     cfg.live_csv_ex = args.livecsvex
     cfg.ops_log_path = args.opslog

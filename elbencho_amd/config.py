@@ -128,6 +128,7 @@ class BenchConfig:
     live_int_ms: int = 2000        # --liveint
     no_live: bool = False          # --nolive
     live1: bool = False            # --live1
+    live1n: bool = False           # --live1n
     live_csv: str = ""             # --livecsv
     live_csv_ex: bool = False      # --livecsvex
     csv_file: str = ""             # --csvfile

@@ -48,6 +48,9 @@ class LocalRunner:
     def poll(self) -> dict[str, Any]:
         return self.engine.poll()
 
+    def poll_workers(self) -> list[dict[str, Any]]:
+        return self.engine.poll_workers()
+
     def wait(self, timeout_ms: int) -> bool:
         return self.engine.wait_phase_done(timeout_ms)
 
@@ -187,8 +190,13 @@ class Coordinator:
                     rc = 1
                     break
                 self._run_sync_and_dropcaches()
-                if i < len(phases) - 1 and cfg.phase_delay_secs:
-                    time.sleep(cfg.phase_delay_secs)
+                if i < len(phases) - 1:
+                    if cfg.phase_delay_secs:
+                        time.sleep(cfg.phase_delay_secs)
+                    if cfg.rotate_hosts:
+                        rot = getattr(self.runner, "rotate_hosts", None)
+                        if rot:
+                            rot(cfg.rotate_hosts)
             if rc:
                 break
         return rc
@@ -215,8 +223,24 @@ class Coordinator:
 
         cpu_first_meter = CpuUtil()  # phase start -> stonewall
         cpu_last_meter = CpuUtil()   # phase start -> phase end
-        live = LiveStatsPrinter(cfg, display, planned_entries, planned_bytes)
-        live_csv = LiveCsvWriter(cfg.live_csv, cfg, name) if cfg.live_csv else None
+
+        # live display: fullscreen per-worker dashboard on a TTY by default,
+        # --live1 single line, --live1n newline mode (reference behavior)
+        from elbencho_amd.livestats import (FullscreenLiveStats, LiveCsvExWriter,
+                                            NewlineLiveStats)
+        if cfg.live1 or not sys.stderr.isatty():
+            live = LiveStatsPrinter(cfg, display, planned_entries, planned_bytes)
+        elif getattr(cfg, "live1n", False):
+            live = NewlineLiveStats(cfg, display)
+        else:
+            live = FullscreenLiveStats(cfg, display, planned_bytes, planned_entries)
+        want_worker_rows = isinstance(live, FullscreenLiveStats) or cfg.live_csv_ex
+        if cfg.live_csv and cfg.live_csv_ex:
+            live_csv = LiveCsvExWriter(cfg.live_csv, cfg, name)
+        elif cfg.live_csv:
+            live_csv = LiveCsvWriter(cfg.live_csv, cfg, name)
+        else:
+            live_csv = None
 
         if self.dist:
             self.dist.barrier()  # lockstep phase start across GPU ranks
@@ -231,9 +255,19 @@ class Coordinator:
             p = self.runner.poll()
             if cpu_first is None and p.get("stonewall_triggered"):
                 cpu_first = cpu_first_meter.percent_since_last()
-            live.update(p)
+            rows = None
+            if want_worker_rows:
+                getter = getattr(self.runner, "poll_workers", None)
+                rows = getter() if getter else None
+            try:
+                live.update(p, rows)
+            except TypeError:
+                live.update(p)
             if live_csv:
-                live_csv.update(p)
+                if isinstance(live_csv, LiveCsvExWriter):
+                    live_csv.update(p, rows or [])
+                else:
+                    live_csv.update(p)
             if deadline and time.monotonic() > deadline:
                 print(f"\nPhase time limit reached ({cfg.timelimit}s); "
                       "interrupting workers...", file=sys.stderr)

@@ -271,6 +271,19 @@ PYBIND11_MODULE(_core, m)
                  d["lat_sum_entries"] = lp.latSumEntries;
                  return d;
              })
+        .def("poll_workers",
+             [](Engine& e) {
+                 py::list out;
+                 for (auto& w : e.workers) {
+                     py::dict d;
+                     d["rank"] = w->globalRank;
+                     d["entries"] = w->liveOps.entries.load();
+                     d["bytes"] = w->liveOps.bytes.load() + w->liveOpsReadMix.bytes.load();
+                     d["iops"] = w->liveOps.iops.load() + w->liveOpsReadMix.iops.load();
+                     out.append(d);
+                 }
+                 return out;
+             })
         .def("finish_phase",
              [](Engine& e) {
                  std::vector<WorkerResult> rs;

@@ -261,6 +261,27 @@ class RemoteRunner:
         return agg
 
     # ------------------------------------------------------------------
+    def poll_workers(self):
+        """Per-service rows for the fullscreen dashboard."""
+        rows = []
+        for i, hs in enumerate(self.hosts):
+            st = hs.last_status
+            if not st or st.get("idle"):
+                continue
+            rows.append({"rank": i, "entries": st.get("entries", 0),
+                         "bytes": st.get("bytes", 0), "iops": st.get("iops", 0)})
+        return rows
+
+    # ------------------------------------------------------------------
+    def rotate_hosts(self, n: int) -> None:
+        """--rotatehosts: rotate the service list by n between phases
+        (reference Coordinator::rotateHosts, Coordinator.cpp:384)."""
+        if not n or len(self.hosts) < 2:
+            return
+        n %= len(self.hosts)
+        self.hosts = self.hosts[n:] + self.hosts[:n]
+
+    # ------------------------------------------------------------------
     def interrupt(self) -> None:
         try:
             self._for_all(lambda i, hs: hs.client.get("/interruptphase"))

@@ -520,7 +520,7 @@ class LiveStatsPrinter:
         self.enabled = (not cfg.no_live) and out.isatty()
         self._printed = False
 
-    def update(self, poll: dict[str, Any]) -> None:
+    def update(self, poll: dict[str, Any], worker_rows=None) -> None:
         if not self.enabled:
             return
         now = time.monotonic()
