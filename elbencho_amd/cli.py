@@ -261,6 +261,11 @@ def build_parser() -> argparse.ArgumentParser:
                         "servers; requires --servers).")
     g.add_argument("--servers", default="", metavar="LIST",
                    help="Comma-separated netbench server hosts.")
+    g.add_argument("--clients", default="", metavar="LIST",
+                   help="Comma-separated netbench client service hosts "
+                        "(added to --hosts).")
+    g.add_argument("--clientsfile", default="", metavar="PATH",
+                   help="File with one netbench client host per line.")
     g.add_argument("--serversfile", default="", metavar="PATH",
                    help="File with one netbench server per line.")
     g.add_argument("--respsize", default="1", metavar="SIZE",
@@ -422,6 +427,14 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
 
     cfg.netbench = args.netbench
     cfg.servers = [s for s in args.servers.split(",") if s] if args.servers else []
+    clients = [s for s in args.clients.split(",") if s] if args.clients else []
+    if args.clientsfile:
+        with open(args.clientsfile) as f:
+            clients = [ln.strip() for ln in f if ln.strip()]
+    if args.netbench and clients:
+        # --servers + --clients spell out the full service set
+        cfg.hosts = (cfg.hosts or []) + [h for h in cfg.servers + clients
+                                         if h not in (cfg.hosts or [])]
     if args.serversfile:
         with open(args.serversfile) as f:
             cfg.servers = [ln.strip() for ln in f if ln.strip()]

@@ -187,6 +187,7 @@ class RemoteRunner:
 
         def prep(i: int, hs: HostState):
             wire = cfg.to_wire()
+            wire["service_index"] = i
             wire["threads"] = cfg.threads
             wire["num_dataset_threads"] = (cfg.threads * num_hosts) if shared else cfg.threads
             if cfg.bench_mode == "netbench":

@@ -65,6 +65,10 @@ class ServiceState:
             cfg.hosts = []
             if wire_cfg.get("service_port"):
                 cfg.service_port = wire_cfg["service_port"]
+            if cfg.gpu_per_service and cfg.gpu_ids:
+                # one GPU (set) per service instead of per thread
+                idx = int(wire_cfg.get("service_index", 0))
+                cfg.gpu_ids = [cfg.gpu_ids[idx % len(cfg.gpu_ids)]]
             self.cfg = cfg
             self.runner = LocalRunner(cfg)
             self.results = None

@@ -1,0 +1,143 @@
+#!/bin/bash
+#
+# Test a few simple benchmark cases — the behavior contract of
+# reference tools/test-examples.sh, ported to elbencho-amd:
+#  * block-device tests on self-created loopback devices (random-read
+#    latency, multi-threaded 4K write IOPS over 2 devices, 1M streaming read)
+#  * multi-file dir-mode create/read/delete with --verify data integrity
+#  * distributed tests without a real cluster: two local services driven by
+#    a master via --hosts host:[PORT1-PORT2], then --quit
+#  * optional random-IO tests asserting file allocation after full-coverage
+#    random writes
+#
+# Requires root for the loopback device setup (-b to skip).
+
+set -u
+
+SCRIPT_PATH=$(dirname "$0")
+EB="python3 -m elbencho_amd"
+export PYTHONPATH="${SCRIPT_PATH}/..:${PYTHONPATH:-}"
+
+LOOP_BACKING_FILES=("backing1" "backing2")
+LOOP_BACKING_FILE_SIZE=$((10*1024*1024))
+LOOPDEV_PATHS=()
+SERVICE_PORT_START=${SERVICE_PORT_START:=1711}
+SERVICE_PORTS=( "$SERVICE_PORT_START" $((SERVICE_PORT_START + 1)) )
+
+SKIP_BLOCKDEV_TESTS=0
+SKIP_MULTIFILE_TESTS=0
+SKIP_DISTRIBUTED_TESTS=0
+RUN_RANDOM_IO_TESTS=0
+unset BASE_DIR
+
+usage()
+{
+  echo "Usage: $0 [-b skip blockdev] [-d skip distributed] [-m skip multifile]"
+  echo "          [-r run random-IO tests] BASEDIR"
+  exit 1
+}
+
+while getopts ":bdhmr" opt; do
+  case "${opt}" in
+    b) SKIP_BLOCKDEV_TESTS=1 ;;
+    d) SKIP_DISTRIBUTED_TESTS=1 ;;
+    m) SKIP_MULTIFILE_TESTS=1 ;;
+    r) RUN_RANDOM_IO_TESTS=1 ;;
+    *) usage ;;
+  esac
+done
+shift $((OPTIND-1))
+[ $# -ne 1 ] && usage
+BASE_DIR=$1
+[ -d "$BASE_DIR" ] || { echo "ERROR: BASEDIR must exist: $BASE_DIR"; exit 1; }
+
+die() { echo "ERROR: $1"; cleanup_loopdev; exit 1; }
+
+run() {
+  echo "  \$ elbencho-amd $*"
+  $EB --nolive "$@" || die "command failed: $*"
+}
+
+# ---------------------------------------------------------------------------
+prep_loopdev()
+{
+  for (( i=0; i < ${#LOOP_BACKING_FILES[@]}; i++ )); do
+    truncate -s "$LOOP_BACKING_FILE_SIZE" "${BASE_DIR}/${LOOP_BACKING_FILES[$i]}" \
+      || die "backing file creation failed"
+    LOOPDEV_PATHS[$i]=$(losetup --show -f "${BASE_DIR}/${LOOP_BACKING_FILES[$i]}") \
+      || die "losetup failed (need root; use -b to skip blockdev tests)"
+    chmod o+rw "${LOOPDEV_PATHS[$i]}"
+  done
+}
+
+cleanup_loopdev()
+{
+  for dev in "${LOOPDEV_PATHS[@]:-}"; do
+    [ -n "$dev" ] && losetup -d "$dev" 2>/dev/null
+  done
+  for f in "${LOOP_BACKING_FILES[@]}"; do
+    rm -f "${BASE_DIR}/${f}"
+  done
+}
+
+blockdev_tests()
+{
+  echo "== Block device tests (loopback) =="
+  prep_loopdev
+  echo "-- 4KiB random read latency of ${LOOPDEV_PATHS[0]}:"
+  run -r -b 4K --lat --cpu --direct --rand --no0usecerr "${LOOPDEV_PATHS[0]}"
+  echo "-- 4KiB 16-thread QD16 write IOPS over both devices:"
+  run -w -b 4K -t 16 --iodepth 16 --direct --rand --no0usecerr \
+      "${LOOPDEV_PATHS[0]}" "${LOOPDEV_PATHS[1]}"
+  echo "-- 1MiB 8-thread streaming read of ${LOOPDEV_PATHS[0]}:"
+  run -r -b 1M -t 8 --iodepth 4 --direct --no0usecerr "${LOOPDEV_PATHS[0]}"
+  cleanup_loopdev
+}
+
+multifile_tests()
+{
+  echo "== Multi-file dir mode tests =="
+  echo "-- 2 threads x 3 dirs x 4 x 1MiB files, write with --verify:"
+  run -t 2 -d -n 3 -w -N 4 -s 1m -b 1m --lat --verify 1 --no0usecerr "$BASE_DIR"
+  echo "-- read back in 128KiB blocks with --verify:"
+  run -t 2 -n 3 -r -N 4 -s 1m -b 128k --verify 1 --no0usecerr "$BASE_DIR"
+  echo "-- delete files and dirs:"
+  run -t 2 -n 3 -N 4 -F -D --no0usecerr "$BASE_DIR"
+}
+
+distributed_tests()
+{
+  echo "== Distributed tests (two local services) =="
+  $EB --service --foreground --port "${SERVICE_PORTS[0]}" --zones 0 2>/dev/null &
+  SVC1=$!
+  $EB --service --foreground --port "${SERVICE_PORTS[1]}" --cores 0 2>/dev/null &
+  SVC2=$!
+  sleep 2
+  echo "-- master drives both services (4 threads x 8 dirs x 16 x 4KiB files):"
+  $EB --nolive --hosts "localhost:[${SERVICE_PORTS[0]}-${SERVICE_PORTS[1]}]" \
+      -t 4 -d -n 8 -w -r -N 16 -s 4k -F -D --verify 1 --no0usecerr "$BASE_DIR" \
+      || { kill $SVC1 $SVC2 2>/dev/null; die "distributed test failed"; }
+  echo "-- stopping services via --quit:"
+  $EB --hosts "localhost:${SERVICE_PORTS[0]},localhost:${SERVICE_PORTS[1]}" --quit
+  wait $SVC1 $SVC2 2>/dev/null
+}
+
+random_io_tests()
+{
+  echo "== Random IO tests =="
+  echo "-- 4 threads x full-coverage random 4K writes into 13 x 1MiB files:"
+  run -w -t 4 -b 4k -s 1m --rand --no0usecerr "${BASE_DIR}/testfile[1-13]"
+  for i in $(seq 1 13); do
+    sz=$(stat -c %s "${BASE_DIR}/testfile$i")
+    [ "$sz" -eq $((1024*1024)) ] || die "testfile$i not fully allocated ($sz)"
+  done
+  rm -f "${BASE_DIR}"/testfile*
+  echo "-- full allocation after random writes verified."
+}
+
+[ "$SKIP_BLOCKDEV_TESTS" -eq 0 ] && blockdev_tests
+[ "$SKIP_MULTIFILE_TESTS" -eq 0 ] && multifile_tests
+[ "$SKIP_DISTRIBUTED_TESTS" -eq 0 ] && distributed_tests
+[ "$RUN_RANDOM_IO_TESTS" -eq 1 ] && random_io_tests
+
+echo "All tests passed."
