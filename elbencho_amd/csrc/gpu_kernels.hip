@@ -257,8 +257,15 @@ GpuCtx::GpuCtx(int deviceId, int numSlots, uint64_t bufSize, bool pinnedHostBufs
     }
 
     impl->slotEvents.resize(numSlots, nullptr);
+    // EB_GPU_EVBLOCK=1: block instead of spin in hipEventSynchronize — frees
+    // host cores for pread/pwrite when many worker threads wait on events
+    static const bool evBlock = [] {
+        const char* v = getenv("EB_GPU_EVBLOCK");
+        return v && v[0] == '1';
+    }();
+    unsigned evFlags = hipEventDisableTiming | (evBlock ? hipEventBlockingSync : 0);
     for (int i = 0; i < numSlots; i++)
-        HIP_CHECK(hipEventCreateWithFlags(&impl->slotEvents[i], hipEventDisableTiming));
+        HIP_CHECK(hipEventCreateWithFlags(&impl->slotEvents[i], evFlags));
 
     HIP_CHECK(hipMalloc(&impl->verifyOutDev, 2 * sizeof(unsigned long long)));
     HIP_CHECK(hipHostMalloc(&impl->verifyOutHost, 2 * sizeof(unsigned long long),
