@@ -135,7 +135,13 @@ random_io_tests()
   echo "-- full allocation after random writes verified."
 }
 
-[ "$SKIP_BLOCKDEV_TESTS" -eq 0 ] && blockdev_tests
+if [ "$SKIP_BLOCKDEV_TESTS" -eq 0 ]; then
+  if losetup -f >/dev/null 2>&1; then
+    blockdev_tests
+  else
+    echo "WARNING: no free loopback device available; skipping blockdev tests."
+  fi
+fi
 [ "$SKIP_MULTIFILE_TESTS" -eq 0 ] && multifile_tests
 [ "$SKIP_DISTRIBUTED_TESTS" -eq 0 ] && distributed_tests
 [ "$RUN_RANDOM_IO_TESTS" -eq 1 ] && random_io_tests
