@@ -57,6 +57,9 @@ class LocalRunner:
     def interrupt(self) -> None:
         self.engine.interrupt()
 
+    def trigger_stonewall(self) -> None:
+        self.engine.trigger_stonewall()
+
     def finish(self) -> list[WorkerStats]:
         return [WorkerStats.from_engine(d) for d in self.engine.finish_phase()]
 
@@ -292,6 +295,11 @@ class Coordinator:
         if not quiet or results.errors:
             if self.dist is None or self.dist.rank == 0:
                 print_phase_results(cfg, results, self.out)
+                if cfg.svc_elapsed:
+                    for hostport, us in getattr(self.runner, "last_service_elapsed", []):
+                        from elbencho_amd.units import elapsed_ms_to_human
+                        print(f"{'':<11} {'Svc ' + hostport:<17}: "
+                              f"{elapsed_ms_to_human(us // 1000):>11}", file=self.out)
 
         if cfg.csv_file and (self.dist is None or self.dist.rank == 0):
             append_csv_result(cfg, results, cfg.csv_file)

@@ -1760,6 +1760,22 @@ bool Engine::waitPhaseDone(int64_t timeoutMs)
 
 void Engine::interrupt() { interruptFlag.store(true); }
 
+void Engine::triggerStonewall()
+{
+    // remote stonewall propagation: another service's first finisher defines
+    // the global first-done point; snapshot local workers now (reference
+    // RemoteWorker stonewall handling)
+    std::lock_guard<std::mutex> lk(doneMtx);
+    if (stonewallTriggered.load()) return;
+    uint64_t elapsed = nowUSecSince(phaseStart);
+    for (auto& peer : workers) {
+        peer->stonewallOps.takeFrom(peer->liveOps);
+        peer->stonewallOpsReadMix.takeFrom(peer->liveOpsReadMix);
+        peer->stonewallElapsedUSec = elapsed;
+    }
+    stonewallTriggered.store(true);
+}
+
 Engine::LivePoll Engine::poll()
 {
     LivePoll lp{};

@@ -226,6 +226,7 @@ public:
     void startPhase(Phase phase);
     bool waitPhaseDone(int64_t timeoutMs); // true when all workers finished
     void interrupt();
+    void triggerStonewall(); // remote stonewall propagation (master RPC)
 
     struct LivePoll {
         uint64_t entries, bytes, iops;

@@ -253,6 +253,7 @@ PYBIND11_MODULE(_core, m)
         .def("wait_phase_done", &Engine::waitPhaseDone,
              py::call_guard<py::gil_scoped_release>())
         .def("interrupt", &Engine::interrupt)
+        .def("trigger_stonewall", &Engine::triggerStonewall)
         .def("poll",
              [](Engine& e) {
                  Engine::LivePoll lp = e.poll();

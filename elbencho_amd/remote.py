@@ -160,6 +160,7 @@ class RemoteRunner:
         self._phase = phase_name
         self._bench_id = str(uuid.uuid4())
         self._poll_ms = FIRST_POLL_MS
+        self._stonewall_sent = False
         shared = not cfg.no_svc_share and cfg.path_type != "dir"
         num_hosts = len(self.hosts)
 
@@ -241,6 +242,23 @@ class RemoteRunner:
                 hs.done = True
 
         self._for_all(poll)
+
+        # stonewall propagation: the first service whose worker finished its
+        # fair share defines the whole job's first-done point — tell the
+        # other services to snapshot now (reference RemoteWorker :453-583)
+        if not self._stonewall_sent and any(
+                hs.last_status.get("stonewall_triggered") for hs in self.hosts):
+            self._stonewall_sent = True
+
+            def sw(i: int, hs: HostState):
+                if not hs.last_status.get("stonewall_triggered"):
+                    hs.client.get("/triggerstonewall")
+
+            try:
+                self._for_all(sw)
+            except RuntimeError:
+                pass  # best effort; local finish times still bound the result
+
         return all(hs.done for hs in self.hosts)
 
     # ------------------------------------------------------------------
@@ -295,10 +313,16 @@ class RemoteRunner:
             return hs.client.get("/benchresult")
 
         all_workers: list[WorkerStats] = []
-        for res in self._for_all(collect):
+        self.last_service_elapsed = []  # (hostport, max elapsed usec)
+        for hs, res in zip(self.hosts, self._for_all(collect)):
+            host_workers = []
             for w in res["workers"]:
                 ws = WorkerStats(**{k: w[k] for k in w if k in WorkerStats.__dataclass_fields__})
-                all_workers.append(ws)
+                host_workers.append(ws)
+            all_workers.extend(host_workers)
+            self.last_service_elapsed.append(
+                (hs.client.hostport,
+                 max((w.elapsed_usec for w in host_workers), default=0)))
         return all_workers
 
     # ------------------------------------------------------------------

@@ -116,6 +116,11 @@ class ServiceState:
             if self.runner:
                 self.runner.interrupt()
 
+    def trigger_stonewall(self) -> None:
+        with self.lock:
+            if self.runner:
+                self.runner.trigger_stonewall()
+
 
 class Handler(BaseHTTPRequestHandler):
     state: ServiceState  # set by run_service
@@ -166,6 +171,9 @@ class Handler(BaseHTTPRequestHandler):
                 phase = q.get("phase", ["IDLE"])[0]
                 bench_id = q.get("benchid", [""])[0]
                 self.state.start_phase(phase, bench_id)
+                self._send_json({"ok": True})
+            elif url.path == "/triggerstonewall":
+                self.state.trigger_stonewall()
                 self._send_json({"ok": True})
             elif url.path == "/interruptphase":
                 self.state.interrupt()
