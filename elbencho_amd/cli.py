@@ -282,6 +282,25 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--s3secret", default="", metavar="SECRET", help="S3 secret key.")
     g.add_argument("--s3region", default="", metavar="REGION", help="S3 region.")
     g.add_argument("--s3objprefix", default="", metavar="STR", help="S3 object name prefix.")
+    g.add_argument("--s3aclput", action="store_true", help="Phase: put object ACLs.")
+    g.add_argument("--s3aclget", action="store_true", help="Phase: get object ACLs.")
+    g.add_argument("--s3aclverify", action="store_true",
+                   help="Verify object ACLs in the get phase.")
+    g.add_argument("--s3aclgrants", default="", metavar="ACL",
+                   help="Canned ACL for --s3aclput (e.g. private, public-read).")
+    g.add_argument("--s3baclput", action="store_true", help="Phase: put bucket ACLs.")
+    g.add_argument("--s3baclget", action="store_true", help="Phase: get bucket ACLs.")
+    g.add_argument("--s3otag", action="store_true", help="Phase: put object tagging.")
+    g.add_argument("--s3otagverify", action="store_true",
+                   help="Verify object tagging after put.")
+    g.add_argument("--s3btag", action="store_true", help="Phase: put bucket tagging.")
+    g.add_argument("--s3btagverify", action="store_true",
+                   help="Verify bucket tagging after put.")
+    g.add_argument("--s3credfile", default="", metavar="PATH",
+                   help="File with one key:secret credential pair per line, "
+                        "round-robined across workers.")
+    g.add_argument("--s3credlist", default="", metavar="LIST",
+                   help="Comma-separated key:secret pairs, round-robined across workers.")
     g.add_argument("--s3randobj", action="store_true",
                    help="Read at random offsets of random objects.")
     g.add_argument("--s3listobj", type=int, default=0, metavar="N",
@@ -453,6 +472,18 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.s3_list_verify = args.s3listverify
     cfg.s3_multi_del = args.s3multidel
     cfg.s3_fastget = args.s3fastget
+    cfg.s3_acl_put = args.s3aclput
+    cfg.s3_acl_get = args.s3aclget
+    cfg.s3_acl_verify = args.s3aclverify
+    cfg.s3_acl_grants = args.s3aclgrants
+    cfg.s3_bacl_put = args.s3baclput
+    cfg.s3_bacl_get = args.s3baclget
+    cfg.s3_otag = args.s3otag
+    cfg.s3_otag_verify = args.s3otagverify
+    cfg.s3_btag = args.s3btag
+    cfg.s3_btag_verify = args.s3btagverify
+    cfg.s3_cred_file = args.s3credfile
+    cfg.s3_cred_list = args.s3credlist
 
     cfg.config_file = args.configfile
 

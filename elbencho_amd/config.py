@@ -211,6 +211,18 @@ class BenchConfig:
     s3_sign_policy: int = 0        # --s3sign
     s3_max_conns: int = 0          # --s3maxconns
     s3_ignore_errors: bool = False
+    s3_acl_put: bool = False       # --s3aclput
+    s3_acl_get: bool = False       # --s3aclget
+    s3_acl_verify: bool = False    # --s3aclverify
+    s3_acl_grants: str = ""        # --s3aclgrants (canned ACL string)
+    s3_bacl_put: bool = False      # --s3baclput
+    s3_bacl_get: bool = False      # --s3baclget
+    s3_otag: bool = False          # --s3otag
+    s3_otag_verify: bool = False   # --s3otagverify
+    s3_btag: bool = False          # --s3btag
+    s3_btag_verify: bool = False   # --s3btagverify
+    s3_cred_file: str = ""         # --s3credfile (lines "key:secret")
+    s3_cred_list: str = ""         # --s3credlist ("key:secret,key:secret")
 
     # --- misc ---
     config_file: str = ""          # -c/--configfile
@@ -349,13 +361,23 @@ class BenchConfig:
         if self.bench_mode == "netbench":
             return ["NETBENCH"] if self.run_write else []
         if self.bench_mode == "s3":
+            # reference phase order, Coordinator.cpp:311-334
             order = [
                 ("MKDIRS", self.run_mkdirs),       # MKBUCKETS
+                ("PUTBACL", self.s3_bacl_put),
+                ("PUTBTAG", self.s3_btag),
+                ("GETBTAG", self.s3_btag and self.s3_btag_verify),
                 ("WRITE", self.run_write),         # PUT objects
+                ("PUTOBJACL", self.s3_acl_put),
+                ("PUTOTAG", self.s3_otag),
                 ("STAT", self.run_stat),           # HEAD objects
+                ("GETOTAG", self.s3_otag and self.s3_otag_verify),
+                ("GETOBJACL", self.s3_acl_get),
                 ("LISTOBJ", bool(self.s3_list_obj)),
                 ("READ", self.run_read),           # GET objects
+                ("DELOTAG", self.s3_otag and self.run_delfiles),
                 ("RMFILES", self.run_delfiles),    # delete objects
+                ("GETBACL", self.s3_bacl_get),
                 ("RMDIRS", self.run_deldirs),      # RMBUCKETS
             ]
             return [name for name, enabled in order if enabled]

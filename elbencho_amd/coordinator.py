@@ -213,7 +213,11 @@ class Coordinator:
     # ------------------------------------------------------------------
     S3_DISPLAY = {"MKDIRS": "MKBUCKETS", "RMDIRS": "RMBUCKETS", "WRITE": "WRITE",
                   "READ": "READ", "STAT": "HEADOBJ", "RMFILES": "RMOBJECTS",
-                  "LISTOBJ": "LISTOBJ"}
+                  "LISTOBJ": "LISTOBJ", "PUTOBJACL": "PUTOBJACL",
+                  "GETOBJACL": "GETOBJACL", "PUTBACL": "PUTBACL",
+                  "GETBACL": "GETBACL", "PUTOTAG": "PUTOBJMD",
+                  "GETOTAG": "GETOBJMD", "DELOTAG": "DELOBJMD",
+                  "PUTBTAG": "PUTBUCKETMD", "GETBTAG": "GETBUCKETMD"}
 
     def run_phase(self, name: str, quiet: bool = False) -> bool:
         cfg = self.cfg

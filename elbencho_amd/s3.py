@@ -200,6 +200,63 @@ class S3Client:
         token_el = root.find(f"{ns}NextContinuationToken")
         return out, (token_el.text if token_el is not None else "")
 
+    # --- ACL (canned) ---
+    def put_object_acl(self, bucket: str, key: str, canned: str):
+        status, data, _ = self.request("PUT", f"/{bucket}/{key}", query={"acl": ""},
+                                       headers={"x-amz-acl": canned})
+        self._check(status, data, f"put acl {bucket}/{key}")
+
+    def get_object_acl(self, bucket: str, key: str) -> bytes:
+        status, data, _ = self.request("GET", f"/{bucket}/{key}", query={"acl": ""})
+        self._check(status, data, f"get acl {bucket}/{key}")
+        return data
+
+    def put_bucket_acl(self, bucket: str, canned: str):
+        status, data, _ = self.request("PUT", f"/{bucket}", query={"acl": ""},
+                                       headers={"x-amz-acl": canned})
+        self._check(status, data, f"put bucket acl {bucket}")
+
+    def get_bucket_acl(self, bucket: str) -> bytes:
+        status, data, _ = self.request("GET", f"/{bucket}", query={"acl": ""})
+        self._check(status, data, f"get bucket acl {bucket}")
+        return data
+
+    # --- tagging ---
+    @staticmethod
+    def _tagset_xml(tags: dict[str, str]) -> bytes:
+        inner = "".join(f"<Tag><Key>{k}</Key><Value>{v}</Value></Tag>"
+                        for k, v in tags.items())
+        return f"<Tagging><TagSet>{inner}</TagSet></Tagging>".encode()
+
+    def put_object_tagging(self, bucket: str, key: str, tags: dict[str, str]):
+        status, data, _ = self.request("PUT", f"/{bucket}/{key}", query={"tagging": ""},
+                                       body=self._tagset_xml(tags))
+        self._check(status, data, f"put tagging {bucket}/{key}")
+
+    def get_object_tagging(self, bucket: str, key: str) -> dict[str, str]:
+        status, data, _ = self.request("GET", f"/{bucket}/{key}", query={"tagging": ""})
+        self._check(status, data, f"get tagging {bucket}/{key}")
+        import re as _re
+        return dict(_re.findall(r"<Key>([^<]*)</Key><Value>([^<]*)</Value>",
+                                data.decode()))
+
+    def delete_object_tagging(self, bucket: str, key: str):
+        status, data, _ = self.request("DELETE", f"/{bucket}/{key}", query={"tagging": ""})
+        if status not in (200, 204):
+            self._check(status, data, f"delete tagging {bucket}/{key}")
+
+    def put_bucket_tagging(self, bucket: str, tags: dict[str, str]):
+        status, data, _ = self.request("PUT", f"/{bucket}", query={"tagging": ""},
+                                       body=self._tagset_xml(tags))
+        self._check(status, data, f"put bucket tagging {bucket}")
+
+    def get_bucket_tagging(self, bucket: str) -> dict[str, str]:
+        status, data, _ = self.request("GET", f"/{bucket}", query={"tagging": ""})
+        self._check(status, data, f"get bucket tagging {bucket}")
+        import re as _re
+        return dict(_re.findall(r"<Key>([^<]*)</Key><Value>([^<]*)</Value>",
+                                data.decode()))
+
     # --- multipart ---
     def create_multipart(self, bucket: str, key: str) -> str:
         status, data, _ = self.request("POST", f"/{bucket}/{key}", query={"uploads": ""})
@@ -304,7 +361,10 @@ class S3Worker(threading.Thread):
         self.elapsed_us = 0
         cfg = runner.cfg
         ep = cfg.s3_endpoints[self.rank % len(cfg.s3_endpoints)]
-        self.client = S3Client(ep, cfg.s3_key, cfg.s3_secret, cfg.s3_region)
+        key, secret = cfg.s3_key, cfg.s3_secret
+        if runner.credentials:  # --s3credfile/--s3credlist round-robin
+            key, secret = runner.credentials[self.rank % len(runner.credentials)]
+        self.client = S3Client(ep, key, secret, cfg.s3_region)
         self.core = load_core()
 
     # --- object name layout mirrors dir mode: r{rank}/d{dir}/r{rank}-f{file} ---
@@ -389,6 +449,58 @@ class S3Worker(threading.Thread):
                     self.ops.entries += 1
         elif ph == "LISTOBJ":
             self._list_objects()
+        elif ph == "PUTOBJACL":
+            for name in self._object_names():
+                self._check_interrupt()
+                self.client.put_object_acl(self._bucket(), name, cfg.s3_acl_grants or "private")
+                self.ops.entries += 1
+        elif ph == "GETOBJACL":
+            for name in self._object_names():
+                self._check_interrupt()
+                acl = self.client.get_object_acl(self._bucket(), name)
+                if cfg.s3_acl_verify and (cfg.s3_acl_grants or "private") not in acl.decode():
+                    raise S3Error(f"object ACL verification failed for {name}")
+                self.ops.entries += 1
+        elif ph == "PUTBACL":
+            if self.local_rank == 0:
+                for b in self.r.buckets:
+                    self.client.put_bucket_acl(b, cfg.s3_acl_grants or "private")
+                    self.ops.entries += 1
+        elif ph == "GETBACL":
+            if self.local_rank == 0:
+                for b in self.r.buckets:
+                    self.client.get_bucket_acl(b)
+                    self.ops.entries += 1
+        elif ph == "PUTOTAG":
+            for name in self._object_names():
+                self._check_interrupt()
+                self.client.put_object_tagging(self._bucket(), name,
+                                               {"elbencho-amd": str(self.rank)})
+                self.ops.entries += 1
+        elif ph == "GETOTAG":
+            for name in self._object_names():
+                self._check_interrupt()
+                tags = self.client.get_object_tagging(self._bucket(), name)
+                if cfg.s3_otag_verify and tags.get("elbencho-amd") != str(self.rank):
+                    raise S3Error(f"object tagging verification failed for {name}: {tags}")
+                self.ops.entries += 1
+        elif ph == "DELOTAG":
+            for name in self._object_names():
+                self._check_interrupt()
+                self.client.delete_object_tagging(self._bucket(), name)
+                self.ops.entries += 1
+        elif ph == "PUTBTAG":
+            if self.local_rank == 0:
+                for b in self.r.buckets:
+                    self.client.put_bucket_tagging(b, {"elbencho-amd": "bucket"})
+                    self.ops.entries += 1
+        elif ph == "GETBTAG":
+            if self.local_rank == 0:
+                for b in self.r.buckets:
+                    tags = self.client.get_bucket_tagging(b)
+                    if cfg.s3_btag_verify and tags.get("elbencho-amd") != "bucket":
+                        raise S3Error(f"bucket tagging verification failed for {b}")
+                    self.ops.entries += 1
         else:
             raise S3Error(f"S3 phase not supported: {ph}")
 
@@ -442,6 +554,8 @@ class S3Worker(threading.Thread):
         cfg = self.r.cfg
         size = cfg.file_size
         bs = cfg.block_size
+        if cfg.s3_rand_obj:
+            return self._get_random_objects()
         for name in self._object_names():
             self._check_interrupt()
             te = time.monotonic()
@@ -464,6 +578,46 @@ class S3Worker(threading.Thread):
                 off += ln
             self.entry_lat.vec = _add_lat(self.entry_lat, te)
             self.ops.entries += 1
+
+    def _get_random_objects(self):
+        """--s3randobj: ranged reads at random offsets of random objects
+        (reference LocalWorker.cpp:4069). Amount: randamount/threads bytes."""
+        import random as _random
+
+        cfg = self.r.cfg
+        size = cfg.file_size
+        bs = min(cfg.block_size, size)
+        amount = (cfg.rand_amount or
+                  cfg.num_dataset_threads * max(cfg.dirs, 1) * cfg.files * size)
+        amount //= cfg.num_dataset_threads
+        rng = _random.Random(0x5EED ^ self.rank)
+        # any rank may read any rank's objects
+        all_names = []
+        for r in range(cfg.num_dataset_threads):
+            for d in range(max(cfg.dirs, 1)):
+                for f in range(cfg.files):
+                    if cfg.dirs > 0:
+                        all_names.append(f"{cfg.s3_obj_prefix}r{r}/d{d}/r{r}-f{f}")
+                    else:
+                        all_names.append(f"{cfg.s3_obj_prefix}r{r}-f{f}")
+        done = 0
+        while done < amount:
+            self._check_interrupt()
+            name = all_names[rng.randrange(len(all_names))]
+            ln = min(bs, amount - done, size)
+            off = rng.randrange(max(size - ln, 0) + 1)
+            t0 = time.monotonic()
+            data = self.client.get_object(self._bucket(), name, (off, off + ln - 1))
+            self.io_lat.vec = _add_lat(self.io_lat, t0)
+            if len(data) != ln:
+                raise S3Error(f"short random read of {name}")
+            if cfg.verify >= 0 and not cfg.s3_fastget:
+                bad = self.core.verify_checksum(data, off, cfg.verify)
+                if bad != 2**64 - 1:
+                    raise S3Error(f"S3 verification failed for {name} at offset {bad}")
+            self.ops.bytes += ln
+            self.ops.iops += 1
+            done += ln
 
     def _list_objects(self):
         cfg = self.r.cfg
@@ -520,6 +674,19 @@ class S3Runner:
                         for p in cfg.paths]
         if not self.buckets:
             raise S3Error("S3 mode requires s3://bucket paths")
+        self.credentials: list[tuple[str, str]] = []
+        if cfg.s3_cred_list:
+            for ent in cfg.s3_cred_list.split(","):
+                if ":" in ent:
+                    k, sec = ent.split(":", 1)
+                    self.credentials.append((k, sec))
+        elif cfg.s3_cred_file:
+            with open(cfg.s3_cred_file) as f:
+                for ln in f:
+                    ln = ln.strip()
+                    if ln and ":" in ln and not ln.startswith("#"):
+                        k, sec = ln.split(":", 1)
+                        self.credentials.append((k, sec))
         self.workers: list[S3Worker] = []
         self.interrupt_flag = threading.Event()
         self.start_gate = threading.Event()
@@ -611,6 +778,8 @@ class S3Runner:
             return nobj, 0
         if phase_name in ("MKDIRS", "RMDIRS"):
             return len(self.buckets), 0
+        if phase_name in ("PUTOBJACL", "GETOBJACL", "PUTOTAG", "GETOTAG", "DELOTAG"):
+            return nobj, 0
         return 0, 0
 
     def close(self) -> None:

@@ -28,6 +28,8 @@ class S3Store:
         self.uploads: dict[str, dict[int, bytes]] = {}  # uploadId -> parts
         self.upload_meta: dict[str, tuple[str, str]] = {}  # uploadId -> (bucket,key)
         self.next_upload = [0]
+        self.acls: dict[tuple, str] = {}      # (bucket, key|None) -> canned acl
+        self.tags: dict[tuple, bytes] = {}    # (bucket, key|None) -> tagging xml
 
 
 def _xml(body: str) -> bytes:
@@ -117,6 +119,12 @@ class S3Handler(BaseHTTPRequestHandler):
         bucket, key, q = self._parse()
         st = self.store
         with st.lock:
+            if "acl" in q:
+                st.acls[(bucket, key)] = self.headers.get("x-amz-acl", "private")
+                return self._send(200)
+            if "tagging" in q:
+                st.tags[(bucket, key)] = body
+                return self._send(200)
             if key is None:  # create bucket
                 if bucket in st.buckets:
                     return self._err(409, "BucketAlreadyOwnedByYou")
@@ -140,6 +148,15 @@ class S3Handler(BaseHTTPRequestHandler):
         bucket, key, q = self._parse()
         st = self.store
         with st.lock:
+            if "acl" in q:
+                acl = st.acls.get((bucket, key), "private")
+                return self._send(200, _xml(
+                    f"<AccessControlPolicy><AccessControlList><Grant>"
+                    f"<Permission>{acl}</Permission></Grant></AccessControlList>"
+                    f"</AccessControlPolicy>"))
+            if "tagging" in q:
+                return self._send(200, st.tags.get((bucket, key),
+                                                   _xml("<Tagging><TagSet></TagSet></Tagging>")))
             if bucket not in st.buckets:
                 return self._err(404, "NoSuchBucket")
             if key is None:  # list objects v2
@@ -192,6 +209,9 @@ class S3Handler(BaseHTTPRequestHandler):
         bucket, key, q = self._parse()
         st = self.store
         with st.lock:
+            if "tagging" in q:
+                st.tags.pop((bucket, key), None)
+                return self._send(204)
             if "uploadId" in q:  # abort multipart
                 st.uploads.pop(q["uploadId"], None)
                 st.upload_meta.pop(q["uploadId"], None)

@@ -122,3 +122,35 @@ def test_s3_listobj_and_multidel(mock_s3, capsys):
     rc = _cli(mock_s3, ["-F", "--s3multidel", "3", "-t", "2", "-N", "4",
                         "-s", "4k", "-b", "4k", "s3://lbkt"])
     assert rc == 0
+
+
+def test_s3_acl_and_tagging_phases(mock_s3, capsys):
+    rc = _cli(mock_s3, ["-d", "-w", "-t", "2", "-N", "2", "-s", "4k", "-b", "4k",
+                        "--s3baclput", "--s3baclget", "--s3btag", "--s3btagverify",
+                        "--s3aclput", "--s3aclget", "--s3aclverify",
+                        "--s3aclgrants", "public-read",
+                        "--s3otag", "--s3otagverify", "-F", "-D", "s3://aclbkt"])
+    out = capsys.readouterr().out
+    assert rc == 0, out
+    for phase in ("PUTBACL", "PUTBUCKETMD", "GETBUCKETMD", "PUTOBJACL", "PUTOBJMD",
+                  "GETOBJMD", "GETOBJACL", "DELOBJMD", "GETBACL"):
+        assert phase in out, f"{phase} missing in:\n{out}"
+
+
+def test_s3_randobj(mock_s3):
+    rc = _cli(mock_s3, ["-d", "-w", "-t", "2", "-N", "4", "-s", "64k", "-b", "16k",
+                        "--verify", "3", "s3://rbkt"])
+    assert rc == 0
+    rc = _cli(mock_s3, ["-r", "-t", "2", "-N", "4", "-s", "64k", "-b", "16k",
+                        "--s3randobj", "--randamount", "256k", "--verify", "3",
+                        "s3://rbkt"])
+    assert rc == 0
+
+
+def test_s3_credentials_roundrobin(mock_s3):
+    # both workers use valid creds from the list -> phases succeed
+    rc = main(["--s3endpoints", mock_s3, "--nolive",
+               "--s3credlist", f"{ACCESS_KEY}:{SECRET_KEY},{ACCESS_KEY}:{SECRET_KEY}",
+               "-d", "-w", "-F", "-D", "-t", "2", "-N", "2", "-s", "4k", "-b", "4k",
+               "s3://credbkt"])
+    assert rc == 0
