@@ -11,6 +11,8 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdio>
+#include <map>
+#include <mutex>
 #include <stdexcept>
 #include <string>
 #include <vector>
@@ -224,8 +226,36 @@ static dim3 gridForBytes(uint64_t workItems)
     return dim3((uint32_t)blocks);
 }
 
+// Optional process-wide shared stream pool (EB_GPU_SHARED_STREAMS=N): many
+// worker threads funnel their staging copies through N HIP streams per
+// device instead of one stream each — fewer SDMA queues, less switching.
+static hipStream_t sharedStream(int devId, int& poolSizeOut)
+{
+    static std::mutex mtx;
+    static std::map<int, std::vector<hipStream_t>> pools;
+    static std::map<int, int> next;
+    static const int poolSize = [] {
+        const char* v = getenv("EB_GPU_SHARED_STREAMS");
+        int n = v ? atoi(v) : 0;
+        return (n >= 1 && n <= 64) ? n : 0;
+    }();
+
+    poolSizeOut = poolSize;
+    if (!poolSize) return nullptr;
+
+    std::lock_guard<std::mutex> lk(mtx);
+    auto& pool = pools[devId];
+    if (pool.empty()) {
+        pool.resize(poolSize);
+        for (int i = 0; i < poolSize; i++)
+            HIP_CHECK(hipStreamCreateWithFlags(&pool[i], hipStreamNonBlocking));
+    }
+    return pool[next[devId]++ % poolSize];
+}
+
 struct GpuCtx::Impl {
     hipStream_t stream = nullptr;
+    bool ownStream = true;
     std::vector<char*> devBufs;
     std::vector<char*> hostBufs;
     std::vector<hipEvent_t> slotEvents;
@@ -240,7 +270,14 @@ GpuCtx::GpuCtx(int deviceId, int numSlots, uint64_t bufSize, bool pinnedHostBufs
     : impl(new Impl), devId(deviceId), slotSize(bufSize), slots(numSlots)
 {
     HIP_CHECK(hipSetDevice(deviceId));
-    HIP_CHECK(hipStreamCreateWithFlags(&impl->stream, hipStreamNonBlocking));
+    int poolSize = 0;
+    hipStream_t shared = sharedStream(deviceId, poolSize);
+    if (shared) {
+        impl->stream = shared;
+        impl->ownStream = false;
+    } else {
+        HIP_CHECK(hipStreamCreateWithFlags(&impl->stream, hipStreamNonBlocking));
+    }
 
     impl->hostPinned = pinnedHostBufs;
     impl->devBufs.resize(numSlots, nullptr);
@@ -294,7 +331,7 @@ GpuCtx::~GpuCtx()
         if (e) (void)hipEventDestroy(e);
     if (impl->verifyOutDev) (void)hipFree(impl->verifyOutDev);
     if (impl->verifyOutHost) (void)hipHostFree(impl->verifyOutHost);
-    if (impl->stream) (void)hipStreamDestroy(impl->stream);
+    if (impl->stream && impl->ownStream) (void)hipStreamDestroy(impl->stream);
     delete impl;
 }
 
