@@ -234,9 +234,12 @@ static hipStream_t sharedStream(int devId, int& poolSizeOut)
     static std::mutex mtx;
     static std::map<int, std::vector<hipStream_t>> pools;
     static std::map<int, int> next;
+    // default 8: measured best on MI355X for 16-worker staging (see
+    // profiles/r01_staging_tuning.md); EB_GPU_SHARED_STREAMS=0 gives each
+    // worker its own stream
     static const int poolSize = [] {
         const char* v = getenv("EB_GPU_SHARED_STREAMS");
-        int n = v ? atoi(v) : 0;
+        int n = v ? atoi(v) : 8;
         return (n >= 1 && n <= 64) ? n : 0;
     }();
 
