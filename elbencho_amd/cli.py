@@ -318,6 +318,7 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--s3nocompress", action="store_true", help=argparse.SUPPRESS)
 
     g = p.add_argument_group("misc")
+    g.add_argument("--hdfs", action="store_true", help=argparse.SUPPRESS)
     g.add_argument("-c", "--configfile", default="", metavar="PATH",
                    help="Read options from a config file (key=value lines).")
     g.add_argument("--version", action="version",
@@ -486,6 +487,11 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.s3_cred_list = args.s3credlist
 
     cfg.config_file = args.configfile
+
+    if args.hdfs:
+        raise ConfigError("this build does not include Hadoop HDFS support "
+                          "(libhdfs is not available in the target image; "
+                          "matching the reference's optional-feature behavior)")
 
     cfg.finalize()
     return cfg
