@@ -543,6 +543,12 @@ def main(argv: list[str] | None = None) -> int:
     from elbencho_amd.coordinator import Coordinator
 
     try:
+        from elbencho_amd import load_core
+        load_core().register_fault_handlers()
+    except (ImportError, RuntimeError):
+        pass
+
+    try:
         return Coordinator(cfg).main()
     except ConfigError as e:
         print(f"ERROR: {e}", file=sys.stderr)

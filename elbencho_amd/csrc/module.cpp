@@ -3,6 +3,14 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <execinfo.h>
+#include <csignal>
+#include <cstdio>
+#include <cstdlib>
+#include <unistd.h>
+
+#include <fcntl.h>
+
 #include "engine.h"
 #include "gpu.h"
 #include "histogram.h"
@@ -123,10 +131,51 @@ py::dict resultToDict(const WorkerResult& r)
     return d;
 }
 
+// Fault signal handlers: dump a native backtrace to stderr and a trace file
+// on SIGSEGV/FPE/BUS/ILL/ABRT (reference analogue: toolkits/SignalTk.cpp:29-53,
+// which uses boost::stacktrace; this uses glibc backtrace).
+extern "C" void ebFaultHandler(int sig)
+{
+    void* frames[64];
+    int n = backtrace(frames, 64);
+
+    const char* name = (sig == SIGSEGV)   ? "SIGSEGV (segmentation fault)"
+                       : (sig == SIGFPE)  ? "SIGFPE (floating point exception)"
+                       : (sig == SIGBUS)  ? "SIGBUS (bus error; can be caused by a "
+                                            "truncated mmap'ed file)"
+                       : (sig == SIGILL)  ? "SIGILL (illegal instruction)"
+                       : (sig == SIGABRT) ? "SIGABRT (abort)"
+                                          : "fatal signal";
+    dprintf(STDERR_FILENO, "\nelbencho-amd: caught %s — native backtrace:\n", name);
+    backtrace_symbols_fd(frames, n, STDERR_FILENO);
+
+    const char* tmpdir = getenv("TMPDIR");
+    char path[256];
+    snprintf(path, sizeof(path), "%s/elbencho_amd_fault_trace.txt",
+             tmpdir ? tmpdir : "/tmp");
+    int fd = open(path, O_WRONLY | O_CREAT | O_APPEND, 0644);
+    if (fd >= 0) {
+        dprintf(fd, "caught %s — native backtrace:\n", name);
+        backtrace_symbols_fd(frames, n, fd);
+        close(fd);
+        dprintf(STDERR_FILENO, "(trace also appended to %s)\n", path);
+    }
+
+    signal(sig, SIG_DFL);
+    raise(sig);
+}
+
+static void registerFaultHandlers()
+{
+    for (int sig : {SIGSEGV, SIGFPE, SIGBUS, SIGILL, SIGABRT}) signal(sig, ebFaultHandler);
+}
+
 } // namespace
 
 PYBIND11_MODULE(_core, m)
 {
+    m.def("register_fault_handlers", &registerFaultHandlers,
+          "Install SIGSEGV/FPE/BUS/ILL/ABRT handlers that dump a native backtrace");
     m.doc() = "elbencho_amd native I/O engine (MI355X / gfx950)";
 
     m.def("gpu_device_count", &gpuDeviceCount);
