@@ -116,15 +116,22 @@ static void mkdirIgnoreExists(const std::string& path)
     if (mkdir(path.c_str(), 0777) && errno != EEXIST) throwErrno("mkdir", path);
 }
 
-// staging ring depth for GPU mode (tunable for experiments)
-static int gpuSlotCount(int ioDepth)
+// staging ring depth for GPU mode (tunable for experiments). Small blocks
+// get a deeper ring so one H2D can batch a ~512 KiB span of consecutive
+// slots (two batches in flight).
+static int gpuSlotCount(int ioDepth, uint64_t blockSize)
 {
     static const int def = [] {
         const char* v = getenv("EB_GPU_SLOTS");
         int n = v ? atoi(v) : 2;
-        return (n >= 1 && n <= 64) ? n : 2;
+        return (n >= 1 && n <= 128) ? n : 2;
     }();
-    return std::max(def, ioDepth);
+    int slots = def;
+    if (blockSize < (256ULL << 10)) {
+        uint64_t batch = (256ULL << 10) / std::max<uint64_t>(blockSize, 1);
+        slots = (int)std::min<uint64_t>(2 * batch, 128);
+    }
+    return std::max(slots, ioDepth);
 }
 
 // ---------------------------------------------------------------------------
@@ -167,7 +174,7 @@ void Worker::allocBuffers()
 {
     const auto& cfg = eng.cfg;
     int slots = std::max(1, cfg.ioDepth);
-    if (!cfg.gpuIDs.empty()) slots = gpuSlotCount(cfg.ioDepth); // staging pipeline
+    if (!cfg.gpuIDs.empty()) slots = gpuSlotCount(cfg.ioDepth, cfg.blockSize);
 
     if (!cfg.gpuIDs.empty()) {
         setupGpu();
@@ -241,7 +248,7 @@ void Worker::setupGpu()
 {
     const auto& cfg = eng.cfg;
     int devId = cfg.gpuIDs[globalRank % cfg.gpuIDs.size()];
-    int slots = gpuSlotCount(cfg.ioDepth);
+    int slots = gpuSlotCount(cfg.ioDepth, cfg.blockSize);
 
     if (gpuDeviceCount() <= 0)
         throw WorkerError("GPU requested (gpuids) but no HIP device is available — "
@@ -690,6 +697,45 @@ void Worker::fileModeBlocks(bool isWrite)
             if (cfg.fsyncPerFile)
                 for (size_t i = 0; i < fg.fds.size(); i++)
                     if (fsync(fg.fds[i])) throwErrno("fsync", cfg.paths[i]);
+        } else if (cfg.blockSize < (256ULL << 10) && !doVerify && nSlots >= 4) {
+            // small-block batched read: fill half the ring with preads, then
+            // one ranged H2D covers all of them (slots are contiguous); the
+            // other half stages while this half reads.
+            const int batch = nSlots / 2;
+            int half = 0; // 0 -> slots [0, batch), 1 -> [batch, 2*batch)
+            bool halfBusy[2] = {false, false};
+
+            int filled = 0;
+            int base = 0;
+            while (gen->next(spec)) {
+                uint64_t fileIdx, inFileOff, ioLen;
+                if (!mapBlock(spec, fileIdx, inFileOff, ioLen)) continue;
+
+                if (filled == 0) {
+                    if ((opCount++ % 4) == 0) checkInterrupt();
+                    base = half * batch;
+                    if (halfBusy[half]) gpu->waitSlotEvent(base); // ring reuse
+                }
+
+                rateLimiter.wait(ioLen);
+                ssize_t res = pread(fg.fds[fileIdx], hostBufs[base + filled], ioLen,
+                                    inFileOff);
+                if (res != (ssize_t)ioLen) throwErrno("read", cfg.paths[fileIdx]);
+                liveOps.bytes.fetch_add(ioLen, std::memory_order_relaxed);
+                liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+                filled++;
+
+                if (filled == batch) {
+                    gpu->copyH2DRangeAsync(base, batch);
+                    gpu->recordSlotEvent(base); // event indexed by ring base
+                    halfBusy[half] = true;
+                    half ^= 1;
+                    filled = 0;
+                }
+            }
+            if (filled) { // tail batch
+                gpu->copyH2DRangeAsync(half * batch, filled);
+            }
         } else {
             int slot = 0;
             while (gen->next(spec)) {

@@ -53,6 +53,9 @@ public:
 
     // --- async staging copies on this worker's stream ---
     void copyH2DAsync(int slot, uint64_t len);
+    // one memcpy covering `count` consecutive slots (slot stride is the
+    // 4 KiB-rounded slot size) — batches small-block staging
+    void copyH2DRangeAsync(int firstSlot, int count);
     void copyD2HAsync(int slot, uint64_t len);
     void syncStream();
 

@@ -259,6 +259,9 @@ static hipStream_t sharedStream(int devId, int& poolSizeOut)
 struct GpuCtx::Impl {
     hipStream_t stream = nullptr;
     bool ownStream = true;
+    char* devBase = nullptr;
+    char* hostBase = nullptr;
+    uint64_t slotStride = 0;
     std::vector<char*> devBufs;
     std::vector<char*> hostBufs;
     std::vector<hipEvent_t> slotEvents;
@@ -286,14 +289,20 @@ GpuCtx::GpuCtx(int deviceId, int numSlots, uint64_t bufSize, bool pinnedHostBufs
     impl->devBufs.resize(numSlots, nullptr);
     impl->hostBufs.resize(numSlots, nullptr);
 
+    // one contiguous slab per side: slot i at base + i*slotStride. Contiguous
+    // slots let small-block staging batch N slots into ONE ranged memcpy.
+    impl->slotStride = (bufSize + 4095) & ~4095ULL; // keep O_DIRECT alignment
+    HIP_CHECK(hipMalloc(&impl->devBase, impl->slotStride * numSlots));
+    if (pinnedHostBufs) {
+        HIP_CHECK(hipHostMalloc(&impl->hostBase, impl->slotStride * numSlots,
+                                hipHostMallocDefault));
+    } else {
+        if (posix_memalign((void**)&impl->hostBase, 4096, impl->slotStride * numSlots))
+            throw std::runtime_error("host buffer alloc failed");
+    }
     for (int i = 0; i < numSlots; i++) {
-        HIP_CHECK(hipMalloc(&impl->devBufs[i], bufSize));
-        if (pinnedHostBufs) {
-            HIP_CHECK(hipHostMalloc(&impl->hostBufs[i], bufSize, hipHostMallocDefault));
-        } else {
-            if (posix_memalign((void**)&impl->hostBufs[i], 4096, bufSize))
-                throw std::runtime_error("host buffer alloc failed");
-        }
+        impl->devBufs[i] = impl->devBase + (uint64_t)i * impl->slotStride;
+        impl->hostBufs[i] = impl->hostBase + (uint64_t)i * impl->slotStride;
     }
 
     impl->slotEvents.resize(numSlots, nullptr);
@@ -321,14 +330,12 @@ GpuCtx::GpuCtx(int deviceId, int numSlots, uint64_t bufSize, bool pinnedHostBufs
 GpuCtx::~GpuCtx()
 {
     (void)hipSetDevice(devId);
-    for (auto p : impl->devBufs)
-        if (p) (void)hipFree(p);
-    for (auto p : impl->hostBufs) {
-        if (!p) continue;
+    if (impl->devBase) (void)hipFree(impl->devBase);
+    if (impl->hostBase) {
         if (impl->hostPinned)
-            (void)hipHostFree(p);
+            (void)hipHostFree(impl->hostBase);
         else
-            free(p);
+            free(impl->hostBase);
     }
     for (auto e : impl->slotEvents)
         if (e) (void)hipEventDestroy(e);
@@ -346,6 +353,14 @@ void GpuCtx::copyH2DAsync(int slot, uint64_t len)
 {
     HIP_CHECK(hipMemcpyAsync(impl->devBufs[slot], impl->hostBufs[slot], len,
                              hipMemcpyHostToDevice, impl->stream));
+}
+
+// one memcpy spanning `count` consecutive slots (small-block batching)
+void GpuCtx::copyH2DRangeAsync(int firstSlot, int count)
+{
+    HIP_CHECK(hipMemcpyAsync(impl->devBufs[firstSlot], impl->hostBufs[firstSlot],
+                             impl->slotStride * count, hipMemcpyHostToDevice,
+                             impl->stream));
 }
 
 void GpuCtx::copyD2HAsync(int slot, uint64_t len)
