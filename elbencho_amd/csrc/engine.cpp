@@ -157,6 +157,57 @@ Worker::~Worker()
     }
 }
 
+// --- rwmix byte-ratio balancer (reference toolkits/RateLimiterRWMixThreads:
+// readers and writers mutually throttle so read bytes stay at rwMixPct% of
+// the combined volume; headroom = blockSize * peer threads; 600s watchdog) ---
+
+bool Worker::rwBalancerActive() const
+{
+    const auto& cfg = eng.cfg;
+    return cfg.rwMixThreads > 0 && cfg.rwMixPct > 0 && cfg.rwMixPct < 100 &&
+           eng.currentPhase == Phase::WRITE;
+}
+
+void Worker::rwBalanceWait(bool isRead, uint64_t nextLen)
+{
+    const auto& cfg = eng.cfg;
+    const uint64_t pct = (uint64_t)cfg.rwMixPct;
+    const unsigned peers = isRead ? (cfg.numThreads - cfg.rwMixThreads) : cfg.rwMixThreads;
+    const uint64_t headroom = cfg.blockSize * std::max(1u, peers);
+    auto waitStart = Clock::now();
+
+    for (;;) {
+        uint64_t rd = eng.rwBalBytesRead.load(std::memory_order_relaxed);
+        uint64_t wr = eng.rwBalBytesWrite.load(std::memory_order_relaxed);
+        uint64_t allowed;
+        if (isRead) // read <= (write + headroom) * pct / (100 - pct)
+            allowed = (wr + headroom) * pct / (100 - pct);
+        else // write <= (read + headroom) * (100 - pct) / pct
+            allowed = (rd + headroom) * (100 - pct) / pct;
+
+        uint64_t mine = isRead ? rd : wr;
+        if (mine + nextLen <= allowed) return;
+
+        // peers all done? then no more balancing possible — proceed
+        if (eng.workersDone.load(std::memory_order_relaxed) >= (int)peers &&
+            eng.workersDone.load(std::memory_order_relaxed) > 0)
+            return;
+
+        checkInterrupt();
+        if (std::chrono::duration_cast<std::chrono::seconds>(Clock::now() - waitStart)
+                .count() > 600)
+            throw WorkerError("rwmix balancer: stalled >600s waiting for " +
+                              std::string(isRead ? "writer" : "reader") + " progress");
+        std::this_thread::sleep_for(std::chrono::milliseconds(2));
+    }
+}
+
+void Worker::rwBalanceAccount(bool isRead, uint64_t len)
+{
+    (isRead ? eng.rwBalBytesRead : eng.rwBalBytesWrite)
+        .fetch_add(len, std::memory_order_relaxed);
+}
+
 bool Worker::rwMixDecideRead()
 {
     bool doRead = (rwMixReads * 100) < (rwMixOps * (uint64_t)eng.cfg.rwMixPct);
@@ -794,8 +845,14 @@ void Worker::fileModeBlocks(bool isWrite)
 
         // rwmix: dedicated readers always read; otherwise hold reads/total
         // at rwMixPct (reference --rwmixpct / --rwmixthr semantics)
-        bool mixRead = rwMixActive && (isDedicatedReader || rwMixDecideRead());
+        // with dedicated reader threads (--rwmixthr) the pct is a BYTE ratio
+        // held by the balancer; without them it is a per-block read probability
+        bool mixRead = rwMixActive &&
+                       (isDedicatedReader ||
+                        (cfg.rwMixThreads == 0 && rwMixDecideRead()));
         bool blockWrite = isWrite && !mixRead;
+
+        if (rwMixActive && rwBalancerActive()) rwBalanceWait(mixRead, ioLen);
 
         auto t0 = lat ? Clock::now() : Clock::time_point();
 
@@ -817,6 +874,7 @@ void Worker::fileModeBlocks(bool isWrite)
         AtomicLiveOps& ops = mixRead ? liveOpsReadMix : liveOps;
         ops.bytes.fetch_add(ioLen, std::memory_order_relaxed);
         ops.iops.fetch_add(1, std::memory_order_relaxed);
+        if (rwMixActive && rwBalancerActive()) rwBalanceAccount(mixRead, ioLen);
     }
 
     if (isWrite && cfg.fsyncPerFile)
@@ -908,7 +966,9 @@ void Worker::fileModeBlocksUring(bool isWrite)
         uint64_t fileIdx, inFileOff, ioLen;
         if (!mapBlock(spec, fileIdx, inFileOff, ioLen)) return true; // skip this block
 
-        bool mixRead = rwMixActive && (isDedicatedReader || rwMixDecideRead());
+        bool mixRead = rwMixActive &&
+                       (isDedicatedReader ||
+                        (cfg.rwMixThreads == 0 && rwMixDecideRead()));
         bool blockWrite = isWrite && !mixRead;
 
         rateLimiter.wait(ioLen);
@@ -1158,8 +1218,9 @@ void Worker::dirModeFiles(Phase phase)
 
                         BlockSpec spec;
                         while (gen->next(spec)) {
-                            bool mixRead =
-                                rwMixActive && (isDedicatedReader || rwMixDecideRead());
+                            bool mixRead = rwMixActive &&
+                                (isDedicatedReader ||
+                                 (cfg.rwMixThreads == 0 && rwMixDecideRead()));
                             bool blockWrite = isWrite && !mixRead;
                             auto t0 = lat ? Clock::now() : Clock::time_point();
                             ssize_t res = blockIO(blockWrite, fd, 0, spec.len, spec.offset,
@@ -1738,6 +1799,8 @@ void Engine::startPhase(Phase phase)
     if (phaseRunning) throw std::runtime_error("phase already running");
 
     currentPhase = phase;
+    rwBalBytesRead.store(0);
+    rwBalBytesWrite.store(0);
     nbAcceptDone = false;
     nbConns.clear();
     if (!cfg.opsLogPath.empty() && !opsLog.isEnabled())

@@ -184,6 +184,9 @@ private:
     // rwmix decision: keep reads/total at rwMixPct (write phase only)
     bool rwMixDecideRead(); // defined in engine.cpp (needs Engine)
     void applyBinding();
+    bool rwBalancerActive() const;
+    void rwBalanceWait(bool isRead, uint64_t nextLen); // throttle to the byte ratio
+    void rwBalanceAccount(bool isRead, uint64_t len);
     void addEntryLat(uint64_t us)
     {
         entryLat.add(us);
@@ -261,6 +264,12 @@ public:
     std::condition_variable doneCv;
 
     std::vector<std::unique_ptr<Worker>> workers;
+
+    // rwmix byte-ratio balancer state (reference RateLimiterRWMixThreads):
+    // active when both --rwmixthr and --rwmixpct are set; holds read bytes at
+    // rwMixPct% of the combined volume with blockSize*peerThreads headroom
+    std::atomic<uint64_t> rwBalBytesRead{0};
+    std::atomic<uint64_t> rwBalBytesWrite{0};
 
     // resolved at prepare()
     std::vector<uint64_t> resolvedFileSizes; // per path (file/bdev mode)
