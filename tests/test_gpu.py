@@ -113,3 +113,44 @@ def test_gpu_missing_fails_loudly(core, tmp_path):
     eng.wait_phase_done(60_000)
     res = eng.finish_phase()
     assert res[0]["error"], "expected an error for invalid GPU id"
+
+
+def test_gpu_small_block_batched_staging(core, tmp_path):
+    """4K blocks take the batched half-ring staging path (one ranged H2D per
+    64 blocks); accounting and completion must match the generic path."""
+    p = str(tmp_path / "gpu_4k")
+    size = 8 * 1024 * 1024
+    cfg = dict(paths=[p], path_type="file", threads=2, num_dataset_threads=2,
+               file_size=size, block_size=4096, gpu_ids=[0])
+    eng = core.Engine(cfg)
+    eng.prepare()
+    for phase in ("WRITE", "READ"):
+        eng.start_phase(core.PHASES[phase])
+        assert eng.wait_phase_done(120_000)
+        res = eng.finish_phase()
+        errs = [r["error"] for r in res if r["error"]]
+        assert not errs, errs
+        assert sum(r["bytes"] for r in res) == size
+        assert sum(r["iops"] for r in res) == size // 4096
+
+
+def test_gpu_small_block_with_verify_uses_checked_path(core, tmp_path):
+    """verify forces the per-block path (on-GPU checked); corruption is found."""
+    p = str(tmp_path / "gpu_4kv")
+    size = 1024 * 1024
+    cfg = dict(paths=[p], path_type="file", threads=1, num_dataset_threads=1,
+               file_size=size, block_size=4096, gpu_ids=[0], verify_salt=17)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    eng.start_phase(core.PHASES["WRITE"])
+    assert eng.wait_phase_done(120_000)
+    assert not [r["error"] for r in eng.finish_phase() if r["error"]]
+
+    with open(p, "r+b") as f:
+        f.seek(500_000)
+        f.write(b"\xba\xad")
+
+    eng.start_phase(core.PHASES["READ"])
+    eng.wait_phase_done(120_000)
+    errs = [r["error"] for r in eng.finish_phase() if r["error"]]
+    assert errs and "verification failed" in errs[0].lower()
