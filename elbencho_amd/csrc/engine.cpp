@@ -116,9 +116,30 @@ static void mkdirIgnoreExists(const std::string& path)
     if (mkdir(path.c_str(), 0777) && errno != EEXIST) throwErrno("mkdir", path);
 }
 
-// staging ring depth for GPU mode (tunable for experiments). Small blocks
-// get a deeper ring so one H2D can batch a ~512 KiB span of consecutive
-// slots (two batches in flight).
+// Staging geometry for GPU mode. Reads batch `batch` consecutive slots into
+// ONE ranged H2D (pinned and device slots are contiguous): 4 MiB copies only
+// reach ~30 GB/s on the SDMA path, ~16 MiB spans run near the 57 GB/s peak.
+// The ring holds two batches so pread-fill overlaps the copy.
+static uint64_t gpuBatchBytes()
+{
+    static const uint64_t v = [] {
+        const char* e = getenv("EB_GPU_BATCH_BYTES");
+        return e ? (uint64_t)atoll(e) : (16ULL << 20);
+    }();
+    return v;
+}
+
+static int gpuBatchSlots(uint64_t blockSize)
+{
+    uint64_t bb = gpuBatchBytes();
+    if (!bb) return 1; // batching disabled
+    uint64_t batch = bb / std::max<uint64_t>(blockSize, 1);
+    // cap the ring (2 batches) at 64 MiB per worker
+    uint64_t maxBatch = (32ULL << 20) / std::max<uint64_t>(blockSize, 1);
+    batch = std::min(batch, std::max<uint64_t>(maxBatch, 1));
+    return (int)std::min<uint64_t>(std::max<uint64_t>(batch, 1), 64);
+}
+
 static int gpuSlotCount(int ioDepth, uint64_t blockSize)
 {
     static const int def = [] {
@@ -126,11 +147,7 @@ static int gpuSlotCount(int ioDepth, uint64_t blockSize)
         int n = v ? atoi(v) : 2;
         return (n >= 1 && n <= 128) ? n : 2;
     }();
-    int slots = def;
-    if (blockSize < (256ULL << 10)) {
-        uint64_t batch = (256ULL << 10) / std::max<uint64_t>(blockSize, 1);
-        slots = (int)std::min<uint64_t>(2 * batch, 128);
-    }
+    int slots = std::max(def, 2 * gpuBatchSlots(blockSize));
     return std::max(slots, ioDepth);
 }
 
@@ -748,11 +765,11 @@ void Worker::fileModeBlocks(bool isWrite)
             if (cfg.fsyncPerFile)
                 for (size_t i = 0; i < fg.fds.size(); i++)
                     if (fsync(fg.fds[i])) throwErrno("fsync", cfg.paths[i]);
-        } else if (cfg.blockSize < (256ULL << 10) && !doVerify && nSlots >= 4) {
-            // small-block batched read: fill half the ring with preads, then
-            // one ranged H2D covers all of them (slots are contiguous); the
-            // other half stages while this half reads.
-            const int batch = nSlots / 2;
+        } else if (!doVerify && gpuBatchSlots(cfg.blockSize) >= 2 && nSlots >= 4) {
+            // batched read: fill half the ring with preads, then one ranged
+            // H2D covers all of them (slots are contiguous); the other half
+            // stages while this half reads.
+            const int batch = gpuBatchSlots(cfg.blockSize);
             int half = 0; // 0 -> slots [0, batch), 1 -> [batch, 2*batch)
             bool halfBusy[2] = {false, false};
 
