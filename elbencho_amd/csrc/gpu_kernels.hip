@@ -294,8 +294,12 @@ GpuCtx::GpuCtx(int deviceId, int numSlots, uint64_t bufSize, bool pinnedHostBufs
     impl->slotStride = (bufSize + 4095) & ~4095ULL; // keep O_DIRECT alignment
     HIP_CHECK(hipMalloc(&impl->devBase, impl->slotStride * numSlots));
     if (pinnedHostBufs) {
-        HIP_CHECK(hipHostMalloc(&impl->hostBase, impl->slotStride * numSlots,
-                                hipHostMallocDefault));
+        // EB_GPU_HOSTALLOC=nc: non-coherent pinned pages (device-optimized)
+        static const unsigned hostFlags = [] {
+            const char* v = getenv("EB_GPU_HOSTALLOC");
+            return (v && v[0] == 'n') ? hipHostMallocNonCoherent : hipHostMallocDefault;
+        }();
+        HIP_CHECK(hipHostMalloc(&impl->hostBase, impl->slotStride * numSlots, hostFlags));
     } else {
         if (posix_memalign((void**)&impl->hostBase, 4096, impl->slotStride * numSlots))
             throw std::runtime_error("host buffer alloc failed");
