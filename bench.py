@@ -73,8 +73,9 @@ def main() -> int:
     if use_gpu:
         import torch
 
-        torch.cuda.set_device(local_rank)
-        device = torch.device("cuda", local_rank)
+        dev_idx = local_rank % max(torch.cuda.device_count(), 1)
+        torch.cuda.set_device(dev_idx)
+        device = torch.device("cuda", dev_idx)
 
     # --- per-rank dataset on tmpfs ---
     os.makedirs(args.dir, exist_ok=True)
@@ -113,7 +114,7 @@ def main() -> int:
     # measured engine: GPU-staged when a GPU is present
     mcfg = dict(base_cfg, paths=[path])
     if use_gpu:
-        mcfg["gpu_ids"] = [local_rank]
+        mcfg["gpu_ids"] = [local_rank % max(core.gpu_device_count(), 1)]
     if args.workload == "randread":
         mcfg["random"] = True
         mcfg["block_size"] = 4096

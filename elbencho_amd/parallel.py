@@ -30,15 +30,17 @@ def init_from_env(device: Optional[torch.device] = None) -> "PhaseSync | None":
     if "RANK" not in os.environ or "WORLD_SIZE" not in os.environ:
         return None
     if not dist.is_initialized():
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get(
+            "EB_DIST_BACKEND", "nccl" if torch.cuda.is_available() else "gloo")
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29511")
         dist.init_process_group(
             backend=backend,
             timeout=datetime.timedelta(seconds=int(os.environ.get("EB_DIST_TIMEOUT", "600"))),
         )
-    if device is None and torch.cuda.is_available():
+    if device is None and torch.cuda.is_available() and dist.get_backend() == "nccl":
         local_rank = int(os.environ.get("LOCAL_RANK", dist.get_rank()))
+        local_rank %= max(torch.cuda.device_count(), 1)
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
     return PhaseSync(device)
