@@ -266,10 +266,7 @@ class Coordinator:
             if want_worker_rows:
                 getter = getattr(self.runner, "poll_workers", None)
                 rows = getter() if getter else None
-            try:
-                live.update(p, rows)
-            except TypeError:
-                live.update(p)
+            live.update(p, rows)
             if live_csv:
                 if isinstance(live_csv, LiveCsvExWriter):
                     live_csv.update(p, rows or [])
