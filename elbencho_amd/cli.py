@@ -299,6 +299,14 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--s3credfile", default="", metavar="PATH",
                    help="File with one key:secret credential pair per line, "
                         "round-robined across workers.")
+    g.add_argument("--s3nompucompl", action="store_true",
+                   help="Do not complete multipart uploads in the write phase "
+                        "(complete them later via --s3mpucompl, possibly from "
+                        "another instance).")
+    g.add_argument("--s3mpucompl", action="store_true",
+                   help="Phase: complete multipart uploads left open by an earlier "
+                        "--s3nompucompl run (uploadIds and part ETags are "
+                        "rediscovered from the S3 endpoint).")
     g.add_argument("--s3credlist", default="", metavar="LIST",
                    help="Comma-separated key:secret pairs, round-robined across workers.")
     g.add_argument("--s3randobj", action="store_true",
@@ -485,6 +493,8 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.s3_btag_verify = args.s3btagverify
     cfg.s3_cred_file = args.s3credfile
     cfg.s3_cred_list = args.s3credlist
+    cfg.s3_no_mpu_compl = args.s3nompucompl
+    cfg.s3_mpu_complete = args.s3mpucompl
 
     cfg.config_file = args.configfile
 

@@ -223,6 +223,8 @@ class BenchConfig:
     s3_btag_verify: bool = False   # --s3btagverify
     s3_cred_file: str = ""         # --s3credfile (lines "key:secret")
     s3_cred_list: str = ""         # --s3credlist ("key:secret,key:secret")
+    s3_no_mpu_compl: bool = False  # --s3nompucompl (leave multipart uploads open)
+    s3_mpu_complete: bool = False  # run the S3MPUCOMPLETE phase
 
     # --- misc ---
     config_file: str = ""          # -c/--configfile
@@ -368,6 +370,7 @@ class BenchConfig:
                 ("PUTBTAG", self.s3_btag),
                 ("GETBTAG", self.s3_btag and self.s3_btag_verify),
                 ("WRITE", self.run_write),         # PUT objects
+                ("S3MPUCOMPLETE", self.s3_mpu_complete),
                 ("PUTOBJACL", self.s3_acl_put),
                 ("PUTOTAG", self.s3_otag),
                 ("STAT", self.run_stat),           # HEAD objects

@@ -217,7 +217,8 @@ class Coordinator:
                   "GETOBJACL": "GETOBJACL", "PUTBACL": "PUTBACL",
                   "GETBACL": "GETBACL", "PUTOTAG": "PUTOBJMD",
                   "GETOTAG": "GETOBJMD", "DELOTAG": "DELOBJMD",
-                  "PUTBTAG": "PUTBUCKETMD", "GETBTAG": "GETBUCKETMD"}
+                  "PUTBTAG": "PUTBUCKETMD", "GETBTAG": "GETBUCKETMD",
+                  "S3MPUCOMPLETE": "MPUCOMPL"}
 
     def run_phase(self, name: str, quiet: bool = False) -> bool:
         cfg = self.cfg

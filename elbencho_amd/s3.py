@@ -283,6 +283,27 @@ class S3Client:
                                        query={"uploadId": upload_id}, body=body)
         self._check(status, data, f"complete multipart {bucket}/{key}")
 
+    def list_multipart_uploads(self, bucket: str,
+                               prefix: str = "") -> list[tuple[str, str]]:
+        """[(key, uploadId)] of in-progress multipart uploads."""
+        q = {"uploads": ""}
+        if prefix:
+            q["prefix"] = prefix
+        status, data, _ = self.request("GET", f"/{bucket}", query=q)
+        self._check(status, data, f"list multipart uploads {bucket}")
+        import re as _re
+        return _re.findall(r"<Key>([^<]*)</Key><UploadId>([^<]*)</UploadId>",
+                           data.decode())
+
+    def list_parts(self, bucket: str, key: str, upload_id: str) -> list[tuple[int, str]]:
+        status, data, _ = self.request("GET", f"/{bucket}/{key}",
+                                       query={"uploadId": upload_id})
+        self._check(status, data, f"list parts {bucket}/{key}")
+        import re as _re
+        return [(int(n), e) for n, e in
+                _re.findall(r"<PartNumber>(\d+)</PartNumber><ETag>([^<]*)</ETag>",
+                            data.decode())]
+
     def abort_multipart(self, bucket: str, key: str, upload_id: str):
         status, data, _ = self.request("DELETE", f"/{bucket}/{key}",
                                        query={"uploadId": upload_id})
@@ -501,6 +522,21 @@ class S3Worker(threading.Thread):
                     if cfg.s3_btag_verify and tags.get("elbencho-amd") != "bucket":
                         raise S3Error(f"bucket tagging verification failed for {b}")
                     self.ops.entries += 1
+        elif ph == "S3MPUCOMPLETE":
+            # complete multipart uploads left open by an earlier --s3nompucompl
+            # run — possibly by ANOTHER instance: uploadIds and part ETags are
+            # rediscovered from S3 itself (reference S3MPUCOMPLETE phase +
+            # shared MPU store, Common.h:196 / S3UploadStore)
+            for name in self._object_names():
+                self._check_interrupt()
+                bucket = self._bucket()
+                uploads = [u for k, u in
+                           self.client.list_multipart_uploads(bucket, prefix=name)
+                           if k == name]
+                for upload_id in uploads:
+                    parts = self.client.list_parts(bucket, name, upload_id)
+                    self.client.complete_multipart(bucket, name, upload_id, parts)
+                    self.ops.entries += 1
         else:
             raise S3Error(f"S3 phase not supported: {ph}")
 
@@ -543,7 +579,9 @@ class S3Worker(threading.Thread):
                         self.ops.iops += 1
                         off += ln
                         part_num += 1
-                    self.client.complete_multipart(bucket, name, upload_id, parts)
+                    if not cfg.s3_no_mpu_compl:
+                        self.client.complete_multipart(bucket, name, upload_id, parts)
+                    # else: left open for a later S3MPUCOMPLETE phase
                 except BaseException:
                     self.client.abort_multipart(bucket, name, upload_id)
                     raise

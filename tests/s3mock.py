@@ -157,6 +157,22 @@ class S3Handler(BaseHTTPRequestHandler):
             if "tagging" in q:
                 return self._send(200, st.tags.get((bucket, key),
                                                    _xml("<Tagging><TagSet></TagSet></Tagging>")))
+            if "uploads" in q and key is None:  # list multipart uploads
+                ups = "".join(
+                    f"<Upload><Key>{k}</Key><UploadId>{uid}</UploadId></Upload>"
+                    for uid, (b, k) in st.upload_meta.items() if b == bucket
+                    and k.startswith(q.get("prefix", "")))
+                return self._send(200, _xml(
+                    f"<ListMultipartUploadsResult>{ups}</ListMultipartUploadsResult>"))
+            if "uploadId" in q and key is not None:  # list parts
+                up = st.uploads.get(q["uploadId"])
+                if up is None:
+                    return self._err(404, "NoSuchUpload")
+                parts = "".join(
+                    f"<Part><PartNumber>{n}</PartNumber>"
+                    f"<ETag>\"{hashlib.md5(d).hexdigest()}\"</ETag></Part>"
+                    for n, d in sorted(up.items()))
+                return self._send(200, _xml(f"<ListPartsResult>{parts}</ListPartsResult>"))
             if bucket not in st.buckets:
                 return self._err(404, "NoSuchBucket")
             if key is None:  # list objects v2

@@ -154,3 +154,29 @@ def test_s3_credentials_roundrobin(mock_s3):
                "-d", "-w", "-F", "-D", "-t", "2", "-N", "2", "-s", "4k", "-b", "4k",
                "s3://credbkt"])
     assert rc == 0
+
+
+def test_s3_mpu_sharing_cross_instance(mock_s3):
+    """--s3nompucompl leaves multipart uploads open; a separate run (another
+    'instance') completes them via --s3mpucompl by rediscovering uploadIds
+    and part ETags from the endpoint."""
+    rc = _cli(mock_s3, ["-d", "-w", "-t", "2", "-N", "2", "-s", "192k", "-b", "64k",
+                        "--s3nompucompl", "--verify", "9", "s3://mpubkt"])
+    assert rc == 0
+    # objects do not exist yet (uploads incomplete)
+    from tests.s3mock import S3Handler
+    with S3Handler.store.lock:
+        assert S3Handler.store.buckets["mpubkt"] == {}
+        assert len(S3Handler.store.uploads) == 4  # 2 threads x 2 objects
+
+    rc = _cli(mock_s3, ["--s3mpucompl", "-t", "2", "-N", "2", "-s", "192k",
+                        "-b", "64k", "s3://mpubkt"])
+    assert rc == 0
+    with S3Handler.store.lock:
+        assert len(S3Handler.store.buckets["mpubkt"]) == 4
+        assert not S3Handler.store.uploads
+
+    # the completed objects verify end to end
+    rc = _cli(mock_s3, ["-r", "-t", "2", "-N", "2", "-s", "192k", "-b", "64k",
+                        "--verify", "9", "s3://mpubkt"])
+    assert rc == 0
