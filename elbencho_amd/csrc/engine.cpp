@@ -627,6 +627,10 @@ void Worker::fileModeBlocks(bool isWrite)
 {
     const auto& cfg = eng.cfg;
     if (cfg.ioDepth > 1) return fileModeBlocksUring(isWrite);
+    if (cfg.useMmap && !cfg.gpuIDs.empty() && !cfg.measureLat && cfg.flockMode == 0 &&
+        !(isWrite && (cfg.rwMixPct > 0 || cfg.rwMixThreads > 0)) && !cfg.verifyDirect &&
+        cfg.pathType == PathType::FILE)
+        return fileModeBlocksGpuMmap(isWrite);
 
     const uint64_t fileSize = eng.effFileSize;
     const uint64_t bs = cfg.blockSize;
@@ -901,6 +905,106 @@ void Worker::fileModeBlocks(bool isWrite)
     if (isWrite && cfg.fsyncPerFile)
         for (size_t i = 0; i < fg.fds.size(); i++)
             if (fsync(fg.fds[i])) throwErrno("fsync", cfg.paths[i]);
+}
+
+// mmap + GPU zero-copy: every block is ONE hipMemcpyAsync between pinned
+// page-cache pages and the worker's HBM slot ring — no pread/pwrite, no host
+// bounce buffer. The MI355X-native answer to cuFile for cache-resident data.
+void Worker::fileModeBlocksGpuMmap(bool isWrite)
+{
+    const auto& cfg = eng.cfg;
+    const uint64_t fileSize = eng.effFileSize;
+    const uint64_t bs = cfg.blockSize;
+    const size_t numFiles = cfg.paths.size();
+    const uint64_t numBlocksPerFile = (fileSize + bs - 1) / bs;
+    const bool doVerify = cfg.verifySalt >= 0;
+
+    // registered mappings (cached in the engine across phases)
+    std::vector<char*> bases(numFiles);
+    for (size_t i = 0; i < numFiles; i++)
+        bases[i] = eng.getMappedReg(cfg.paths[i], fileSize, isWrite).base;
+
+    const uint64_t mapBPF = (cfg.random || cfg.strided) ? (fileSize / bs) : numBlocksPerFile;
+    const uint64_t virtFileLen = mapBPF * bs;
+
+    std::unique_ptr<OffsetGen> gen;
+    if (cfg.random || cfg.strided) {
+        uint64_t numBlocksTotal = mapBPF * numFiles;
+        uint64_t rangeLen = bs * (numBlocksTotal / cfg.numDataSetThreads);
+        uint64_t rangeOff = (uint64_t)globalRank * rangeLen;
+        if (cfg.strided) {
+            gen = std::make_unique<OffsetGenStrided>(bs, globalRank, cfg.numDataSetThreads);
+            gen->reset(0, bs * numBlocksTotal);
+        } else {
+            gen = makeOffsetGen(rangeOff, rangeLen);
+        }
+    } else {
+        uint64_t myStart, myLen;
+        fairShareSlice(virtFileLen * numFiles, myStart, myLen);
+        if (!myLen) return;
+        gen = makeOffsetGen(myStart, myLen);
+    }
+
+    const int nSlots = (int)hostBufs.size();
+    constexpr uint64_t VERIFY_FETCH_INTERVAL = 64;
+    uint64_t sinceFetch = 0;
+    BlockSpec spec;
+    uint64_t opCount = 0;
+    int slot = 0;
+
+    auto fetchVerify = [&]() {
+        GpuVerifyResult r = gpu->fetchVerifyResult();
+        if (r.numMismatches)
+            throw WorkerError("Data verification failed (GPU). First bad file offset: " +
+                              std::to_string(r.firstBadFileOffset) +
+                              "; mismatching 8-byte words: " +
+                              std::to_string(r.numMismatches));
+    };
+
+    while (gen->next(spec)) {
+        if ((opCount++ % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+
+        uint64_t fileIdx, inFileOff, ioLen;
+        if (!virtFileLen) break;
+        fileIdx = spec.offset / virtFileLen;
+        if (fileIdx >= numFiles) continue;
+        inFileOff = spec.offset - fileIdx * virtFileLen;
+        if (inFileOff >= fileSize) continue;
+        ioLen = std::min(spec.len, fileSize - inFileOff);
+
+        rateLimiter.wait(ioLen);
+
+        if (isWrite) {
+            // fill the HBM slot, then DMA into the mapped file pages
+            preWriteFill(slot, ioLen, inFileOff);
+            gpu->copyToHostAsync(slot, bases[fileIdx] + inFileOff, ioLen);
+        } else {
+            gpu->copyFromHostAsync(slot, bases[fileIdx] + inFileOff, ioLen);
+            if (doVerify && (inFileOff % 8 == 0) && (ioLen % 16 == 0)) {
+                gpu->verifyChecksumDevAsync(slot, ioLen, inFileOff,
+                                            (uint64_t)cfg.verifySalt);
+                if (++sinceFetch >= VERIFY_FETCH_INTERVAL) {
+                    fetchVerify();
+                    sinceFetch = 0;
+                }
+            } else if (doVerify) { // odd tail: check the mapped pages directly
+                uint64_t bad = verifyChecksumCPU(bases[fileIdx] + inFileOff, ioLen,
+                                                 inFileOff, (uint64_t)cfg.verifySalt);
+                if (bad != UINT64_MAX)
+                    throw WorkerError("Data verification failed. First bad file offset: " +
+                                      std::to_string(bad));
+            }
+        }
+
+        liveOps.bytes.fetch_add(ioLen, std::memory_order_relaxed);
+        liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+        slot = (slot + 1) % nSlots;
+
+        if ((opCount & 127) == 0) gpu->syncStream(); // bound the async queue
+    }
+
+    gpu->syncStream();
+    if (doVerify && !isWrite) fetchVerify();
 }
 
 void Worker::fileModeBlocksUring(bool isWrite)
@@ -1774,6 +1878,7 @@ Engine::~Engine()
     interrupt();
     for (auto& t : threads)
         if (t.joinable()) t.join();
+    dropMappedRegs();
 }
 
 void Engine::prepare()
@@ -1959,6 +2064,49 @@ std::vector<WorkerResult> Engine::finishPhase()
     workers.clear();
     currentPhase = Phase::IDLE;
     return results;
+}
+
+// mmap+GPU zero-copy: map the file and pin its pages so SDMA copies move
+// data straight between the page cache and HBM3E (no bounce buffer, no
+// read/write syscalls). Mapping+registration cached across phases.
+Engine::MappedReg& Engine::getMappedReg(const std::string& path, uint64_t len, bool writable)
+{
+    std::lock_guard<std::mutex> lk(mmapRegMtx);
+    auto& reg = mmapRegCache[path];
+    if (reg.base && reg.len >= len) return reg;
+    if (reg.base) { // grow: drop the old mapping first
+        if (reg.registered) gpuHostUnregister(reg.base);
+        munmap(reg.base, reg.len);
+        reg = MappedReg{};
+    }
+
+    int fd = open(path.c_str(), writable ? (O_RDWR | O_CREAT) : O_RDONLY, 0644);
+    if (fd < 0) throwErrno("open", path);
+    if (writable && ftruncate(fd, len)) {
+        close(fd);
+        throwErrno("truncate-to-size", path);
+    }
+    int prot = PROT_READ | (writable ? PROT_WRITE : 0);
+    void* p = mmap(nullptr, len, prot, MAP_SHARED | MAP_POPULATE, fd, 0);
+    close(fd); // mapping keeps the file alive
+    if (p == MAP_FAILED)
+        throw WorkerError("mmap failed. Path: " + path + "; SysErr: " + strerror(errno));
+
+    reg.base = (char*)p;
+    reg.len = len;
+    gpuHostRegister(reg.base, len); // pin for full-speed DMA
+    reg.registered = true;
+    return reg;
+}
+
+void Engine::dropMappedRegs()
+{
+    std::lock_guard<std::mutex> lk(mmapRegMtx);
+    for (auto& [path, reg] : mmapRegCache) {
+        if (reg.registered) gpuHostUnregister(reg.base);
+        if (reg.base) munmap(reg.base, reg.len);
+    }
+    mmapRegCache.clear();
 }
 
 std::pair<uint64_t, uint64_t> Engine::plannedWork(Phase phase) const

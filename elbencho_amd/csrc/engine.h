@@ -14,6 +14,7 @@
 
 #include <chrono>
 #include <condition_variable>
+#include <map>
 #include <memory>
 #include <mutex>
 #include <string>
@@ -161,6 +162,7 @@ private:
 
     // phase bodies
     void fileModeBlocks(bool isWrite);
+    void fileModeBlocksGpuMmap(bool isWrite);
     void fileModeBlocksUring(bool isWrite);
     void fileModeDelete();
     void fileModeStat();
@@ -281,6 +283,18 @@ public:
     std::vector<std::unique_ptr<GpuCtx>> gpuCtxCache;
 
     OpsLogger opsLog;
+
+    // mmap+GPU zero-copy path: per-path registered file mappings, cached
+    // across phases (hipHostRegister of multi-GiB regions is expensive)
+    struct MappedReg {
+        char* base = nullptr;
+        uint64_t len = 0;
+        bool registered = false;
+    };
+    std::mutex mmapRegMtx;
+    std::map<std::string, MappedReg> mmapRegCache;
+    MappedReg& getMappedReg(const std::string& path, uint64_t len, bool writable);
+    void dropMappedRegs();
 
     // netbench server: worker 0 accepts all connections, peers take a subset
     std::mutex nbMtx;

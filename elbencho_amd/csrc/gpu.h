@@ -27,6 +27,15 @@ struct GpuVerifyResult {
     uint64_t firstBadFileOffset; // UINT64_MAX if none
 };
 
+// Pin (hipHostRegister) / unpin an arbitrary host region — used to register
+// mmap'ed file pages so storage<->HBM copies DMA straight from/to the page
+// cache with no bounce buffer (the MI355X-native --mmap + GPU path).
+void gpuHostRegister(void* ptr, uint64_t len);
+void gpuHostUnregister(void* ptr);
+
+// Copy between a registered host region and a device slot buffer.
+class GpuCtx;
+
 // Returns number of visible HIP devices; 0 when no GPU or no driver.
 int gpuDeviceCount();
 
@@ -52,6 +61,9 @@ public:
     void bindThread();
 
     // --- async staging copies on this worker's stream ---
+    // direct copies between a registered host region (mmap'ed file) and HBM
+    void copyFromHostAsync(int slot, const void* src, uint64_t len);
+    void copyToHostAsync(int slot, void* dst, uint64_t len);
     void copyH2DAsync(int slot, uint64_t len);
     // one memcpy covering `count` consecutive slots (slot stride is the
     // 4 KiB-rounded slot size) — batches small-block staging

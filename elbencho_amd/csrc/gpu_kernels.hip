@@ -201,6 +201,13 @@ __global__ __launch_bounds__(256) void ebBlockVarKernel(ulonglong2* __restrict__
 // host wrappers
 // ---------------------------------------------------------------------------
 
+void gpuHostRegister(void* ptr, uint64_t len)
+{
+    HIP_CHECK(hipHostRegister(ptr, len, hipHostRegisterDefault));
+}
+
+void gpuHostUnregister(void* ptr) { (void)hipHostUnregister(ptr); }
+
 int gpuDeviceCount()
 {
     int n = 0;
@@ -352,6 +359,18 @@ GpuCtx::~GpuCtx()
 char* GpuCtx::hostBuf(int slot) const { return impl->hostBufs[slot]; }
 
 void GpuCtx::bindThread() { HIP_CHECK(hipSetDevice(devId)); }
+
+void GpuCtx::copyFromHostAsync(int slot, const void* src, uint64_t len)
+{
+    HIP_CHECK(hipMemcpyAsync(impl->devBufs[slot], src, len, hipMemcpyHostToDevice,
+                             impl->stream));
+}
+
+void GpuCtx::copyToHostAsync(int slot, void* dst, uint64_t len)
+{
+    HIP_CHECK(hipMemcpyAsync(dst, impl->devBufs[slot], len, hipMemcpyDeviceToHost,
+                             impl->stream));
+}
 
 void GpuCtx::copyH2DAsync(int slot, uint64_t len)
 {
