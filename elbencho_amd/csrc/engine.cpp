@@ -2073,8 +2073,8 @@ Engine::MappedReg& Engine::getMappedReg(const std::string& path, uint64_t len, b
 {
     std::lock_guard<std::mutex> lk(mmapRegMtx);
     auto& reg = mmapRegCache[path];
-    if (reg.base && reg.len >= len) return reg;
-    if (reg.base) { // grow: drop the old mapping first
+    if (reg.base && reg.len >= len && (reg.writable || !writable)) return reg;
+    if (reg.base) { // grow or upgrade to writable: drop the old mapping first
         if (reg.registered) gpuHostUnregister(reg.base);
         munmap(reg.base, reg.len);
         reg = MappedReg{};
@@ -2094,6 +2094,7 @@ Engine::MappedReg& Engine::getMappedReg(const std::string& path, uint64_t len, b
 
     reg.base = (char*)p;
     reg.len = len;
+    reg.writable = writable;
     gpuHostRegister(reg.base, len); // pin for full-speed DMA
     reg.registered = true;
     return reg;

@@ -290,6 +290,7 @@ public:
         char* base = nullptr;
         uint64_t len = 0;
         bool registered = false;
+        bool writable = false;
     };
     std::mutex mmapRegMtx;
     std::map<std::string, MappedReg> mmapRegCache;
