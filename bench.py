@@ -113,8 +113,13 @@ def main() -> int:
 
     # measured engine: GPU-staged when a GPU is present
     mcfg = dict(base_cfg, paths=[path])
+    use_mmap = os.environ.get("EB_BENCH_MMAP", "1") != "0"
     if use_gpu:
         mcfg["gpu_ids"] = [local_rank % max(core.gpu_device_count(), 1)]
+        # zero-copy: file pages pinned, each block is one hipMemcpyAsync
+        # between the page cache and HBM (fastest seq path; see profiles/)
+        if use_mmap and args.workload in ("seqread", "seqwrite"):
+            mcfg["mmap"] = True
     if args.workload == "randread":
         mcfg["random"] = True
         mcfg["block_size"] = 4096
@@ -183,6 +188,8 @@ def main() -> int:
                 "iodepth": args.iodepth,
                 "bench_dir": args.dir,
                 "gpu_staged": use_gpu,
+                "mmap_zero_copy": bool(use_gpu and use_mmap
+                                       and args.workload != "randread"),
                 "parallelism": f"dp{world}" if world > 1 else "single",
             },
         }
