@@ -407,6 +407,20 @@ class S3Worker(threading.Thread):
         if self.r.interrupt_flag.is_set():
             raise KeyboardInterrupt
 
+    def _oplog(self, op: str, entry: str, offset: int, length: int, pre: bool,
+               err: bool = False):
+        log = self.r.ops_log
+        if not log:
+            return
+        import json as _json
+        line = _json.dumps({"rank": self.rank, "op": op, "entry": entry,
+                            "offset": offset, "len": length,
+                            "type": "pre" if pre else "post",
+                            **({"error": True} if err else {})})
+        with self.r.ops_log_lock:
+            log.write(line + "\n")
+            log.flush()
+
     def run(self):
         try:
             self.r.start_gate.wait()
@@ -557,7 +571,9 @@ class S3Worker(threading.Thread):
             bucket = self._bucket()
             if size <= bs:  # single part
                 t0 = time.monotonic()
+                self._oplog("PutObject", name, 0, size, True)
                 self.client.put_object(bucket, name, self._make_block(size, 0))
+                self._oplog("PutObject", name, 0, size, False)
                 self.io_lat.vec = _add_lat(self.io_lat, t0)
                 self.ops.bytes += size
                 self.ops.iops += 1
@@ -602,7 +618,9 @@ class S3Worker(threading.Thread):
             while off < size:
                 ln = min(bs, size - off)
                 t0 = time.monotonic()
+                self._oplog("GetObject", name, off, ln, True)
                 data = self.client.get_object(bucket, name, (off, off + ln - 1))
+                self._oplog("GetObject", name, off, ln, False)
                 self.io_lat.vec = _add_lat(self.io_lat, t0)
                 if len(data) != ln:
                     raise S3Error(f"short ranged read of {name}: {len(data)} != {ln}")
@@ -725,6 +743,10 @@ class S3Runner:
                     if ln and ":" in ln and not ln.startswith("#"):
                         k, sec = ln.split(":", 1)
                         self.credentials.append((k, sec))
+        self.ops_log = None
+        if cfg.ops_log_path:
+            self.ops_log_lock = threading.Lock()
+            self.ops_log = open(cfg.ops_log_path, "a")
         self.workers: list[S3Worker] = []
         self.interrupt_flag = threading.Event()
         self.start_gate = threading.Event()

@@ -180,3 +180,14 @@ def test_s3_mpu_sharing_cross_instance(mock_s3):
     rc = _cli(mock_s3, ["-r", "-t", "2", "-N", "2", "-s", "192k", "-b", "64k",
                         "--verify", "9", "s3://mpubkt"])
     assert rc == 0
+
+
+def test_s3_opslog(mock_s3, tmp_path):
+    log = tmp_path / "s3ops.jsonl"
+    rc = _cli(mock_s3, ["-d", "-w", "-r", "-t", "1", "-N", "2", "-s", "8k", "-b", "8k",
+                        "--opslog", str(log), "s3://opsbkt"])
+    assert rc == 0
+    import json as _json
+    lines = [_json.loads(ln) for ln in log.read_text().splitlines()]
+    assert any(l["op"] == "PutObject" and l["type"] == "pre" for l in lines)
+    assert any(l["op"] == "GetObject" and l["type"] == "post" for l in lines)
