@@ -39,6 +39,9 @@ class GpuCtx;
 // Returns number of visible HIP devices; 0 when no GPU or no driver.
 int gpuDeviceCount();
 
+// Diagnostic: the hipGetDeviceCount error string (or "ok: N device(s)").
+std::string gpuProbeError();
+
 // Human-readable device name (empty if unavailable).
 std::string gpuDeviceName(int deviceId);
 

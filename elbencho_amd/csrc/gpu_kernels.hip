@@ -216,6 +216,14 @@ int gpuDeviceCount()
     return n;
 }
 
+std::string gpuProbeError()
+{
+    int n = 0;
+    hipError_t e = hipGetDeviceCount(&n);
+    if (e != hipSuccess) return hipGetErrorString(e);
+    return "ok: " + std::to_string(n) + " device(s)";
+}
+
 std::string gpuDeviceName(int deviceId)
 {
     hipDeviceProp_t prop;

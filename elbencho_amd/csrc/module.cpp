@@ -179,6 +179,7 @@ PYBIND11_MODULE(_core, m)
     m.doc() = "elbencho_amd native I/O engine (MI355X / gfx950)";
 
     m.def("gpu_device_count", &gpuDeviceCount);
+    m.def("gpu_probe_error", &gpuProbeError);
     m.def("gpu_device_name", &gpuDeviceName);
 
     m.def("hist_bucket_lower_bound", &LatencyHistogram::bucketLowerBound);
