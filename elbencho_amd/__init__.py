@@ -20,7 +20,18 @@ HTTP_PROTOCOL_VERSION = "1.0.0"  # master<->service wire compatibility
 
 
 def load_core():
-    """Import the native engine, building it in-tree if necessary."""
+    """Import the native engine, building it in-tree if necessary.
+
+    torch is imported first when present: PyTorch bundles its own HIP
+    runtime, and loading _core's /opt/rocm runtime before torch's leaves
+    hipGetDeviceCount reporting "no ROCm-capable device" once torch loads
+    (two HSA runtimes in one process). torch-first keeps a single runtime.
+    """
+    try:
+        import torch  # noqa: F401
+    except ImportError:
+        pass
+
     try:
         from elbencho_amd import _core  # type: ignore
         return _core
