@@ -293,6 +293,54 @@ PYBIND11_MODULE(_core, m)
         return d;
     }, py::arg("len") = (uint64_t)256 << 20, py::arg("iters") = 10, py::arg("dev") = 0);
 
+    // Persistent GPU buffer ops for Python-side engines (S3): keeps one
+    // GpuCtx alive so per-object verify/fill costs one H2D + kernel, not a
+    // context setup. Used for the "on-GPU verify" S3 path (BASELINE config 5).
+    py::class_<GpuCtx>(m, "GpuBufferOps")
+        .def(py::init([](int dev, uint64_t bufSize) {
+                 return std::make_unique<GpuCtx>(dev, 1, bufSize, true);
+             }),
+             py::arg("dev") = 0, py::arg("buf_size") = (uint64_t)64 << 20)
+        .def("verify",
+             [](GpuCtx& ctx, py::bytes data, uint64_t fileOff, uint64_t salt) {
+                 std::string buf = data;
+                 if (buf.size() > ctx.bufSize())
+                     throw std::runtime_error("GpuBufferOps: data larger than buffer");
+                 {
+                     py::gil_scoped_release rel;
+                     std::memcpy(ctx.hostBuf(0), buf.data(), buf.size());
+                     ctx.copyH2DAsync(0, buf.size());
+                     ctx.syncStream();
+                 }
+                 GpuVerifyResult r = ctx.verifyChecksumDev(0, buf.size(), fileOff, salt);
+                 return py::make_tuple(r.numMismatches, r.firstBadFileOffset);
+             },
+             py::arg("data"), py::arg("file_off"), py::arg("salt"))
+        .def("fill_checksum",
+             [](GpuCtx& ctx, uint64_t len, uint64_t fileOff, uint64_t salt) {
+                 if (len > ctx.bufSize())
+                     throw std::runtime_error("GpuBufferOps: len larger than buffer");
+                 py::gil_scoped_release rel;
+                 ctx.fillChecksumDev(0, len, fileOff, salt);
+                 ctx.copyD2HAsync(0, len);
+                 ctx.syncStream();
+                 py::gil_scoped_acquire acq;
+                 return py::bytes(ctx.hostBuf(0), len);
+             },
+             py::arg("len"), py::arg("file_off"), py::arg("salt"))
+        .def("fill_rand",
+             [](GpuCtx& ctx, uint64_t len, uint64_t seed) {
+                 if (len > ctx.bufSize())
+                     throw std::runtime_error("GpuBufferOps: len larger than buffer");
+                 py::gil_scoped_release rel;
+                 ctx.fillRandDev(0, len, seed);
+                 ctx.copyD2HAsync(0, len);
+                 ctx.syncStream();
+                 py::gil_scoped_acquire acq;
+                 return py::bytes(ctx.hostBuf(0), len);
+             },
+             py::arg("len"), py::arg("seed"));
+
     py::class_<Engine>(m, "Engine")
         .def(py::init([](const py::dict& cfg) {
             return std::make_unique<Engine>(configFromDict(cfg));

@@ -387,6 +387,14 @@ class S3Worker(threading.Thread):
             key, secret = runner.credentials[self.rank % len(runner.credentials)]
         self.client = S3Client(ep, key, secret, cfg.s3_region)
         self.core = load_core()
+        # --gpuids + --verify: verify/generate object data with the gfx950
+        # kernels in HBM (BASELINE config 5); one persistent context per worker
+        self.gpu = None
+        if cfg.gpu_ids and cfg.verify >= 0 and self.core.gpu_device_count() > 0:
+            dev = cfg.gpu_ids[self.rank % len(cfg.gpu_ids)]
+            self.gpu = self.core.GpuBufferOps(dev, cfg.block_size)
+        elif cfg.gpu_ids and self.core.gpu_device_count() <= 0:
+            raise S3Error("GPU requested (gpuids) but no HIP device is available")
 
     # --- object name layout mirrors dir mode: r{rank}/d{dir}/r{rank}-f{file} ---
     def _object_names(self):
@@ -558,6 +566,8 @@ class S3Worker(threading.Thread):
     def _make_block(self, length: int, obj_off: int) -> bytes:
         cfg = self.r.cfg
         if cfg.verify >= 0:
+            if self.gpu:  # generate the pattern in HBM with the fill kernel
+                return self.gpu.fill_checksum(length, obj_off, cfg.verify)
             return self.core.fill_checksum(length, obj_off, cfg.verify)
         return bytes(self.r.rand_block[:length])
 
@@ -625,10 +635,16 @@ class S3Worker(threading.Thread):
                 if len(data) != ln:
                     raise S3Error(f"short ranged read of {name}: {len(data)} != {ln}")
                 if cfg.verify >= 0 and not cfg.s3_fastget:
-                    bad = self.core.verify_checksum(data, off, cfg.verify)
-                    if bad != 2**64 - 1:
-                        raise S3Error(f"S3 data verification failed for {name} at "
-                                      f"object offset {bad}")
+                    if self.gpu and off % 8 == 0 and ln % 16 == 0:
+                        nbad, first = self.gpu.verify(data, off, cfg.verify)
+                        if nbad:
+                            raise S3Error(f"S3 data verification failed (GPU) for "
+                                          f"{name} at object offset {first}")
+                    else:
+                        bad = self.core.verify_checksum(data, off, cfg.verify)
+                        if bad != 2**64 - 1:
+                            raise S3Error(f"S3 data verification failed for {name} at "
+                                          f"object offset {bad}")
                 self.ops.bytes += ln
                 self.ops.iops += 1
                 off += ln

@@ -176,3 +176,34 @@ def test_gpu_mmap_zero_copy_roundtrip(core, tmp_path):
     # file content carries the GPU-written checksum pattern
     with open(p, "rb") as f:
         assert core.verify_checksum(f.read(), 0, 29) == 2**64 - 1
+
+
+def test_gpu_s3_on_gpu_verify(core, tmp_path):
+    """S3 PUT+GET with --gpuids: object data is generated by the gfx950 fill
+    kernel and GETs are verified on-GPU (BASELINE config 5 shape)."""
+    import sys
+    sys.path.insert(0, str(tmp_path.parents[len(tmp_path.parents) - 1]))
+    from tests.s3mock import ACCESS_KEY, SECRET_KEY, start_mock
+    from elbencho_amd.cli import main
+
+    server, port = start_mock()
+    try:
+        rc = main(["--s3endpoints", f"http://127.0.0.1:{port}", "--s3key", ACCESS_KEY,
+                   "--s3secret", SECRET_KEY, "--nolive", "--gpuids", "0",
+                   "-d", "-w", "-r", "-t", "2", "-N", "2", "-s", "24m", "-b", "8m",
+                   "--verify", "7", "s3://gpubkt"])
+        assert rc == 0
+        # corrupt one object and confirm the GPU verify catches it
+        from tests.s3mock import S3Handler
+        with S3Handler.store.lock:
+            key = next(iter(S3Handler.store.buckets["gpubkt"]))
+            data = bytearray(S3Handler.store.buckets["gpubkt"][key])
+            data[12345] ^= 0xFF
+            S3Handler.store.buckets["gpubkt"][key] = bytes(data)
+        rc = main(["--s3endpoints", f"http://127.0.0.1:{port}", "--s3key", ACCESS_KEY,
+                   "--s3secret", SECRET_KEY, "--nolive", "--gpuids", "0",
+                   "-r", "-t", "2", "-N", "2", "-s", "24m", "-b", "8m",
+                   "--verify", "7", "s3://gpubkt"])
+        assert rc == 1
+    finally:
+        server.shutdown()
