@@ -773,10 +773,10 @@ class S3Runner:
         self.phase_start = 0.0
         # pre-generated random payload block (like the engine's host buffers)
         import os as _os
-        self.rand_block = bytearray(_os.urandom(min(cfg.block_size, 1 << 22)))
-        while len(self.rand_block) < cfg.block_size:
-            self.rand_block += self.rand_block
-        self.rand_block = self.rand_block[:cfg.block_size]
+        blk = _os.urandom(min(max(cfg.block_size, 1), 1 << 22))
+        while len(blk) < cfg.block_size:
+            blk = blk + blk  # bytes concat (a bytearray cannot resize itself)
+        self.rand_block = blk[:cfg.block_size]
         self.upload_store = SharedUploadStore()
 
     # --- runner interface ---

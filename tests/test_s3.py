@@ -191,3 +191,11 @@ def test_s3_opslog(mock_s3, tmp_path):
     lines = [_json.loads(ln) for ln in log.read_text().splitlines()]
     assert any(l["op"] == "PutObject" and l["type"] == "pre" for l in lines)
     assert any(l["op"] == "GetObject" and l["type"] == "post" for l in lines)
+
+
+def test_s3_large_block_payload(mock_s3):
+    """block size above the 4 MiB urandom seed chunk (regression: bytearray
+    self-append raised BufferError)."""
+    rc = _cli(mock_s3, ["-d", "-w", "-r", "-F", "-D", "-t", "1", "-N", "1",
+                        "-s", "16m", "-b", "8m", "s3://bigblk"])
+    assert rc == 0
