@@ -33,7 +33,7 @@ def services():
              "--port", str(p)],
             env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
     # wait for readiness
-    deadline = time.monotonic() + 15
+    deadline = time.monotonic() + 40
     for p in ports:
         while True:
             try:
@@ -134,3 +134,37 @@ def test_netbench(services, tmp_path):
             break
     else:
         raise AssertionError("no Total MiB row in:\n" + res.stdout)
+
+
+def test_master_interrupt_rpc(services, tmp_path):
+    """--interrupt stops a running phase on the services."""
+    import threading
+
+    hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+    f = tmp_path / "big"
+
+    results = {}
+
+    def run():
+        results["res"] = run_master(["--hosts", hosts, "-t", "1", "-w", "-s", "1g",
+                                     "-b", "64k", "--limitwrite", "2m", str(f)])
+
+    t = threading.Thread(target=run)
+    t.start()
+    time.sleep(3)  # let the write phase start
+    intr = run_master(["--hosts", hosts, "--interrupt"])
+    assert intr.returncode == 0
+    t.join(60)
+    assert not t.is_alive()
+    # interrupted phase -> master exits nonzero with the interrupt reported
+    assert results["res"].returncode != 0
+    assert "interrupt" in (results["res"].stdout + results["res"].stderr).lower()
+
+
+def test_rotatehosts(services, tmp_path):
+    """--rotatehosts shifts service rank assignment between phases."""
+    hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+    res = run_master(["--hosts", hosts, "--rotatehosts", "1", "-t", "1", "-d",
+                      "-n", "1", "-w", "-r", "-N", "2", "-s", "4k", "-F", "-D",
+                      str(tmp_path)])
+    assert res.returncode == 0, res.stdout + res.stderr
