@@ -296,6 +296,18 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--s3btag", action="store_true", help="Phase: put bucket tagging.")
     g.add_argument("--s3btagverify", action="store_true",
                    help="Verify bucket tagging after put.")
+    g.add_argument("--s3bversion", action="store_true",
+                   help="Phase: enable bucket versioning.")
+    g.add_argument("--s3bversionverify", action="store_true",
+                   help="Verify bucket versioning state after enabling.")
+    g.add_argument("--s3olockcfg", action="store_true",
+                   help="Phase: put a bucket object-lock configuration.")
+    g.add_argument("--s3olockcfgverify", action="store_true",
+                   help="Verify the object-lock configuration after putting it.")
+    g.add_argument("--s3statdirs", action="store_true",
+                   help="Phase: HEAD the buckets (dir-style stat).")
+    g.add_argument("--s3listobjpar", action="store_true",
+                   help="Phase: parallel per-worker prefix listing.")
     g.add_argument("--s3credfile", default="", metavar="PATH",
                    help="File with one key:secret credential pair per line, "
                         "round-robined across workers.")
@@ -495,6 +507,12 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.s3_cred_list = args.s3credlist
     cfg.s3_no_mpu_compl = args.s3nompucompl
     cfg.s3_mpu_complete = args.s3mpucompl
+    cfg.s3_bversion = args.s3bversion
+    cfg.s3_bversion_verify = args.s3bversionverify
+    cfg.s3_olock = args.s3olockcfg
+    cfg.s3_olock_verify = args.s3olockcfgverify
+    cfg.run_statdirs = args.s3statdirs
+    cfg.s3_list_par = args.s3listobjpar
 
     cfg.config_file = args.configfile
 

@@ -225,6 +225,11 @@ class BenchConfig:
     s3_cred_list: str = ""         # --s3credlist ("key:secret,key:secret")
     s3_no_mpu_compl: bool = False  # --s3nompucompl (leave multipart uploads open)
     s3_mpu_complete: bool = False  # run the S3MPUCOMPLETE phase
+    s3_bversion: bool = False      # --s3bversion
+    s3_bversion_verify: bool = False  # --s3bversionverify
+    s3_olock: bool = False         # --s3olockcfg
+    s3_olock_verify: bool = False  # --s3olockcfgverify
+    s3_list_par: bool = False      # --s3listobjpar
 
     # --- misc ---
     config_file: str = ""          # -c/--configfile
@@ -366,6 +371,8 @@ class BenchConfig:
             # reference phase order, Coordinator.cpp:311-334
             order = [
                 ("MKDIRS", self.run_mkdirs),       # MKBUCKETS
+                ("BVERSION", self.s3_bversion),
+                ("OLOCKCFG", self.s3_olock),
                 ("PUTBACL", self.s3_bacl_put),
                 ("PUTBTAG", self.s3_btag),
                 ("GETBTAG", self.s3_btag and self.s3_btag_verify),
@@ -376,7 +383,9 @@ class BenchConfig:
                 ("STAT", self.run_stat),           # HEAD objects
                 ("GETOTAG", self.s3_otag and self.s3_otag_verify),
                 ("GETOBJACL", self.s3_acl_get),
+                ("STATDIRS", self.run_statdirs),
                 ("LISTOBJ", bool(self.s3_list_obj)),
+                ("LISTOBJPAR", self.s3_list_par),
                 ("READ", self.run_read),           # GET objects
                 ("DELOTAG", self.s3_otag and self.run_delfiles),
                 ("RMFILES", self.run_delfiles),    # delete objects

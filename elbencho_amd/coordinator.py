@@ -218,7 +218,9 @@ class Coordinator:
                   "GETBACL": "GETBACL", "PUTOTAG": "PUTOBJMD",
                   "GETOTAG": "GETOBJMD", "DELOTAG": "DELOBJMD",
                   "PUTBTAG": "PUTBUCKETMD", "GETBTAG": "GETBUCKETMD",
-                  "S3MPUCOMPLETE": "MPUCOMPL"}
+                  "S3MPUCOMPLETE": "MPUCOMPL", "BVERSION": "BVERSION",
+                  "OLOCKCFG": "OLOCKCFG", "STATDIRS": "STATDIRS",
+                  "LISTOBJPAR": "LISTOBJ_P"}
 
     def run_phase(self, name: str, quiet: bool = False) -> bool:
         cfg = self.cfg

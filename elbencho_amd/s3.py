@@ -257,6 +257,37 @@ class S3Client:
         return dict(_re.findall(r"<Key>([^<]*)</Key><Value>([^<]*)</Value>",
                                 data.decode()))
 
+    # --- versioning / object lock ---
+    def put_bucket_versioning(self, bucket: str, enabled: bool):
+        body = (f"<VersioningConfiguration><Status>"
+                f"{'Enabled' if enabled else 'Suspended'}</Status>"
+                f"</VersioningConfiguration>").encode()
+        status, data, _ = self.request("PUT", f"/{bucket}", query={"versioning": ""},
+                                       body=body)
+        self._check(status, data, f"put bucket versioning {bucket}")
+
+    def get_bucket_versioning(self, bucket: str) -> str:
+        status, data, _ = self.request("GET", f"/{bucket}", query={"versioning": ""})
+        self._check(status, data, f"get bucket versioning {bucket}")
+        import re as _re
+        m = _re.search(r"<Status>([^<]*)</Status>", data.decode())
+        return m.group(1) if m else ""
+
+    def put_object_lock_config(self, bucket: str, mode: str = "GOVERNANCE",
+                               days: int = 1):
+        body = (f"<ObjectLockConfiguration><ObjectLockEnabled>Enabled"
+                f"</ObjectLockEnabled><Rule><DefaultRetention><Mode>{mode}</Mode>"
+                f"<Days>{days}</Days></DefaultRetention></Rule>"
+                f"</ObjectLockConfiguration>").encode()
+        status, data, _ = self.request("PUT", f"/{bucket}", query={"object-lock": ""},
+                                       body=body)
+        self._check(status, data, f"put object lock config {bucket}")
+
+    def get_object_lock_config(self, bucket: str) -> bytes:
+        status, data, _ = self.request("GET", f"/{bucket}", query={"object-lock": ""})
+        self._check(status, data, f"get object lock config {bucket}")
+        return data
+
     # --- multipart ---
     def create_multipart(self, bucket: str, key: str) -> str:
         status, data, _ = self.request("POST", f"/{bucket}/{key}", query={"uploads": ""})
@@ -544,6 +575,44 @@ class S3Worker(threading.Thread):
                     if cfg.s3_btag_verify and tags.get("elbencho-amd") != "bucket":
                         raise S3Error(f"bucket tagging verification failed for {b}")
                     self.ops.entries += 1
+        elif ph == "BVERSION":
+            if self.local_rank == 0:
+                for b in self.r.buckets:
+                    self.client.put_bucket_versioning(b, True)
+                    if cfg.s3_bversion_verify:
+                        st = self.client.get_bucket_versioning(b)
+                        if st != "Enabled":
+                            raise S3Error(f"bucket versioning verification failed "
+                                          f"for {b}: {st!r}")
+                    self.ops.entries += 1
+        elif ph == "OLOCKCFG":
+            if self.local_rank == 0:
+                for b in self.r.buckets:
+                    self.client.put_object_lock_config(b)
+                    if cfg.s3_olock_verify:
+                        data = self.client.get_object_lock_config(b).decode()
+                        if "Enabled" not in data:
+                            raise S3Error(f"object lock verification failed for {b}")
+                    self.ops.entries += 1
+        elif ph == "STATDIRS":
+            # HEAD the buckets (reference --statdirs in S3 mode)
+            for i, b in enumerate(self.r.buckets):
+                if i % cfg.num_dataset_threads == self.rank:
+                    if not self.client.head_bucket(b):
+                        raise S3Error(f"bucket {b} does not exist")
+                    self.ops.entries += 1
+        elif ph == "LISTOBJPAR":
+            # parallel listing: each worker lists its own prefix slice
+            prefix = f"{cfg.s3_obj_prefix}r{self.rank}"
+            for bucket in self.r.buckets:
+                token = ""
+                while True:
+                    self._check_interrupt()
+                    objs, token = self.client.list_objects(
+                        bucket, prefix=prefix, max_keys=1000, continuation=token)
+                    self.ops.entries += len(objs)
+                    if not token:
+                        break
         elif ph == "S3MPUCOMPLETE":
             # complete multipart uploads left open by an earlier --s3nompucompl
             # run — possibly by ANOTHER instance: uploadIds and part ETags are

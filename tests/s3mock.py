@@ -30,6 +30,7 @@ class S3Store:
         self.next_upload = [0]
         self.acls: dict[tuple, str] = {}      # (bucket, key|None) -> canned acl
         self.tags: dict[tuple, bytes] = {}    # (bucket, key|None) -> tagging xml
+        self.misc: dict[tuple, bytes] = {}    # (bucket, kind) -> config xml
 
 
 def _xml(body: str) -> bytes:
@@ -125,6 +126,12 @@ class S3Handler(BaseHTTPRequestHandler):
             if "tagging" in q:
                 st.tags[(bucket, key)] = body
                 return self._send(200)
+            if "versioning" in q:
+                st.misc[(bucket, "versioning")] = body
+                return self._send(200)
+            if "object-lock" in q:
+                st.misc[(bucket, "object-lock")] = body
+                return self._send(200)
             if key is None:  # create bucket
                 if bucket in st.buckets:
                     return self._err(409, "BucketAlreadyOwnedByYou")
@@ -157,6 +164,12 @@ class S3Handler(BaseHTTPRequestHandler):
             if "tagging" in q:
                 return self._send(200, st.tags.get((bucket, key),
                                                    _xml("<Tagging><TagSet></TagSet></Tagging>")))
+            if "versioning" in q:
+                return self._send(200, st.misc.get(
+                    (bucket, "versioning"), _xml("<VersioningConfiguration/>")))
+            if "object-lock" in q:
+                return self._send(200, st.misc.get(
+                    (bucket, "object-lock"), _xml("<ObjectLockConfiguration/>")))
             if "uploads" in q and key is None:  # list multipart uploads
                 ups = "".join(
                     f"<Upload><Key>{k}</Key><UploadId>{uid}</UploadId></Upload>"

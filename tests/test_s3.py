@@ -199,3 +199,14 @@ def test_s3_large_block_payload(mock_s3):
     rc = _cli(mock_s3, ["-d", "-w", "-r", "-F", "-D", "-t", "1", "-N", "1",
                         "-s", "16m", "-b", "8m", "s3://bigblk"])
     assert rc == 0
+
+
+def test_s3_versioning_objectlock_statdirs_listpar(mock_s3, capsys):
+    rc = _cli(mock_s3, ["-d", "-w", "-t", "2", "-N", "2", "-s", "4k", "-b", "4k",
+                        "--s3bversion", "--s3bversionverify",
+                        "--s3olockcfg", "--s3olockcfgverify",
+                        "--s3statdirs", "--s3listobjpar", "s3://verbkt"])
+    out = capsys.readouterr().out
+    assert rc == 0, out
+    for phase in ("BVERSION", "OLOCKCFG", "STATDIRS", "LISTOBJ_P"):
+        assert phase in out, f"{phase} missing:\n{out}"
