@@ -556,8 +556,42 @@ def apply_config_file(argv: list[str]) -> list[str]:
     return extra + argv
 
 
+HELP_TOPICS = {
+    "--help-large": (
+        "Large shared files or block devices (e.g. streaming or random IOPS)",
+        ["workload", "access pattern", "GPU (MI355X)", "results & latency"]),
+    "--help-multi": (
+        "Multiple dirs and files per thread (e.g. lots of small files)",
+        ["benchmark phases", "workload", "data integrity & variance"]),
+    "--help-s3": ("S3 object storage", ["S3 object storage", "workload"]),
+    "--help-dist": ("Multiple clients (e.g. shared file systems)",
+                    ["distributed mode", "network benchmark"]),
+    "--help-all": ("Overview of all available options", None),
+}
+
+
+def print_topic_help(topic: str) -> None:
+    parser = build_parser()
+    title, groups = HELP_TOPICS[topic]
+    print(title + "\n")
+    if groups is None:
+        parser.print_help()
+        return
+    for g in parser._action_groups:
+        if g.title in groups:
+            fmt = parser._get_formatter()
+            fmt.start_section(g.title)
+            fmt.add_arguments(g._group_actions)
+            fmt.end_section()
+            print(fmt.format_help())
+
+
 def main(argv: list[str] | None = None) -> int:
     argv = list(sys.argv[1:] if argv is None else argv)
+    for topic in HELP_TOPICS:
+        if topic in argv:
+            print_topic_help(topic)
+            return 0
     argv = apply_config_file(argv)
     parser = build_parser()
     args = parser.parse_args(argv)

@@ -255,6 +255,10 @@ class Coordinator:
         if self.dist:
             self.dist.barrier()  # lockstep phase start across GPU ranks
 
+        if cfg.log_level >= 1:
+            print(f"[phase {display}] id={phase_id} planned_entries={planned_entries} "
+                  f"planned_bytes={planned_bytes}", file=sys.stderr)
+
         self.runner.start(name)
 
         deadline = time.monotonic() + cfg.timelimit if cfg.timelimit else None
@@ -312,6 +316,16 @@ class Coordinator:
         if cfg.res_file and (self.dist is None or self.dist.rank == 0):
             with open(cfg.res_file, "a") as f:
                 print_phase_results(cfg, results, f)
+
+        # sub-microsecond phases usually mean a mis-configured benchmark
+        # (reference --no0usecerr contract)
+        did_work = results.bytes or results.entries or results.iops
+        if (did_work and results.last_finish_usec == 0
+                and not cfg.ignore_0usec_errors and name not in ("SYNC", "DROPCACHES")):
+            print(f"ERROR: phase {display} completed in less than a microsecond — "
+                  "the configuration likely measures nothing "
+                  "(--no0usecerr to ignore)", file=sys.stderr)
+            return False
 
         interrupted_only = all(e.endswith("interrupted") for e in results.errors)
         if results.errors and not (self.interrupted and interrupted_only):
