@@ -224,7 +224,8 @@ def build_parser() -> argparse.ArgumentParser:
                    help="Show elapsed time per service host.")
     g.add_argument("--svcpwfile", default="", metavar="PATH",
                    help="Shared-secret file to authorize master<->service communication.")
-    g.add_argument("--svcwait", action="store_true", help=argparse.SUPPRESS)
+    g.add_argument("--svcwait", action="store_true",
+                   help="Wait indefinitely for services to become reachable.")
     g.add_argument("--svcping", action="store_true", help=argparse.SUPPRESS)
     g.add_argument("--interrupt", action="store_true",
                    help="Interrupt the current phase on the given service hosts.")
@@ -334,7 +335,12 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--s3fastput", action="store_true", help=argparse.SUPPRESS)
     g.add_argument("--s3sign", type=int, default=0, help=argparse.SUPPRESS)
     g.add_argument("--s3maxconns", type=int, default=0, help=argparse.SUPPRESS)
-    g.add_argument("--s3ignoreerrors", action="store_true", help=argparse.SUPPRESS)
+    g.add_argument("--s3ignoreerrors", action="store_true",
+                   help="Record S3 op errors per worker but keep the phase running.")
+    g.add_argument("--s3sse", action="store_true",
+                   help="Request server-side encryption (SSE-S3/AES256) on uploads.")
+    g.add_argument("--s3sseckey", default="", help=argparse.SUPPRESS)
+    g.add_argument("--s3ssekmskey", default="", help=argparse.SUPPRESS)
     g.add_argument("--s3nocompress", action="store_true", help=argparse.SUPPRESS)
 
     g = p.add_argument_group("misc")
@@ -450,6 +456,7 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.svc_update_int_ms = args.svcupint
     cfg.svc_elapsed = args.svcelapsed
     cfg.svc_pw_file = args.svcpwfile
+    cfg.svc_wait = args.svcwait
     cfg.interrupt_services = args.interrupt
     cfg.quit_services = args.quit
     cfg.rank_offset = args.rankoffset
@@ -513,6 +520,8 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.s3_olock_verify = args.s3olockcfgverify
     cfg.run_statdirs = args.s3statdirs
     cfg.s3_list_par = args.s3listobjpar
+    cfg.s3_sse = args.s3sse
+    cfg.s3_ignore_errors = args.s3ignoreerrors
 
     cfg.config_file = args.configfile
 

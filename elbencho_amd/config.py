@@ -170,6 +170,7 @@ class BenchConfig:
     svc_update_int_ms: int = 500   # --svcupint
     svc_elapsed: bool = False      # --svcelapsed
     svc_pw_file: str = ""          # --svcpwfile
+    svc_wait: bool = False         # --svcwait
     interrupt_services: bool = False  # --interrupt
     quit_services: bool = False    # --quit
     rank_offset: int = 0           # --rankoffset
@@ -230,6 +231,7 @@ class BenchConfig:
     s3_olock: bool = False         # --s3olockcfg
     s3_olock_verify: bool = False  # --s3olockcfgverify
     s3_list_par: bool = False      # --s3listobjpar
+    s3_sse: bool = False           # --s3sse (SSE-S3 AES256 header)
 
     # --- misc ---
     config_file: str = ""          # -c/--configfile
@@ -471,6 +473,7 @@ class BenchConfig:
             resp_size=self.resp_size,
             send_buf=self.send_buf,
             recv_buf=self.recv_buf,
+            netdevs=self.netdevs,
             tree_dirs=self.tree_dirs_resolved,
             tree_files=self.tree_files_resolved,
             sharesize=self.sharesize,

@@ -1691,7 +1691,9 @@ void Worker::netbenchClient()
         port = std::stoi(srv.substr(pos + 1)) + 1000; // service port + 1000
     }
 
-    int fd = netConnect(host, port, 0, 30, eng.interruptFlag);
+    const std::string bindDev =
+        cfg.netDevs.empty() ? "" : cfg.netDevs[globalRank % cfg.netDevs.size()];
+    int fd = netConnect(host, port, bindDev, 30, eng.interruptFlag);
     setSockBufs(fd, cfg.sendBufSize, cfg.recvBufSize);
 
     std::vector<char> resp(cfg.respSize);

@@ -86,6 +86,7 @@ struct EngineConfig {
     int netbenchNumConns = 0;  // server: total client connections to expect
     uint64_t respSize = 1;
     int sendBufSize = 0, recvBufSize = 0;
+    std::vector<std::string> netDevs; // --netdevs round-robin client binding
 
     // custom tree mode (reference PathStore / --treefile):
     // dirs + (relpath, size) files under paths[0]; files >= shareSize are

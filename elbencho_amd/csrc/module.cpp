@@ -89,6 +89,8 @@ EngineConfig configFromDict(const py::dict& d)
     c.respSize = getU64("resp_size", 1);
     c.sendBufSize = (int)getI("send_buf", 0);
     c.recvBufSize = (int)getI("recv_buf", 0);
+    if (d.contains("netdevs"))
+        c.netDevs = d["netdevs"].cast<std::vector<std::string>>();
     c.blockVarPct = (int)getI("blockvar_pct", 100);
     c.blockVarAlgo = getS("blockvar_algo", "fast");
     c.randAlgo = getS("rand_algo", "balanced_single");

@@ -58,8 +58,9 @@ inline int netListen(int port, int backlog)
     return fd;
 }
 
-// connect with retry (client side; reference retries while services come up)
-inline int netConnect(const std::string& host, int port, int bindDevIdxUnused,
+// connect with retry (client side; reference retries while services come up);
+// bindDev non-empty binds the outgoing socket to that device (--netdevs)
+inline int netConnect(const std::string& host, int port, const std::string& bindDev,
                       int timeoutSecs, const std::atomic<bool>& interruptFlag)
 {
     struct addrinfo hints{}, *res = nullptr;
@@ -73,6 +74,9 @@ inline int netConnect(const std::string& host, int port, int bindDevIdxUnused,
             throw InterruptedError();
         if (getaddrinfo(host.c_str(), portStr.c_str(), &hints, &res) == 0) {
             int fd = socket(res->ai_family, res->ai_socktype, res->ai_protocol);
+            if (fd >= 0 && !bindDev.empty())
+                setsockopt(fd, SOL_SOCKET, SO_BINDTODEVICE, bindDev.c_str(),
+                           (socklen_t)bindDev.size());
             if (fd >= 0 && connect(fd, res->ai_addr, res->ai_addrlen) == 0) {
                 freeaddrinfo(res);
                 return fd;

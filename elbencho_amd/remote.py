@@ -114,6 +114,8 @@ class RemoteRunner:
 
     # ------------------------------------------------------------------
     def _wait_for_services(self, timeout: float = 10.0) -> None:
+        if getattr(self.cfg, "svc_wait", False):
+            timeout = 365 * 24 * 3600.0  # --svcwait: wait for services forever
         deadline = time.monotonic() + timeout
         for hs in self.hosts:
             while True:

@@ -45,7 +45,9 @@ class S3Client:
     """One S3 endpoint connection with AWS Signature V4 (path-style URLs)."""
 
     def __init__(self, endpoint: str, access_key: str, secret_key: str,
-                 region: str = "us-east-1", timeout: float = 60.0):
+                 region: str = "us-east-1", timeout: float = 60.0,
+                 extra_put_headers: dict[str, str] | None = None):
+        self.extra_put_headers = extra_put_headers or {}
         u = urllib.parse.urlparse(endpoint if "//" in endpoint else "http://" + endpoint)
         self.host = u.hostname or "localhost"
         self.port = u.port or (443 if u.scheme == "https" else 80)
@@ -152,7 +154,8 @@ class S3Client:
 
     # --- object ops ---
     def put_object(self, bucket: str, key: str, body: bytes):
-        status, data, _ = self.request("PUT", f"/{bucket}/{key}", body=body)
+        status, data, _ = self.request("PUT", f"/{bucket}/{key}", body=body,
+                                       headers=dict(self.extra_put_headers))
         self._check(status, data, f"put {bucket}/{key}")
 
     def get_object(self, bucket: str, key: str,
@@ -416,7 +419,11 @@ class S3Worker(threading.Thread):
         key, secret = cfg.s3_key, cfg.s3_secret
         if runner.credentials:  # --s3credfile/--s3credlist round-robin
             key, secret = runner.credentials[self.rank % len(runner.credentials)]
-        self.client = S3Client(ep, key, secret, cfg.s3_region)
+        put_headers = {}
+        if cfg.s3_sse:  # SSE-S3 (AES256) passthrough header
+            put_headers["x-amz-server-side-encryption"] = "AES256"
+        self.client = S3Client(ep, key, secret, cfg.s3_region,
+                               extra_put_headers=put_headers)
         self.core = load_core()
         # --gpuids + --verify: verify/generate object data with the gfx950
         # kernels in HBM (BASELINE config 5); one persistent context per worker
@@ -464,7 +471,13 @@ class S3Worker(threading.Thread):
         try:
             self.r.start_gate.wait()
             t0 = time.monotonic()
-            self._run_phase()
+            try:
+                self._run_phase()
+            except S3Error as e:
+                if not self.r.cfg.s3_ignore_errors:
+                    raise
+                # --s3ignoreerrors: record and carry on (stress-mode knob)
+                self.error = f"(ignored) {e}"
             self.elapsed_us = int((time.monotonic() - t0) * 1e6)
         except KeyboardInterrupt:
             self.error = "interrupted"
