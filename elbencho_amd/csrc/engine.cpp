@@ -1111,6 +1111,9 @@ void Worker::fileModeBlocksUring(bool isWrite)
 
         slots[slot] = {inFileOff, ioLen, lat ? Clock::now() : Clock::time_point(),
                        (int)fileIdx, blockWrite};
+        if (eng.opsLog.isEnabled())
+            eng.opsLog.log(globalRank, blockWrite ? "uring_write" : "uring_read",
+                           cfg.paths[fileIdx], inFileOff, ioLen, true, false);
         if (!ring.prep(blockWrite, fg.fds[fileIdx], hostBufs[slot], ioLen, inFileOff,
                        (uint64_t)slot))
             throw WorkerError("io_uring SQ unexpectedly full");
@@ -1134,6 +1137,10 @@ void Worker::fileModeBlocksUring(bool isWrite)
             inFlight--;
 
             const bool wasWrite = st.isWriteOp;
+            if (eng.opsLog.isEnabled())
+                eng.opsLog.log(globalRank, wasWrite ? "uring_write" : "uring_read",
+                               cfg.paths[st.fileIdx], st.inFileOff, st.len, false,
+                               comps[i].res < 0);
             if (comps[i].res < 0)
                 throw WorkerError(std::string("async ") + (wasWrite ? "write" : "read") +
                                   " failed. Path: " + cfg.paths[st.fileIdx] +

@@ -11,6 +11,8 @@
 #include <cstdint>
 #include <cstring>
 #include <random>
+#include <stdexcept>
+#include <string>
 
 namespace eb {
 
@@ -57,6 +59,36 @@ public:
         return state ^ (state >> 29);
     }
 
+    // tight non-virtual fill loop (the per-block variance refill is on the
+    // CPU write hot path; one virtual call per 8 bytes costs GB/s)
+    void fillBuf(char* buf, uint64_t len) override
+    {
+        uint64_t s = state;
+        while (len >= 32) { // 4-way unroll, dependency chain stays serial
+            uint64_t v0 = s * 0x9E3779B97F4A7C15ULL + 0xD1B54A32D192ED03ULL;
+            uint64_t v1 = v0 * 0x9E3779B97F4A7C15ULL + 0xD1B54A32D192ED03ULL;
+            uint64_t v2 = v1 * 0x9E3779B97F4A7C15ULL + 0xD1B54A32D192ED03ULL;
+            uint64_t v3 = v2 * 0x9E3779B97F4A7C15ULL + 0xD1B54A32D192ED03ULL;
+            uint64_t o0 = v0 ^ (v0 >> 29), o1 = v1 ^ (v1 >> 29);
+            uint64_t o2 = v2 ^ (v2 >> 29), o3 = v3 ^ (v3 >> 29);
+            std::memcpy(buf, &o0, 8);
+            std::memcpy(buf + 8, &o1, 8);
+            std::memcpy(buf + 16, &o2, 8);
+            std::memcpy(buf + 24, &o3, 8);
+            s = v3;
+            buf += 32;
+            len -= 32;
+        }
+        state = s;
+        while (len) {
+            uint64_t v = next();
+            uint64_t n = len < 8 ? len : 8;
+            std::memcpy(buf, &v, n);
+            buf += n;
+            len -= n;
+        }
+    }
+
 private:
     uint64_t state;
 };
@@ -81,6 +113,29 @@ public:
         s[2] ^= t;
         s[3] = rotl64(s[3], 45);
         return result;
+    }
+
+    void fillBuf(char* buf, uint64_t len) override
+    {
+        uint64_t a = s[0], b = s[1], c = s[2], d = s[3];
+        while (len >= 8) {
+            const uint64_t result = rotl64(b * 5, 7) * 9;
+            const uint64_t t = b << 17;
+            c ^= a;
+            d ^= b;
+            b ^= c;
+            a ^= d;
+            c ^= t;
+            d = rotl64(d, 45);
+            std::memcpy(buf, &result, 8);
+            buf += 8;
+            len -= 8;
+        }
+        s[0] = a; s[1] = b; s[2] = c; s[3] = d;
+        if (len) {
+            uint64_t v = next();
+            std::memcpy(buf, &v, len);
+        }
     }
 
 private:
