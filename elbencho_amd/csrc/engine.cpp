@@ -209,10 +209,10 @@ void Worker::rwBalanceWait(bool isRead, uint64_t nextLen)
         uint64_t mine = isRead ? rd : wr;
         if (mine + nextLen <= allowed) return;
 
-        // peers all done? then no more balancing possible — proceed
-        if (eng.workersDone.load(std::memory_order_relaxed) >= (int)peers &&
-            eng.workersDone.load(std::memory_order_relaxed) > 0)
-            return;
+        // the peer GROUP is done? then no more balancing possible — proceed
+        int peersDone = (isRead ? eng.rwWritersDone : eng.rwReadersDone)
+                            .load(std::memory_order_relaxed);
+        if (peers > 0 && peersDone >= (int)peers) return;
 
         checkInterrupt();
         if (std::chrono::duration_cast<std::chrono::seconds>(Clock::now() - waitStart)
@@ -629,7 +629,7 @@ void Worker::fileModeBlocks(bool isWrite)
     if (cfg.ioDepth > 1) return fileModeBlocksUring(isWrite);
     if (cfg.useMmap && !cfg.gpuIDs.empty() && !cfg.measureLat && cfg.flockMode == 0 &&
         !(isWrite && (cfg.rwMixPct > 0 || cfg.rwMixThreads > 0)) && !cfg.verifyDirect &&
-        cfg.pathType == PathType::FILE)
+        cfg.pathType == PathType::FILE && !eng.opsLog.isEnabled())
         return fileModeBlocksGpuMmap(isWrite);
 
     const uint64_t fileSize = eng.effFileSize;
@@ -707,7 +707,8 @@ void Worker::fileModeBlocks(bool isWrite)
     }();
     const bool gpuPipelined = gpu && !pipelineDisabled && cfg.ioDepth == 1 &&
                               !cfg.measureLat && !cfg.useMmap && cfg.flockMode == 0 &&
-                              !rwMixActive && !cfg.verifyDirect && hostBufs.size() >= 2;
+                              !rwMixActive && !cfg.verifyDirect && hostBufs.size() >= 2 &&
+                              !eng.opsLog.isEnabled(); // tracing needs per-op hooks
     if (gpuPipelined) {
         constexpr uint64_t VERIFY_FETCH_INTERVAL = 64;
         const bool doVerify = cfg.verifySalt >= 0;
@@ -1842,6 +1843,7 @@ void Worker::threadMain()
         // --rwmixthr: first N threads of a write phase only read
         isDedicatedReader =
             (eng.currentPhase == Phase::WRITE) && (localRank < cfg.rwMixThreads);
+        dedicatedReader = isDedicatedReader;
 
         // buffers only needed for data phases
         if (eng.currentPhase == Phase::WRITE || eng.currentPhase == Phase::READ ||
@@ -1936,6 +1938,8 @@ void Engine::startPhase(Phase phase)
     currentPhase = phase;
     rwBalBytesRead.store(0);
     rwBalBytesWrite.store(0);
+    rwReadersDone.store(0);
+    rwWritersDone.store(0);
     nbAcceptDone = false;
     nbConns.clear();
     if (!cfg.opsLogPath.empty() && !opsLog.isEnabled())
@@ -1964,6 +1968,8 @@ void Engine::startPhase(Phase phase)
 void Engine::onWorkerDone(Worker& w, bool hadError)
 {
     std::lock_guard<std::mutex> lk(doneMtx);
+
+    (w.dedicatedReader ? rwReadersDone : rwWritersDone).fetch_add(1);
 
     if (hadError) {
         workersWithError.fetch_add(1);

@@ -154,6 +154,7 @@ public:
     std::string error;
     int localRank;
     int globalRank;
+    bool dedicatedReader = false; // rwmix role (read by Engine::onWorkerDone)
 
 private:
     using Clock = std::chrono::steady_clock;
@@ -273,6 +274,8 @@ public:
     // rwMixPct% of the combined volume with blockSize*peerThreads headroom
     std::atomic<uint64_t> rwBalBytesRead{0};
     std::atomic<uint64_t> rwBalBytesWrite{0};
+    std::atomic<int> rwReadersDone{0};
+    std::atomic<int> rwWritersDone{0};
 
     // resolved at prepare()
     std::vector<uint64_t> resolvedFileSizes; // per path (file/bdev mode)
