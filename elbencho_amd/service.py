@@ -96,7 +96,7 @@ class ServiceState:
             p = self.runner.poll()
             p["bench_id"] = self.bench_id
             p["phase_name"] = self.phase_name
-            p["cpu_util_pct"] = 0
+            p["cpu_util_pct"] = self.cpu.percent_since_last()
             p["idle"] = False
             return p
 
