@@ -381,7 +381,8 @@ std::unique_ptr<OffsetGen> Worker::makeOffsetGen(uint64_t rangeStart, uint64_t r
         // random aligned writes: full coverage so the resulting file is fully
         // allocated (reference behavior, LocalWorker.cpp:1177-1185)
         gen = std::make_unique<OffsetGenRandomAlignedFullCoverage>(
-            cfg.blockSize, cfg.benchSeed ^ (0x9E3779B97F4A7C15ULL * (globalRank + 1)));
+            cfg.blockSize, cfg.benchSeed + 0x9E3779B97F4A7C15ULL * (uint64_t)eng.phaseSeq +
+                               0x94D049BB133111EBULL * (uint64_t)(globalRank + 1));
     } else {
         gen = std::make_unique<OffsetGenRandomAligned>(cfg.blockSize, *rng, perWorkerRandAmount);
     }
@@ -1835,10 +1836,12 @@ void Worker::threadMain()
 
         applyBinding();
 
+        const uint64_t phaseSeed =
+            cfg.benchSeed + 0x9E3779B97F4A7C15ULL * (uint64_t)eng.phaseSeq;
         rng.reset(makeRandAlgo(cfg.randAlgo,
-                               cfg.benchSeed ^ (0xBF58476D1CE4E5B9ULL * (globalRank + 1))));
+                               phaseSeed ^ (0xBF58476D1CE4E5B9ULL * (globalRank + 1))));
         fillRng.reset(makeRandAlgo(cfg.blockVarAlgo,
-                                   cfg.benchSeed ^ (0x94D049BB133111EBULL * (globalRank + 1))));
+                                   phaseSeed ^ (0x94D049BB133111EBULL * (globalRank + 1))));
 
         // --rwmixthr: first N threads of a write phase only read
         isDedicatedReader =
@@ -1936,6 +1939,7 @@ void Engine::startPhase(Phase phase)
     if (phaseRunning) throw std::runtime_error("phase already running");
 
     currentPhase = phase;
+    phaseSeq++; // new offset/fill streams every phase (iterations differ)
     rwBalBytesRead.store(0);
     rwBalBytesWrite.store(0);
     rwReadersDone.store(0);

@@ -276,6 +276,7 @@ public:
     std::atomic<uint64_t> rwBalBytesWrite{0};
     std::atomic<int> rwReadersDone{0};
     std::atomic<int> rwWritersDone{0};
+    int phaseSeq = 0; // bumped per startPhase; decorrelates iteration streams
 
     // resolved at prepare()
     std::vector<uint64_t> resolvedFileSizes; // per path (file/bdev mode)

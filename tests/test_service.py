@@ -168,3 +168,24 @@ def test_rotatehosts(services, tmp_path):
                       "-n", "1", "-w", "-r", "-N", "2", "-s", "4k", "-F", "-D",
                       str(tmp_path)])
     assert res.returncode == 0, res.stdout + res.stderr
+
+
+def test_distributed_custom_tree(services, tmp_path):
+    """Custom tree partitioning across two services (tree shipped in the
+    wire config; shared big file range-sliced across all ranks)."""
+    from elbencho_amd.pathstore import CustomTree, write_treefile
+
+    base = tmp_path / "bench"
+    base.mkdir()
+    tf = tmp_path / "tree.txt"
+    tree = CustomTree(dirs=["a", "a/b"],
+                      files=[("a/small1", 4096), ("a/b/small2", 8192),
+                             ("bigshared", 1 << 20)])
+    write_treefile(tree, str(tf))
+    hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+    res = run_master(["--hosts", hosts, "-t", "2", "-d", "-w", "-r", "-F", "-D",
+                      "-b", "64k", "--treefile", str(tf), "--sharesize", "512k",
+                      "--verify", "6", str(base)])
+    assert res.returncode == 0, res.stdout + res.stderr
+    assert not (base / "a").exists()
+    assert not (base / "bigshared").exists()
