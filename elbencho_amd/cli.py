@@ -33,6 +33,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("paths", nargs="*", metavar="PATH",
                    help="Benchmark paths (files, block devices, directories or s3:// buckets). "
                         "Supports bracket expansion, e.g. /mnt/file[1-4].")
+    p.add_argument("--path", dest="path_opt", action="append", default=[],
+                   metavar="PATH", help="Benchmark path (alternative to positional paths; "
+                                        "may be given multiple times).")
 
     g = p.add_argument_group("benchmark phases")
     g.add_argument("-w", "--write", action="store_true", help="Write/create files or objects.")
@@ -167,6 +170,9 @@ def build_parser() -> argparse.ArgumentParser:
                    help="Serialize ops log writes across processes via flock.")
     g.add_argument("--log", type=int, default=0, metavar="LEVEL",
                    help="Log level: 0 normal, 1 verbose, 2 debug.")
+    g.add_argument("--base10", action="store_true",
+                   help="Show throughput in base10 instead of base2 numbers "
+                        "(e.g. MB/s instead of MiB/s).")
 
     g = p.add_argument_group("limits & timing")
     g.add_argument("--timelimit", type=int, default=0, metavar="SECS",
@@ -210,8 +216,12 @@ def build_parser() -> argparse.ArgumentParser:
                    help="Run as service for distributed mode (control via master).")
     g.add_argument("--port", type=int, default=1611, metavar="N",
                    help="TCP port of service. (Default: 1611)")
-    g.add_argument("--foreground", action="store_true",
-                   help="Run service in foreground instead of daemonizing.")
+    g.add_argument("--foreground", "--nodetach", action="store_true",
+                   help="Run service in foreground instead of daemonizing. "
+                        "(--nodetach is the reference-compatible alias.)")
+    g.add_argument("--althttpsvc", action="store_true",
+                   help="Use alternative (single-threaded) implementation of the "
+                        "HTTP service, for testing.")
     g.add_argument("--nosvcshare", action="store_true",
                    help="Benchmark paths are not shared between service hosts.")
     g.add_argument("--numhosts", type=int, default=-1, metavar="N",
@@ -247,6 +257,9 @@ def build_parser() -> argparse.ArgumentParser:
                    help="Scan an existing directory tree into a tree file.")
     g.add_argument("--treerand", action="store_true",
                    help="Process custom tree files in random order.")
+    g.add_argument("--treeroundrob", action="store_true",
+                   help="Assign shared-file blocks round-robin to workers instead of "
+                        "maximizing consecutive ranges per worker.")
 
     g = p.add_argument_group("mixed read/write")
     g.add_argument("--rwmixpct", type=int, default=0, metavar="PCT",
@@ -269,6 +282,8 @@ def build_parser() -> argparse.ArgumentParser:
                    help="File with one netbench client host per line.")
     g.add_argument("--serversfile", default="", metavar="PATH",
                    help="File with one netbench server per line.")
+    g.add_argument("--numservers", type=int, default=0, metavar="N",
+                   help="Use only the first N hosts of the --servers list.")
     g.add_argument("--respsize", default="1", metavar="SIZE",
                    help="Netbench server response size per received block. (Default: 1)")
     g.add_argument("--sendbuf", default="0", metavar="SIZE", help="Socket send buffer size.")
@@ -337,6 +352,48 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--s3maxconns", type=int, default=0, help=argparse.SUPPRESS)
     g.add_argument("--s3ignoreerrors", action="store_true",
                    help="Record S3 op errors per worker but keep the phase running.")
+    g.add_argument("--s3sessiontoken", default="", metavar="TOKEN",
+                   help="S3 session token (x-amz-security-token header).")
+    g.add_argument("--s3chksumalgo", default="", metavar="ALGO",
+                   help="S3 checksum algorithm (CRC32, CRC32C, SHA1, SHA256): sets the "
+                        "x-amz-sdk-checksum-algorithm and x-amz-checksum-* headers on "
+                        "uploads.")
+    g.add_argument("--s3checksumalgo", dest="s3chksumalgo", help=argparse.SUPPRESS)
+    g.add_argument("--s3aclgrantee", default="", metavar="NAME",
+                   help="S3 ACL grantee; special canned values private, public-read, "
+                        "public-read-write, authenticated-read ignore the grantee "
+                        "type/permissions.")
+    g.add_argument("--s3aclgtype", default="", metavar="TYPE",
+                   help="S3 ACL grantee type: id, emailAddress, uri, group.")
+    g.add_argument("--s3aclputinl", action="store_true",
+                   help="Set S3 object ACL inline in the object upload (requires "
+                        "grantee and permissions).")
+    g.add_argument("--s3mpusharing", action="store_true",
+                   help="Shared multipart upload mode: object names are given as "
+                        "parameters (e.g. mybucket/myobj[1-10]) and all workers "
+                        "upload disjoint parts of each object.")
+    g.add_argument("--s3mpusizevar", default="0", metavar="SIZE",
+                   help="Maximum number of bytes to subtract from the multipart part "
+                        "size for random part-size variance (last part absorbs the "
+                        "difference).")
+    g.add_argument("--s3mpusplit", default="0", metavar="SIZE",
+                   help="Multipart part size override (default: the -b block size).")
+    g.add_argument("--s3nompcheck", action="store_true",
+                   help="Don't check for multipart uploads exceeding 10,000 parts.")
+    g.add_argument("--s3single", action="store_true",
+                   help="Use a single shared S3 client instance for all threads.")
+    g.add_argument("--s3targetgbps", type=int, default=0, metavar="N",
+                   help="Throughput target per S3 client in Gbps (informational; "
+                        "this client sizes connections per worker thread).")
+    g.add_argument("--s3virtaddr", action="store_true",
+                   help="Use S3 virtual-hosted addressing (bucket as subdomain of "
+                        "the endpoint DNS name).")
+    g.add_argument("--s3log", type=int, default=0, metavar="LEVEL",
+                   help="S3 client request trace log level (0=disabled). "
+                        "See --s3logprefix for the file name.")
+    g.add_argument("--s3logprefix", default="", metavar="PREFIX",
+                   help="Path and filename prefix of the S3 client trace log; "
+                        "\"DATE.log\" gets appended. (Default: s3_client_)")
     g.add_argument("--s3sse", action="store_true",
                    help="Request server-side encryption (SSE-S3/AES256) on uploads.")
     g.add_argument("--s3sseckey", default="", help=argparse.SUPPRESS)
@@ -355,7 +412,7 @@ def build_parser() -> argparse.ArgumentParser:
 
 def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg = BenchConfig()
-    cfg.paths = list(args.paths)
+    cfg.paths = list(args.paths) + list(args.path_opt)
     cfg.run_write = args.write
     cfg.run_read = args.read
     cfg.run_stat = args.stat
@@ -415,6 +472,7 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.no_csv_labels = args.nocsvlabels
     cfg.label = args.label
     cfg.show_dir_stats = args.dirstats
+    cfg.show_base10 = args.base10
     cfg.live_int_ms = args.liveint
     cfg.no_live = args.nolive
     cfg.live1 = args.live1
@@ -450,6 +508,7 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.service_mode = args.service
     cfg.service_port = args.port
     cfg.foreground = args.foreground
+    cfg.alt_http_svc = args.althttpsvc
     cfg.no_svc_share = args.nosvcshare
     cfg.num_hosts = args.numhosts
     cfg.rotate_hosts = args.rotatehosts
@@ -467,6 +526,7 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.tree_round_up = parse_size(args.treeroundup)
     cfg.tree_scan = args.treescan
     cfg.tree_rand = args.treerand
+    cfg.tree_round_robin = args.treeroundrob
 
     cfg.rwmix_pct = args.rwmixpct
     cfg.rwmix_threads = args.rwmixthr
@@ -485,6 +545,7 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     if args.serversfile:
         with open(args.serversfile) as f:
             cfg.servers = [ln.strip() for ln in f if ln.strip()]
+    cfg.num_servers = args.numservers
     cfg.resp_size = parse_size(args.respsize)
     cfg.send_buf = parse_size(args.sendbuf)
     cfg.recv_buf = parse_size(args.recvbuf)
@@ -521,6 +582,22 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.run_statdirs = args.s3statdirs
     cfg.s3_list_par = args.s3listobjpar
     cfg.s3_sse = args.s3sse
+    cfg.s3_sse_c_key = args.s3sseckey
+    cfg.s3_sse_kms_key = args.s3ssekmskey
+    cfg.s3_session_token = args.s3sessiontoken
+    cfg.s3_chksum_algo = args.s3chksumalgo
+    cfg.s3_acl_grantee = args.s3aclgrantee
+    cfg.s3_acl_gtype = args.s3aclgtype
+    cfg.s3_acl_put_inline = args.s3aclputinl
+    cfg.s3_mpu_sharing = args.s3mpusharing
+    cfg.s3_mpu_size_var = parse_size(args.s3mpusizevar)
+    cfg.s3_mpu_split = parse_size(args.s3mpusplit)
+    cfg.s3_no_mp_check = args.s3nompcheck
+    cfg.s3_single = args.s3single
+    cfg.s3_target_gbps = args.s3targetgbps
+    cfg.s3_virt_addr = args.s3virtaddr
+    cfg.s3_log = args.s3log
+    cfg.s3_log_prefix = args.s3logprefix
     cfg.s3_ignore_errors = args.s3ignoreerrors
 
     cfg.config_file = args.configfile

@@ -137,6 +137,7 @@ class BenchConfig:
     no_csv_labels: bool = False    # --nocsvlabels
     label: str = ""                # --label
     show_dir_stats: bool = False   # --dirstats
+    show_base10: bool = False      # --base10 (MB/s instead of MiB/s in output)
     log_level: int = 0             # --log
     ops_log_path: str = ""         # --opslog
     ops_log_lock: bool = False     # --opsloglock
@@ -163,7 +164,8 @@ class BenchConfig:
     hosts: list[str] = field(default_factory=list)  # --hosts
     service_mode: bool = False     # --service
     service_port: int = 1611       # --port
-    foreground: bool = False       # --foreground
+    foreground: bool = False       # --foreground / --nodetach
+    alt_http_svc: bool = False     # --althttpsvc (single-threaded HTTP server)
     no_svc_share: bool = False     # --nosvcshare
     num_hosts: int = -1            # --numhosts
     rotate_hosts: int = 0          # --rotatehosts
@@ -182,6 +184,7 @@ class BenchConfig:
     tree_round_up: int = 0         # --treeroundup
     tree_scan: str = ""            # --treescan
     tree_rand: bool = False        # --treerand
+    tree_round_robin: bool = False  # --treeroundrob (shared blocks round-robin)
 
     # --- rwmix ---
     rwmix_pct: int = 0             # --rwmixpct
@@ -191,6 +194,7 @@ class BenchConfig:
     # --- netbench ---
     netbench: bool = False         # --netbench
     servers: list[str] = field(default_factory=list)  # --servers
+    num_servers: int = 0           # --numservers (limit servers list; 0 = all)
     resp_size: int = 1             # --respsize
     send_buf: int = 0              # --sendbuf
     recv_buf: int = 0              # --recvbuf
@@ -232,6 +236,22 @@ class BenchConfig:
     s3_olock_verify: bool = False  # --s3olockcfgverify
     s3_list_par: bool = False      # --s3listobjpar
     s3_sse: bool = False           # --s3sse (SSE-S3 AES256 header)
+    s3_sse_c_key: str = ""         # --s3sseckey (SSE-C base64 key)
+    s3_sse_kms_key: str = ""       # --s3ssekmskey (SSE-KMS key id)
+    s3_session_token: str = ""     # --s3sessiontoken (x-amz-security-token)
+    s3_chksum_algo: str = ""       # --s3chksumalgo (CRC32|CRC32C|SHA1|SHA256)
+    s3_acl_grantee: str = ""       # --s3aclgrantee (canned ACL or grantee name)
+    s3_acl_gtype: str = ""         # --s3aclgtype (id|emailAddress|uri|group)
+    s3_acl_put_inline: bool = False  # --s3aclputinl (ACL headers on object PUT)
+    s3_mpu_sharing: bool = False   # --s3mpusharing (workers share one MPU/object)
+    s3_mpu_size_var: int = 0       # --s3mpusizevar (max bytes subtracted per part)
+    s3_mpu_split: int = 0          # --s3mpusplit (part size override)
+    s3_no_mp_check: bool = False   # --s3nompcheck (skip 10k part-count check)
+    s3_single: bool = False        # --s3single (one shared client for all workers)
+    s3_target_gbps: int = 0        # --s3targetgbps (per-client throughput target)
+    s3_virt_addr: bool = False     # --s3virtaddr (virtual-hosted addressing)
+    s3_log: int = 0                # --s3log (client trace level, 0=off)
+    s3_log_prefix: str = ""        # --s3logprefix (trace file prefix)
 
     # --- misc ---
     config_file: str = ""          # -c/--configfile
@@ -263,6 +283,10 @@ class BenchConfig:
 
         if self.bench_mode == "posix" and self.paths:
             self.path_type = self._infer_path_type()
+
+        # --numservers: use only the first N hosts of the netbench servers list
+        if self.num_servers and self.servers:
+            self.servers = self.servers[:self.num_servers]
 
         if self.treefile and not self.tree_scan:
             from elbencho_amd.pathstore import parse_treefile
@@ -323,8 +347,23 @@ class BenchConfig:
         if self.bench_mode == "s3":
             if not self.s3_endpoints:
                 raise ConfigError("S3 mode requires --s3endpoints")
-            if (self.run_write or self.run_read) and self.files < 1:
+            if (self.run_write or self.run_read) and self.files < 1 \
+                    and not self.s3_mpu_sharing:
                 raise ConfigError("S3 object read/write requires -N/--files >= 1")
+            # reference ProgArgs.cpp:1505-1515: refuse MPUs above the S3
+            # 10,000-part limit unless --s3nompcheck
+            part_size = self.s3_mpu_split or self.block_size
+            if (self.run_write and not self.s3_no_mp_check and part_size
+                    and self.file_size > part_size
+                    and self.file_size / part_size > 10000):
+                raise ConfigError(
+                    "object size and part block size would result in a multipart "
+                    "upload exceeding the S3 limit of 10,000 parts. "
+                    "(--s3nompcheck disables this check.)")
+            if self.s3_chksum_algo and self.s3_chksum_algo.upper() not in (
+                    "CRC32", "CRC32C", "SHA1", "SHA256"):
+                raise ConfigError("--s3chksumalgo must be one of "
+                                  "CRC32, CRC32C, SHA1, SHA256")
 
         if self.threads < 1:
             raise ConfigError("number of threads must be >= 1")
@@ -477,6 +516,7 @@ class BenchConfig:
             tree_dirs=self.tree_dirs_resolved,
             tree_files=self.tree_files_resolved,
             sharesize=self.sharesize,
+            tree_round_robin=self.tree_round_robin,
             limit_read_bps=self.limit_read,
             limit_write_bps=self.limit_write,
             ignore_del_errors=self.ignore_del_errors,

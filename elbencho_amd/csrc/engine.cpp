@@ -1514,11 +1514,16 @@ void Worker::customTreeFiles(Phase phase)
     if (cfg.directIO) openFlags |= O_DIRECT;
 
     std::unique_ptr<OffsetGen> gen;
+    std::unique_ptr<OffsetGen> rrGen; // --treeroundrob strided generator
     size_t nonSharedIdx = 0; // running index over the non-shared sublist
 
     for (size_t i = 0; i < cfg.treeFiles.size(); i++) {
         const auto& [rel, size] = cfg.treeFiles[i];
         const bool shared = cfg.shareSize && size >= cfg.shareSize;
+        // round-robin block interleaving of shared files across ranks
+        // (reference --treeroundrob) instead of consecutive range slices
+        const bool roundRobin = shared && cfg.treeRoundRobin &&
+                                (phase == Phase::WRITE || phase == Phase::READ);
 
         uint64_t rangeStart = 0, rangeLen = size;
         if (!shared) {
@@ -1526,6 +1531,14 @@ void Worker::customTreeFiles(Phase phase)
             bool mine = (int)(nonSharedIdx % numRanks) == globalRank;
             nonSharedIdx++;
             if (!mine) continue;
+        } else if (roundRobin) {
+            // strided interleave over the whole file; skip files where this
+            // rank's first stride position is already past the end
+            if (!rrGen)
+                rrGen = std::make_unique<OffsetGenStrided>(cfg.blockSize, globalRank,
+                                                           numRanks);
+            rrGen->reset(0, size);
+            if (!rrGen->totalBytes()) continue;
         } else if (phase == Phase::WRITE || phase == Phase::READ) {
             // blockwise range slice of each shared file per rank
             fairShareSlice(size, rangeStart, rangeLen);
@@ -1551,10 +1564,16 @@ void Worker::customTreeFiles(Phase phase)
                 }
                 if (fd < 0) throwErrno("open", full);
                 try {
-                    if (!gen) gen = makeOffsetGen(rangeStart, rangeLen);
-                    else gen->reset(rangeStart, rangeLen);
+                    OffsetGen* og;
+                    if (roundRobin) {
+                        og = rrGen.get(); // reset above during the skip check
+                    } else {
+                        if (!gen) gen = makeOffsetGen(rangeStart, rangeLen);
+                        else gen->reset(rangeStart, rangeLen);
+                        og = gen.get();
+                    }
                     BlockSpec spec;
-                    while (gen->next(spec)) {
+                    while (og->next(spec)) {
                         uint64_t ioLen = std::min(spec.len, size - spec.offset);
                         auto t0 = lat ? Clock::now() : Clock::time_point();
                         ssize_t res = blockIO(isWrite, fd, 0, ioLen, spec.offset,

@@ -94,6 +94,9 @@ struct EngineConfig {
     std::vector<std::string> treeDirs;
     std::vector<std::pair<std::string, uint64_t>> treeFiles;
     uint64_t shareSize = 0;
+    // assign shared-file blocks round-robin to ranks instead of consecutive
+    // ranges (reference --treeroundrob, PathStore getWorkerSublistSharedRoundRobin)
+    bool treeRoundRobin = false;
 
     std::vector<int> gpuIDs; // empty = CPU buffers only
     bool gpuPinnedHostBufs = true;

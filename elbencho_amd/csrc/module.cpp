@@ -7,6 +7,8 @@
 #include <csignal>
 #include <cstdio>
 #include <cstdlib>
+#include <cstring>
+#include <mutex>
 #include <unistd.h>
 
 #include <fcntl.h>
@@ -81,6 +83,7 @@ EngineConfig configFromDict(const py::dict& d)
     if (d.contains("tree_files"))
         c.treeFiles = d["tree_files"].cast<std::vector<std::pair<std::string, uint64_t>>>();
     c.shareSize = getU64("sharesize", 0);
+    c.treeRoundRobin = getB("tree_round_robin", false);
     c.netbenchIsServer = getB("netbench_is_server", false);
     if (d.contains("netbench_servers"))
         c.netbenchServers = d["netbench_servers"].cast<std::vector<std::string>>();
@@ -197,6 +200,44 @@ PYBIND11_MODULE(_core, m)
     m.def("verify_checksum", [](py::bytes data, uint64_t fileOff, uint64_t salt) {
         std::string buf = data;
         return verifyChecksumCPU(buf.data(), buf.size(), fileOff, salt);
+    });
+
+    // CRC32C (Castagnoli, slice-by-8) for the S3 client's
+    // x-amz-checksum-crc32c header (--s3chksumalgo CRC32C)
+    m.def("crc32c", [](py::bytes data) {
+        static uint32_t table[8][256];
+        static std::once_flag once;
+        std::call_once(once, [] {
+            const uint32_t poly = 0x82F63B78u; // reflected CRC-32C
+            for (uint32_t i = 0; i < 256; i++) {
+                uint32_t c = i;
+                for (int k = 0; k < 8; k++)
+                    c = (c & 1) ? (poly ^ (c >> 1)) : (c >> 1);
+                table[0][i] = c;
+            }
+            for (uint32_t i = 0; i < 256; i++)
+                for (int s = 1; s < 8; s++)
+                    table[s][i] = table[0][table[s - 1][i] & 0xFF] ^
+                                  (table[s - 1][i] >> 8);
+        });
+        std::string buf = data;
+        const unsigned char* p = (const unsigned char*)buf.data();
+        size_t n = buf.size();
+        uint32_t crc = ~0u;
+        while (n >= 8) {
+            uint64_t w;
+            memcpy(&w, p, 8);
+            w ^= crc;
+            crc = table[7][w & 0xFF] ^ table[6][(w >> 8) & 0xFF] ^
+                  table[5][(w >> 16) & 0xFF] ^ table[4][(w >> 24) & 0xFF] ^
+                  table[3][(w >> 32) & 0xFF] ^ table[2][(w >> 40) & 0xFF] ^
+                  table[1][(w >> 48) & 0xFF] ^ table[0][(w >> 56) & 0xFF];
+            p += 8;
+            n -= 8;
+        }
+        while (n--)
+            crc = table[0][(crc ^ *p++) & 0xFF] ^ (crc >> 8);
+        return (uint32_t)~crc;
     });
 
     // --- offset generator test hook ---

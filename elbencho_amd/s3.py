@@ -46,7 +46,9 @@ class S3Client:
 
     def __init__(self, endpoint: str, access_key: str, secret_key: str,
                  region: str = "us-east-1", timeout: float = 60.0,
-                 extra_put_headers: dict[str, str] | None = None):
+                 extra_put_headers: dict[str, str] | None = None,
+                 session_token: str = "", virtual_addressing: bool = False,
+                 checksum_algo: str = "", trace=None):
         self.extra_put_headers = extra_put_headers or {}
         u = urllib.parse.urlparse(endpoint if "//" in endpoint else "http://" + endpoint)
         self.host = u.hostname or "localhost"
@@ -56,6 +58,13 @@ class S3Client:
         self.secret_key = secret_key
         self.region = region or "us-east-1"
         self.timeout = timeout
+        self.session_token = session_token        # --s3sessiontoken
+        self.virtual_addressing = virtual_addressing  # --s3virtaddr
+        self.checksum_algo = checksum_algo.upper()    # --s3chksumalgo
+        self.trace = trace                        # --s3log sink: fn(str)
+        # --s3single shares one client across worker threads; the connection
+        # handles one request at a time, so serialize (uncontended otherwise)
+        self.lock = threading.Lock()
         self._conn: Optional[http.client.HTTPConnection] = None
 
     # --- low level ---
@@ -69,15 +78,18 @@ class S3Client:
             self._conn = None
 
     def _sign(self, method: str, path: str, query: dict[str, str],
-              headers: dict[str, str], payload_hash: str) -> dict[str, str]:
+              headers: dict[str, str], payload_hash: str,
+              host: str | None = None) -> dict[str, str]:
         now = datetime.datetime.now(datetime.timezone.utc)
         amz_date = now.strftime("%Y%m%dT%H%M%SZ")
         datestamp = now.strftime("%Y%m%d")
 
         headers = dict(headers)
-        headers["host"] = f"{self.host}:{self.port}"
+        headers["host"] = host or f"{self.host}:{self.port}"
         headers["x-amz-date"] = amz_date
         headers["x-amz-content-sha256"] = payload_hash
+        if self.session_token:
+            headers["x-amz-security-token"] = self.session_token
 
         canonical_query = "&".join(
             f"{urllib.parse.quote(k, safe='')}={urllib.parse.quote(v, safe='')}"
@@ -113,25 +125,59 @@ class S3Client:
                 want_body: bool = True) -> tuple[int, bytes, dict[str, str]]:
         query = query or {}
         headers = headers or {}
+        host = None
+        if self.virtual_addressing and len(path) > 1:
+            # bucket as subdomain of the endpoint host (the TCP connection
+            # stays on the configured endpoint, like curl --resolve)
+            bucket, _, rest = path[1:].partition("/")
+            path = "/" + rest
+            host = f"{bucket}.{self.host}:{self.port}"
         payload_hash = hashlib.sha256(body).hexdigest() if body else EMPTY_SHA256
-        headers = self._sign(method, path, query, headers, payload_hash)
+        headers = self._sign(method, path, query, headers, payload_hash, host=host)
 
         qs = urllib.parse.urlencode(query)
         url = path + ("?" + qs if qs else "")
 
-        for attempt in (0, 1):  # one reconnect retry on stale connections
-            if self._conn is None:
-                self._connect()
-            try:
-                self._conn.request(method, url, body=body or None, headers=headers)
-                resp = self._conn.getresponse()
-                data = resp.read() if want_body else resp.read()
-                return resp.status, data, dict(resp.getheaders())
-            except (ConnectionError, http.client.HTTPException, OSError):
-                self.close()
-                if attempt:
-                    raise
+        with self.lock:
+            for attempt in (0, 1):  # one reconnect retry on stale connections
+                if self._conn is None:
+                    self._connect()
+                try:
+                    self._conn.request(method, url, body=body or None, headers=headers)
+                    resp = self._conn.getresponse()
+                    data = resp.read() if want_body else resp.read()
+                    if self.trace:
+                        self.trace(f"{method} {headers['host']}{url} "
+                                   f"len={len(body)} -> {resp.status}")
+                    return resp.status, data, dict(resp.getheaders())
+                except (ConnectionError, http.client.HTTPException, OSError):
+                    self.close()
+                    if attempt:
+                        if self.trace:
+                            self.trace(f"{method} {url} -> connection error")
+                        raise
         raise S3Error("unreachable")
+
+    # --s3chksumalgo: x-amz-sdk-checksum-algorithm + the computed
+    # x-amz-checksum-* trailer header (the reference delegates to the AWS SDK;
+    # this client computes the digest itself — CRC32C via the native core)
+    def _checksum_headers(self, body: bytes) -> dict[str, str]:
+        if not self.checksum_algo:
+            return {}
+        import base64
+        import struct
+        import zlib
+        algo = self.checksum_algo
+        if algo == "CRC32":
+            digest = struct.pack(">I", zlib.crc32(body) & 0xFFFFFFFF)
+        elif algo == "CRC32C":
+            digest = struct.pack(">I", load_core().crc32c(body))
+        elif algo == "SHA1":
+            digest = hashlib.sha1(body).digest()
+        else:  # SHA256
+            digest = hashlib.sha256(body).digest()
+        return {"x-amz-sdk-checksum-algorithm": algo,
+                f"x-amz-checksum-{algo.lower()}": base64.b64encode(digest).decode()}
 
     def _check(self, status: int, data: bytes, what: str):
         if status >= 300:
@@ -154,8 +200,10 @@ class S3Client:
 
     # --- object ops ---
     def put_object(self, bucket: str, key: str, body: bytes):
+        headers = dict(self.extra_put_headers)
+        headers.update(self._checksum_headers(body))
         status, data, _ = self.request("PUT", f"/{bucket}/{key}", body=body,
-                                       headers=dict(self.extra_put_headers))
+                                       headers=headers)
         self._check(status, data, f"put {bucket}/{key}")
 
     def get_object(self, bucket: str, key: str,
@@ -203,10 +251,11 @@ class S3Client:
         token_el = root.find(f"{ns}NextContinuationToken")
         return out, (token_el.text if token_el is not None else "")
 
-    # --- ACL (canned) ---
-    def put_object_acl(self, bucket: str, key: str, canned: str):
+    # --- ACL (canned or grantee headers) ---
+    def put_object_acl(self, bucket: str, key: str, acl):
+        headers = {"x-amz-acl": acl} if isinstance(acl, str) else dict(acl)
         status, data, _ = self.request("PUT", f"/{bucket}/{key}", query={"acl": ""},
-                                       headers={"x-amz-acl": canned})
+                                       headers=headers)
         self._check(status, data, f"put acl {bucket}/{key}")
 
     def get_object_acl(self, bucket: str, key: str) -> bytes:
@@ -214,9 +263,10 @@ class S3Client:
         self._check(status, data, f"get acl {bucket}/{key}")
         return data
 
-    def put_bucket_acl(self, bucket: str, canned: str):
+    def put_bucket_acl(self, bucket: str, acl):
+        headers = {"x-amz-acl": acl} if isinstance(acl, str) else dict(acl)
         status, data, _ = self.request("PUT", f"/{bucket}", query={"acl": ""},
-                                       headers={"x-amz-acl": canned})
+                                       headers=headers)
         self._check(status, data, f"put bucket acl {bucket}")
 
     def get_bucket_acl(self, bucket: str) -> bytes:
@@ -293,7 +343,8 @@ class S3Client:
 
     # --- multipart ---
     def create_multipart(self, bucket: str, key: str) -> str:
-        status, data, _ = self.request("POST", f"/{bucket}/{key}", query={"uploads": ""})
+        status, data, _ = self.request("POST", f"/{bucket}/{key}", query={"uploads": ""},
+                                       headers=dict(self.extra_put_headers))
         self._check(status, data, f"initiate multipart {bucket}/{key}")
         root = ET.fromstring(data)
         ns = root.tag.split("}")[0] + "}" if "}" in root.tag else ""
@@ -303,7 +354,8 @@ class S3Client:
                     body: bytes) -> str:
         status, data, headers = self.request(
             "PUT", f"/{bucket}/{key}",
-            query={"partNumber": str(part_num), "uploadId": upload_id}, body=body)
+            query={"partNumber": str(part_num), "uploadId": upload_id}, body=body,
+            headers=self._checksum_headers(body))
         self._check(status, data, f"upload part {part_num} of {bucket}/{key}")
         return headers.get("ETag", headers.get("etag", f'"{part_num}"'))
 
@@ -343,6 +395,58 @@ class S3Client:
                                        query={"uploadId": upload_id})
         if status not in (204, 404):
             self._check(status, data, f"abort multipart {bucket}/{key}")
+
+
+# canned ACL values accepted as --s3aclgrantee (reference ProgArgs.cpp:576-579)
+_CANNED_ACLS = {"private", "public-read", "public-read-write", "authenticated-read"}
+_ACL_PERM_HEADERS = {"READ": "x-amz-grant-read", "WRITE": "x-amz-grant-write",
+                     "READ_ACP": "x-amz-grant-read-acp",
+                     "WRITE_ACP": "x-amz-grant-write-acp",
+                     "FULL_CONTROL": "x-amz-grant-full-control"}
+
+
+def build_put_headers(cfg: BenchConfig) -> dict[str, str]:
+    """Upload headers from SSE (--s3sse/--s3sseckey/--s3ssekmskey) and
+    inline-ACL (--s3aclputinl) options."""
+    put_headers: dict[str, str] = {}
+    if cfg.s3_sse:  # SSE-S3 (AES256) passthrough header
+        put_headers["x-amz-server-side-encryption"] = "AES256"
+    if cfg.s3_sse_kms_key:  # --s3ssekmskey
+        put_headers["x-amz-server-side-encryption"] = "aws:kms"
+        put_headers["x-amz-server-side-encryption-aws-kms-key-id"] = cfg.s3_sse_kms_key
+    if cfg.s3_sse_c_key:  # --s3sseckey (base64 customer key + MD5)
+        import base64 as _b64
+        raw = _b64.b64decode(cfg.s3_sse_c_key)
+        put_headers["x-amz-server-side-encryption-customer-algorithm"] = "AES256"
+        put_headers["x-amz-server-side-encryption-customer-key"] = cfg.s3_sse_c_key
+        put_headers["x-amz-server-side-encryption-customer-key-MD5"] = \
+            _b64.b64encode(hashlib.md5(raw).digest()).decode()
+    if cfg.s3_acl_put_inline:  # --s3aclputinl: ACL headers on every PUT
+        acl = acl_value(cfg)
+        put_headers.update({"x-amz-acl": acl} if isinstance(acl, str) else acl)
+    return put_headers
+
+
+def acl_value(cfg: BenchConfig):
+    """Canned-ACL string or x-amz-grant-* header dict from --s3aclgrantee /
+    --s3aclgtype / --s3aclgrants (reference S3 ACL grant building)."""
+    grantee = cfg.s3_acl_grantee
+    if not grantee:
+        return cfg.s3_acl_grants or "private"  # legacy: grants as canned ACL
+    if grantee in _CANNED_ACLS:
+        return grantee
+    gtype = cfg.s3_acl_gtype or "id"
+    headers: dict[str, str] = {}
+    for perm in (p.strip().upper() for p in (cfg.s3_acl_grants or "").split(",")):
+        if not perm:
+            continue
+        if perm not in _ACL_PERM_HEADERS:
+            raise S3Error(f"unknown S3 ACL permission: {perm}")
+        headers[_ACL_PERM_HEADERS[perm]] = f'{gtype}="{grantee}"'
+    if not headers:
+        raise S3Error("--s3aclgrantee requires --s3aclgrants permissions "
+                      "(READ, WRITE, READ_ACP, WRITE_ACP, FULL_CONTROL)")
+    return headers
 
 
 # ---------------------------------------------------------------------------
@@ -419,11 +523,16 @@ class S3Worker(threading.Thread):
         key, secret = cfg.s3_key, cfg.s3_secret
         if runner.credentials:  # --s3credfile/--s3credlist round-robin
             key, secret = runner.credentials[self.rank % len(runner.credentials)]
-        put_headers = {}
-        if cfg.s3_sse:  # SSE-S3 (AES256) passthrough header
-            put_headers["x-amz-server-side-encryption"] = "AES256"
-        self.client = S3Client(ep, key, secret, cfg.s3_region,
-                               extra_put_headers=put_headers)
+        if cfg.s3_single:
+            # --s3single: one shared client instance for all worker threads
+            self.client = runner.shared_client
+        else:
+            self.client = S3Client(ep, key, secret, cfg.s3_region,
+                                   extra_put_headers=build_put_headers(cfg),
+                                   session_token=cfg.s3_session_token,
+                                   virtual_addressing=cfg.s3_virt_addr,
+                                   checksum_algo=cfg.s3_chksum_algo,
+                                   trace=runner.trace)
         self.core = load_core()
         # --gpuids + --verify: verify/generate object data with the gfx950
         # kernels in HBM (BASELINE config 5); one persistent context per worker
@@ -508,6 +617,17 @@ class S3Worker(threading.Thread):
         elif ph == "READ":
             self._get_objects()
         elif ph == "STAT":  # HEADOBJ
+            num_ranks = max(cfg.num_dataset_threads, 1)
+            if cfg.s3_mpu_sharing:
+                for i, (b, k) in enumerate(self.r.shared_objects):
+                    if i % num_ranks != self.rank % num_ranks:
+                        continue
+                    self._check_interrupt()
+                    t0 = time.monotonic()
+                    self.client.head_object(b, k)
+                    self.entry_lat.vec = _add_lat(self.entry_lat, t0)
+                    self.ops.entries += 1
+                return
             for name in self._object_names():
                 self._check_interrupt()
                 t0 = time.monotonic()
@@ -515,6 +635,17 @@ class S3Worker(threading.Thread):
                 self.entry_lat.vec = _add_lat(self.entry_lat, t0)
                 self.ops.entries += 1
         elif ph == "RMFILES":  # RMOBJECTS / MULTIDEL
+            num_ranks = max(cfg.num_dataset_threads, 1)
+            if cfg.s3_mpu_sharing:
+                for i, (b, k) in enumerate(self.r.shared_objects):
+                    if i % num_ranks != self.rank % num_ranks:
+                        continue
+                    self._check_interrupt()
+                    t0 = time.monotonic()
+                    self.client.delete_object(b, k)
+                    self.entry_lat.vec = _add_lat(self.entry_lat, t0)
+                    self.ops.entries += 1
+                return
             if cfg.s3_multi_del > 0:
                 batch: list[str] = []
                 for name in self._object_names():
@@ -537,21 +668,25 @@ class S3Worker(threading.Thread):
         elif ph == "LISTOBJ":
             self._list_objects()
         elif ph == "PUTOBJACL":
+            acl = acl_value(cfg)
             for name in self._object_names():
                 self._check_interrupt()
-                self.client.put_object_acl(self._bucket(), name, cfg.s3_acl_grants or "private")
+                self.client.put_object_acl(self._bucket(), name, acl)
                 self.ops.entries += 1
         elif ph == "GETOBJACL":
+            expect = acl_value(cfg)
+            expect_str = expect if isinstance(expect, str) else cfg.s3_acl_grantee
             for name in self._object_names():
                 self._check_interrupt()
                 acl = self.client.get_object_acl(self._bucket(), name)
-                if cfg.s3_acl_verify and (cfg.s3_acl_grants or "private") not in acl.decode():
+                if cfg.s3_acl_verify and expect_str not in acl.decode():
                     raise S3Error(f"object ACL verification failed for {name}")
                 self.ops.entries += 1
         elif ph == "PUTBACL":
             if self.local_rank == 0:
+                acl = acl_value(cfg)
                 for b in self.r.buckets:
-                    self.client.put_bucket_acl(b, cfg.s3_acl_grants or "private")
+                    self.client.put_bucket_acl(b, acl)
                     self.ops.entries += 1
         elif ph == "GETBACL":
             if self.local_rank == 0:
@@ -653,15 +788,42 @@ class S3Worker(threading.Thread):
             return self.core.fill_checksum(length, obj_off, cfg.verify)
         return bytes(self.r.rand_block[:length])
 
+    def _part_sizes(self, size: int, seed_extra: int = 0) -> list[tuple[int, int]]:
+        """[(offset, length)] parts of one object: part size is -b (or
+        --s3mpusplit) with optional --s3mpusizevar random shrink per part;
+        the final part absorbs the difference (reference s3MpuSizeVariance).
+        seed_extra must be identical across ranks for shared uploads."""
+        cfg = self.r.cfg
+        ps = cfg.s3_mpu_split or cfg.block_size
+        out = []
+        off = 0
+        if cfg.s3_mpu_size_var:
+            import random as _random
+            rng = _random.Random((cfg.bench_seed or 1) ^ 0x9E3779B9 ^ seed_extra)
+            while off < size:
+                ln = min(ps, size - off)
+                if size - off > ln:  # not the last part: shrink by variance
+                    ln = max(1, ln - rng.randrange(cfg.s3_mpu_size_var + 1))
+                out.append((off, ln))
+                off += ln
+        else:
+            while off < size:
+                ln = min(ps, size - off)
+                out.append((off, ln))
+                off += ln
+        return out
+
     def _put_objects(self):
         cfg = self.r.cfg
+        if cfg.s3_mpu_sharing:
+            return self._put_objects_shared()
         size = cfg.file_size
-        bs = cfg.block_size
+        ps = cfg.s3_mpu_split or cfg.block_size
         for name in self._object_names():
             self._check_interrupt()
             te = time.monotonic()
             bucket = self._bucket()
-            if size <= bs:  # single part
+            if size <= ps and not cfg.s3_mpu_size_var:  # single part
                 t0 = time.monotonic()
                 self._oplog("PutObject", name, 0, size, True)
                 self.client.put_object(bucket, name, self._make_block(size, 0))
@@ -672,12 +834,10 @@ class S3Worker(threading.Thread):
             else:  # multipart: block size = part size (reference -b semantics)
                 upload_id = self.client.create_multipart(bucket, name)
                 parts = []
-                off = 0
-                part_num = 1
                 try:
-                    while off < size:
+                    for part_num, (off, ln) in enumerate(
+                            self._part_sizes(size, self.rank), 1):
                         self._check_interrupt()
-                        ln = min(bs, size - off)
                         t0 = time.monotonic()
                         etag = self.client.upload_part(bucket, name, upload_id,
                                                        part_num, self._make_block(ln, off))
@@ -685,8 +845,6 @@ class S3Worker(threading.Thread):
                         parts.append((part_num, etag))
                         self.ops.bytes += ln
                         self.ops.iops += 1
-                        off += ln
-                        part_num += 1
                     if not cfg.s3_no_mpu_compl:
                         self.client.complete_multipart(bucket, name, upload_id, parts)
                     # else: left open for a later S3MPUCOMPLETE phase
@@ -696,10 +854,89 @@ class S3Worker(threading.Thread):
             self.entry_lat.vec = _add_lat(self.entry_lat, te)
             self.ops.entries += 1
 
+    def _put_objects_shared(self):
+        """--s3mpusharing: every worker uploads its round-robin share of parts
+        of each named object through one shared multipart upload; whichever
+        worker records the final part completes the upload (reference
+        S3UploadStore first-writer-wins + LocalWorker.cpp:5455)."""
+        cfg = self.r.cfg
+        size = cfg.file_size
+        ps = cfg.s3_mpu_split or cfg.block_size
+        num_ranks = max(cfg.num_dataset_threads, 1)
+        store = self.r.upload_store
+        for bucket, key in self.r.shared_objects:
+            self._check_interrupt()
+            te = time.monotonic()
+            # seed by object name so every rank computes identical part bounds
+            name_seed = int(hashlib.md5(f"{bucket}/{key}".encode()).hexdigest()[:8], 16)
+            all_parts = [(n, off, ln) for n, (off, ln)
+                         in enumerate(self._part_sizes(size, name_seed), 1)]
+            upload_id = store.get_or_create(self.client, bucket, key, len(all_parts))
+            did_any = False
+            for part_num, off, ln in all_parts:
+                if (part_num - 1) % num_ranks != self.rank % num_ranks:
+                    continue
+                self._check_interrupt()
+                t0 = time.monotonic()
+                self._oplog("UploadPart", f"{bucket}/{key}", off, ln, True)
+                etag = self.client.upload_part(bucket, key, upload_id, part_num,
+                                               self._make_block(ln, off))
+                self._oplog("UploadPart", f"{bucket}/{key}", off, ln, False)
+                self.io_lat.vec = _add_lat(self.io_lat, t0)
+                self.ops.bytes += ln
+                self.ops.iops += 1
+                did_any = True
+                is_last = store.add_part(bucket, key, part_num, etag)
+                if is_last and not cfg.s3_no_mpu_compl:
+                    self.client.complete_multipart(bucket, key, upload_id,
+                                                   store.get_parts(bucket, key))
+            if did_any:
+                self.entry_lat.vec = _add_lat(self.entry_lat, te)
+                self.ops.entries += 1
+
+    def _get_objects_shared(self):
+        """--s3mpusharing READ: each rank downloads its round-robin block
+        share of every named shared object (ranged GETs)."""
+        cfg = self.r.cfg
+        size = cfg.file_size
+        bs = cfg.block_size
+        num_ranks = max(cfg.num_dataset_threads, 1)
+        for bucket, key in self.r.shared_objects:
+            self._check_interrupt()
+            te = time.monotonic()
+            did_any = False
+            off = 0
+            blk = 0
+            while off < size:
+                ln = min(bs, size - off)
+                if blk % num_ranks == self.rank % num_ranks:
+                    self._check_interrupt()
+                    t0 = time.monotonic()
+                    data = self.client.get_object(bucket, key, (off, off + ln - 1))
+                    self.io_lat.vec = _add_lat(self.io_lat, t0)
+                    if len(data) != ln:
+                        raise S3Error(f"short ranged read of {bucket}/{key}: "
+                                      f"{len(data)} != {ln}")
+                    if cfg.verify >= 0 and not cfg.s3_fastget:
+                        bad = self.core.verify_checksum(data, off, cfg.verify)
+                        if bad != 2**64 - 1:
+                            raise S3Error(f"S3 data verification failed for "
+                                          f"{bucket}/{key} at object offset {bad}")
+                    self.ops.bytes += ln
+                    self.ops.iops += 1
+                    did_any = True
+                off += ln
+                blk += 1
+            if did_any:
+                self.entry_lat.vec = _add_lat(self.entry_lat, te)
+                self.ops.entries += 1
+
     def _get_objects(self):
         cfg = self.r.cfg
         size = cfg.file_size
         bs = cfg.block_size
+        if cfg.s3_mpu_sharing:
+            return self._get_objects_shared()
         if cfg.s3_rand_obj:
             return self._get_random_objects()
         for name in self._object_names():
@@ -824,10 +1061,47 @@ class S3Runner:
         if not cfg.s3_endpoints:
             raise S3Error("S3 mode requires --s3endpoints")
         self.cfg = cfg
-        self.buckets = [p[len("s3://"):] if p.startswith("s3://") else p
-                        for p in cfg.paths]
+        stripped = [p[len("s3://"):] if p.startswith("s3://") else p
+                    for p in cfg.paths]
+        # --s3mpusharing: "bucket/object" path entries name shared objects that
+        # all workers upload/download together (reference s3mpusharing mode,
+        # object names as parameters)
+        self.shared_objects: list[tuple[str, str]] = []
+        if cfg.s3_mpu_sharing:
+            for p in stripped:
+                if "/" not in p:
+                    raise S3Error("--s3mpusharing requires bucket/object paths")
+                b, _, k = p.partition("/")
+                self.shared_objects.append((b, k))
+            self.buckets = sorted({b for b, _ in self.shared_objects})
+        else:
+            self.buckets = stripped
         if not self.buckets:
             raise S3Error("S3 mode requires s3://bucket paths")
+        # --s3log: client request trace (reference AWS SDK logging)
+        self.trace = None
+        self._trace_file = None
+        if cfg.s3_log > 0:
+            import datetime as _dt
+            prefix = cfg.s3_log_prefix or "s3_client_"
+            path = f"{prefix}{_dt.date.today().strftime('%Y%m%d')}.log"
+            self._trace_file = open(path, "a")
+            tl = threading.Lock()
+
+            def _trace(line: str, _f=self._trace_file, _l=tl):
+                with _l:
+                    _f.write(f"{time.time():.6f} {line}\n")
+                    _f.flush()
+            self.trace = _trace
+        # --s3single: one client shared by every worker thread
+        self.shared_client = None
+        if cfg.s3_single:
+            self.shared_client = S3Client(
+                cfg.s3_endpoints[0], cfg.s3_key, cfg.s3_secret, cfg.s3_region,
+                extra_put_headers=build_put_headers(cfg),
+                session_token=cfg.s3_session_token,
+                virtual_addressing=cfg.s3_virt_addr,
+                checksum_algo=cfg.s3_chksum_algo, trace=self.trace)
         self.credentials: list[tuple[str, str]] = []
         if cfg.s3_cred_list:
             for ent in cfg.s3_cred_list.split(","):
@@ -929,6 +1203,13 @@ class S3Runner:
 
     def planned_work(self, phase_name: str) -> tuple[int, int]:
         cfg = self.cfg
+        if cfg.s3_mpu_sharing:
+            nobj = len(self.shared_objects)
+            if phase_name in ("WRITE", "READ"):
+                return nobj, nobj * cfg.file_size
+            if phase_name in ("STAT", "RMFILES"):
+                return nobj, 0
+            return 0, 0
         nobj = max(cfg.dirs, 1) * cfg.files * cfg.threads
         if phase_name in ("WRITE", "READ"):
             return nobj, nobj * cfg.file_size
@@ -950,3 +1231,6 @@ class S3Runner:
                 self.upload_store.abort_unfinished(self.workers[0].client)
             except (S3Error, OSError):
                 pass
+        if self._trace_file:
+            self._trace_file.close()
+            self._trace_file = None

@@ -254,8 +254,14 @@ def run_service(cfg: BenchConfig) -> int:
 
     state = ServiceState(cfg)
     Handler.state = state
-    server = ThreadingHTTPServer(("", cfg.service_port), Handler)
-    server.daemon_threads = True
+    if cfg.alt_http_svc:
+        # --althttpsvc: alternative single-threaded server implementation
+        # (reference HTTPServiceUWS analogue, "for testing")
+        from http.server import HTTPServer
+        server = HTTPServer(("", cfg.service_port), Handler)
+    else:
+        server = ThreadingHTTPServer(("", cfg.service_port), Handler)
+        server.daemon_threads = True
 
     print(f"elbencho-amd service v{VERSION} listening on port {cfg.service_port}")
 

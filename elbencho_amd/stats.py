@@ -230,7 +230,11 @@ def print_phase_results(cfg: BenchConfig, r: PhaseResults, out=None) -> None:
                          r.per_sec_last(r.entries) // cfg.files))
 
     is_rwmix = bool(r.rm_bytes or r.rm_iops or r.rm_entries)
-    mib = 1024 * 1024
+    # --base10: MB/s (10^6) instead of MiB/s (2^20) in console output
+    # (reference Statistics.cpp:190-191; CSV schema stays MiB)
+    mib = 1000 * 1000 if cfg.show_base10 else 1024 * 1024
+    tp_unit = "MB/s" if cfg.show_base10 else "MiB/s"
+    tot_unit = "MB" if cfg.show_base10 else "MiB"
 
     if r.iops:
         # suppress IOPS when it would equal files/s (dir mode, 1 block per file)
@@ -243,19 +247,19 @@ def print_phase_results(cfg: BenchConfig, r: PhaseResults, out=None) -> None:
                      r.per_sec_last(r.iops + r.rm_iops)))
 
     if r.bytes:
-        rows.append(("MiB/s write" if is_rwmix else "Throughput MiB/s",
+        rows.append((f"{tp_unit} write" if is_rwmix else f"Throughput {tp_unit}",
                      r.per_sec_first(r.sw_bytes) // mib, r.per_sec_last(r.bytes) // mib))
     if is_rwmix and r.rm_bytes:
-        rows.append(("MiB/s read", r.per_sec_first(r.rm_sw_bytes) // mib,
+        rows.append((f"{tp_unit} read", r.per_sec_first(r.rm_sw_bytes) // mib,
                      r.per_sec_last(r.rm_bytes) // mib))
-        rows.append(("MiB/s total", r.per_sec_first(r.sw_bytes + r.rm_sw_bytes) // mib,
+        rows.append((f"{tp_unit} total", r.per_sec_first(r.sw_bytes + r.rm_sw_bytes) // mib,
                      r.per_sec_last(r.bytes + r.rm_bytes) // mib))
 
     if r.bytes:
-        rows.append(("MiB write" if is_rwmix else "Total MiB",
+        rows.append((f"{tot_unit} write" if is_rwmix else f"Total {tot_unit}",
                      r.sw_bytes // mib, r.bytes // mib))
     if is_rwmix and r.rm_bytes:
-        rows.append(("MiB read", r.rm_sw_bytes // mib, r.rm_bytes // mib))
+        rows.append((f"{tot_unit} read", r.rm_sw_bytes // mib, r.rm_bytes // mib))
 
     if r.entries:
         rows.append((f"{entry_type} total", r.sw_entries, r.entries))
@@ -540,7 +544,10 @@ class LiveStatsPrinter:
             pct = min(100, 100 * poll["entries"] // self.planned_entries)
             parts.append(f"{pct}%")
         if bps:
-            parts.append(f"{bps // (1024 * 1024)} MiB/s")
+            if self.cfg.show_base10:
+                parts.append(f"{bps // (1000 * 1000)} MB/s")
+            else:
+                parts.append(f"{bps // (1024 * 1024)} MiB/s")
         if iops:
             parts.append(f"{iops} IOPS")
         if eps:

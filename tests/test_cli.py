@@ -128,3 +128,34 @@ def test_config_file(tmp_path):
     rc = main(["--nolive", "-c", str(conf), str(f)])
     assert rc == 0
     assert f.stat().st_size == 128 * 1024
+
+
+def test_base10_output(tmp_path, capsys):
+    f = tmp_path / "b10"
+    rc = main(["-w", "-t", "1", "-s", "4m", "-b", "1m", "--base10", "--nolive",
+               str(f)])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "MB/s" in out
+    assert "MiB/s" not in out
+
+
+def test_path_option_and_nodetach_alias(tmp_path):
+    from elbencho_amd.cli import args_to_config, build_parser
+    p = build_parser()
+    cfg = args_to_config(p.parse_args(
+        ["-w", "-s", "1m", "--path", str(tmp_path / "x"), "--path",
+         str(tmp_path / "y")]))
+    assert len(cfg.paths) == 2
+    # --nodetach is the reference-compatible alias of --foreground
+    args = p.parse_args(["--service", "--nodetach"])
+    assert args.foreground
+
+
+def test_numservers_limits_servers():
+    from elbencho_amd.cli import args_to_config, build_parser
+    p = build_parser()
+    cfg = args_to_config(p.parse_args(
+        ["--netbench", "-w", "-s", "1m", "-b", "64k",
+         "--hosts", "h1,h2,h3", "--servers", "h1,h2", "--numservers", "1"]))
+    assert cfg.servers == ["h1:1611"] or cfg.servers == ["h1"]

@@ -96,3 +96,29 @@ def test_treefile_cli_run(tmp_path):
                "--treefile", str(tf), str(base)])
     assert rc == 0
     assert not (base / "d").exists()
+
+
+def test_custom_tree_round_robin(core, tmp_path):
+    """--treeroundrob: shared-file blocks interleave round-robin across ranks
+    and still cover the file exactly (reference
+    PathStore getWorkerSublistSharedRoundRobin)."""
+    base = tmp_path / "bench"
+    base.mkdir()
+    size = 1024 * 1024 + 777  # odd tail
+    files = [("big", size)]
+    cfg = dict(paths=[str(base)], path_type="dir", threads=3, num_dataset_threads=3,
+               block_size=64 * 1024, tree_files=files, tree_dirs=[],
+               sharesize=512 * 1024, tree_round_robin=True, verify_salt=11)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    res = run_phase(core, eng, "WRITE")
+    assert sum(r["bytes"] for r in res) == size
+    # every rank did interleaved work (21 blocks over 3 ranks: 7 each)
+    by_rank = {r["rank"]: r["bytes"] for r in res}
+    assert all(b > 0 for b in by_rank.values())
+    with open(base / "big", "rb") as f:
+        data = f.read()
+    assert len(data) == size
+    assert core.verify_checksum(data, 0, 11) == 2**64 - 1
+    run_phase(core, eng, "READ")  # verify on read too
+
