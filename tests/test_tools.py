@@ -59,3 +59,31 @@ def test_launcher_script():
 
 def test_hdfs_unsupported(tmp_path):
     assert _run_cli(["--hdfs", "-w", "-s", "1m", "--nolive", str(tmp_path / "f")]) == 1
+
+
+def test_sweep_dry_run(tmp_path):
+    """elbencho-amd-sweep (contrib/storage_sweep analogue): dry run prints one
+    command per power-of-two point with scaled file counts."""
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "elbencho-amd-sweep"), str(tmp_path),
+         "-r", "s", "--budget", "8K", "-t", "2", "-n"],
+        capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
+    lines = [ln for ln in out.stdout.splitlines() if ln.strip()]
+    assert len(lines) == 4  # 1K, 2K, 4K, 8K points within the 8K budget
+    assert "-s 1024" in lines[0] and "--dirsharing" in lines[0]
+    assert "--direct" in lines[-1]      # >= fs block size
+    assert "--direct" not in lines[0]   # tiny files skip O_DIRECT
+
+
+def test_sweep_real_micro(tmp_path):
+    csvf = tmp_path / "sweep.csv"
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "elbencho-amd-sweep"),
+         str(tmp_path / "bench"), "-r", "s", "--budget", "4K", "-t", "1",
+         "-B", "--csv", str(csvf), "--plot", str(tmp_path / "plot")],
+        capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr + out.stdout
+    assert "Gbps" in out.stdout
+    assert csvf.exists()
+    assert (tmp_path / "plot.gp").exists()
