@@ -517,6 +517,7 @@ class BenchConfig:
             tree_files=self.tree_files_resolved,
             sharesize=self.sharesize,
             tree_round_robin=self.tree_round_robin,
+            tree_rand=self.tree_rand,
             limit_read_bps=self.limit_read,
             limit_write_bps=self.limit_write,
             ignore_del_errors=self.ignore_del_errors,

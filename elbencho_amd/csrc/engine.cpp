@@ -20,6 +20,7 @@
 #include <unistd.h>
 
 #include <algorithm>
+#include <random>
 #include <cerrno>
 #include <cstdio>
 #include <cstring>
@@ -1517,8 +1518,16 @@ void Worker::customTreeFiles(Phase phase)
     std::unique_ptr<OffsetGen> rrGen; // --treeroundrob strided generator
     size_t nonSharedIdx = 0; // running index over the non-shared sublist
 
+    // pass 1: ownership/partitioning in treefile order (must be identical on
+    // every rank), collecting this rank's work items
+    struct TreeWorkItem {
+        size_t idx;
+        uint64_t rangeStart, rangeLen;
+        bool roundRobin;
+    };
+    std::vector<TreeWorkItem> work;
     for (size_t i = 0; i < cfg.treeFiles.size(); i++) {
-        const auto& [rel, size] = cfg.treeFiles[i];
+        const uint64_t size = cfg.treeFiles[i].second;
         const bool shared = cfg.shareSize && size >= cfg.shareSize;
         // round-robin block interleaving of shared files across ranks
         // (reference --treeroundrob) instead of consecutive range slices
@@ -1547,8 +1556,26 @@ void Worker::customTreeFiles(Phase phase)
             // stat/unlink of shared files: one rank per file
             if ((int)(i % numRanks) != globalRank) continue;
         }
+        work.push_back({i, rangeStart, rangeLen, roundRobin});
+    }
 
-        if ((i % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+    // --treerand: randomize this worker's processing order (reference
+    // PathStore randomShuffle, LocalWorker.cpp:1591); partitioning above is
+    // untouched, so coverage stays exact
+    if (cfg.treeRandomize) {
+        std::mt19937_64 shuffleRng(cfg.benchSeed ^ (0x7EEF11EULL + globalRank));
+        std::shuffle(work.begin(), work.end(), shuffleRng);
+    }
+
+    for (size_t w = 0; w < work.size(); w++) {
+        const size_t i = work[w].idx;
+        const auto& [rel, size] = cfg.treeFiles[i];
+        const uint64_t rangeStart = work[w].rangeStart;
+        const uint64_t rangeLen = work[w].rangeLen;
+        const bool roundRobin = work[w].roundRobin;
+        if (roundRobin) rrGen->reset(0, size); // re-arm for this file
+
+        if ((w % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
         std::string full = base + "/" + rel;
         auto tEntry0 = lat ? Clock::now() : Clock::time_point();
 

@@ -97,6 +97,9 @@ struct EngineConfig {
     // assign shared-file blocks round-robin to ranks instead of consecutive
     // ranges (reference --treeroundrob, PathStore getWorkerSublistSharedRoundRobin)
     bool treeRoundRobin = false;
+    // randomize each worker's custom-tree processing order (reference
+    // --treerand, PathStore randomShuffle)
+    bool treeRandomize = false;
 
     std::vector<int> gpuIDs; // empty = CPU buffers only
     bool gpuPinnedHostBufs = true;

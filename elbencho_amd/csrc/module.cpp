@@ -84,6 +84,7 @@ EngineConfig configFromDict(const py::dict& d)
         c.treeFiles = d["tree_files"].cast<std::vector<std::pair<std::string, uint64_t>>>();
     c.shareSize = getU64("sharesize", 0);
     c.treeRoundRobin = getB("tree_round_robin", false);
+    c.treeRandomize = getB("tree_rand", false);
     c.netbenchIsServer = getB("netbench_is_server", false);
     if (d.contains("netbench_servers"))
         c.netbenchServers = d["netbench_servers"].cast<std::vector<std::string>>();

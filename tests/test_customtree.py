@@ -122,3 +122,23 @@ def test_custom_tree_round_robin(core, tmp_path):
     assert core.verify_checksum(data, 0, 11) == 2**64 - 1
     run_phase(core, eng, "READ")  # verify on read too
 
+
+
+def test_custom_tree_randomized_order(core, tmp_path):
+    """--treerand shuffles each worker's processing order; partitioning and
+    coverage stay exact (reference PathStore randomShuffle)."""
+    base = tmp_path / "bench"
+    base.mkdir()
+    files = [(f"f{i}", 16 * 1024) for i in range(20)]
+    cfg = dict(paths=[str(base)], path_type="dir", threads=2, num_dataset_threads=2,
+               block_size=16 * 1024, tree_files=files, tree_dirs=[],
+               tree_rand=True, verify_salt=5)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    res = run_phase(core, eng, "WRITE")
+    assert sum(r["entries"] for r in res) == 20
+    for p, s in files:
+        assert os.path.getsize(base / p) == s
+        with open(base / p, "rb") as f:
+            assert core.verify_checksum(f.read(), 0, 5) == 2**64 - 1
+    run_phase(core, eng, "READ")
