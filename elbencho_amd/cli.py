@@ -347,8 +347,12 @@ def build_parser() -> argparse.ArgumentParser:
                    help="Delete objects in multi-delete batches of N.")
     g.add_argument("--s3fastget", action="store_true",
                    help="Discard downloaded objects instead of keeping them in RAM.")
-    g.add_argument("--s3fastput", action="store_true", help=argparse.SUPPRESS)
-    g.add_argument("--s3sign", type=int, default=0, help=argparse.SUPPRESS)
+    g.add_argument("--s3fastput", action="store_true",
+                   help="Reduce CPU overhead for uploads: enables --s3sign 2 "
+                        "(unsigned payloads) and --s3nocompress.")
+    g.add_argument("--s3sign", type=int, default=0, metavar="N",
+                   help="S3 payload signing policy: 0=RequestDependent, 1=Always, "
+                        "2=Never (skips the per-block payload SHA256). (Default: 0)")
     g.add_argument("--s3maxconns", type=int, default=0, help=argparse.SUPPRESS)
     g.add_argument("--s3ignoreerrors", action="store_true",
                    help="Record S3 op errors per worker but keep the phase running.")
@@ -561,6 +565,13 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.s3_list_verify = args.s3listverify
     cfg.s3_multi_del = args.s3multidel
     cfg.s3_fastget = args.s3fastget
+    cfg.s3_fastput = args.s3fastput
+    cfg.s3_sign_policy = args.s3sign
+    cfg.s3_no_compress = args.s3nocompress
+    cfg.s3_max_conns = args.s3maxconns
+    if args.s3fastput:  # reference ProgArgs.cpp:1293-1296
+        cfg.s3_sign_policy = 2
+        cfg.s3_no_compress = True
     cfg.s3_acl_put = args.s3aclput
     cfg.s3_acl_get = args.s3aclget
     cfg.s3_acl_verify = args.s3aclverify

@@ -355,3 +355,15 @@ def test_s3_10k_part_check():
         args_to_config(p.parse_args(base))
     cfg = args_to_config(p.parse_args(base + ["--s3nompcheck"]))
     assert cfg.s3_no_mp_check
+
+
+def test_s3_fastput_unsigned_payload(mock_s3, capsys):
+    """--s3fastput: uploads go out with UNSIGNED-PAYLOAD (no per-block
+    SHA256); the SigV4 mock accepts the signature."""
+    rc = _cli(mock_s3, ["-d", "-w", "-r", "-t", "1", "-N", "2", "-s", "64k",
+                        "-b", "64k", "--s3fastput", "s3://fastbkt"])
+    assert rc == 0
+    c = S3Client(mock_s3, ACCESS_KEY, SECRET_KEY, sign_payload=False)
+    c.create_bucket("fp2")
+    c.put_object("fp2", "o", b"x" * 100)
+    assert c.get_object("fp2", "o") == b"x" * 100
