@@ -1074,6 +1074,17 @@ void Worker::fileModeBlocksUring(bool isWrite)
     IoUring ring;
     ring.init(depth);
 
+    // fixed buffers + registered files: one-time page pin / fd ref instead of
+    // per-op (EB_URING_NOFIXED=1 disables; silent fallback if the kernel
+    // refuses, e.g. RLIMIT_MEMLOCK without CAP_IPC_LOCK)
+    if (!getenv("EB_URING_NOFIXED")) {
+        std::vector<struct iovec> iovs(depth);
+        for (int s = 0; s < depth; s++)
+            iovs[s] = {hostBufs[s], (size_t)bs};
+        ring.registerBuffers(iovs.data(), depth);
+        ring.registerFiles(fg.fds.data(), (unsigned)fg.fds.size());
+    }
+
     struct SlotState {
         uint64_t inFileOff = 0;
         uint64_t len = 0;
@@ -1117,8 +1128,10 @@ void Worker::fileModeBlocksUring(bool isWrite)
         if (eng.opsLog.isEnabled())
             eng.opsLog.log(globalRank, blockWrite ? "uring_write" : "uring_read",
                            cfg.paths[fileIdx], inFileOff, ioLen, true, false);
-        if (!ring.prep(blockWrite, fg.fds[fileIdx], hostBufs[slot], ioLen, inFileOff,
-                       (uint64_t)slot))
+        int ringFd = ring.hasFixedFiles() ? (int)fileIdx : fg.fds[fileIdx];
+        if (!ring.prep(blockWrite, ringFd, hostBufs[slot], ioLen, inFileOff,
+                       (uint64_t)slot, ring.hasFixedBuffers() ? slot : -1,
+                       ring.hasFixedFiles()))
             throw WorkerError("io_uring SQ unexpectedly full");
         inFlight++;
         return true;
