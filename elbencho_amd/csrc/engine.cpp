@@ -1085,6 +1085,106 @@ void Worker::fileModeBlocksUring(bool isWrite)
         ring.registerFiles(fg.fds.data(), (unsigned)fg.fds.size());
     }
 
+    // Batched GPU staging for small-block async reads (BASELINE config 3:
+    // 4K random read, io_uring QD128, buffers in HBM): per-block 4K
+    // hipMemcpyAsync caps at ~150K IOPS (launch-bound); instead the ring is
+    // split into two halves — when every read of a half has completed, ONE
+    // ranged H2D covers all its contiguous slots while the other half's
+    // storage reads stay in flight (same geometry as the sync path's
+    // half-ring batching above).
+    const int gpuBatch = gpu ? gpuBatchSlots(bs) : 1;
+    const bool batchedRead = gpu && !isWrite && cfg.verifySalt < 0 &&
+                             gpuBatch >= 2 && depth >= 4;
+    if (batchedRead) {
+        const int halfSize = std::min(depth / 2, 64);
+        struct Op {
+            uint64_t off = 0, len = 0;
+            Clock::time_point start;
+            int fileIdx = 0;
+        };
+        std::vector<Op> ops(2 * halfSize);
+        std::vector<IoUring::Completion> bcomps(depth);
+        int submitted[2] = {0, 0}, done[2] = {0, 0};
+        bool exhausted2 = false;
+        BlockSpec bspec;
+
+        auto fillHalf = [&](int h) -> int {
+            int base = h * halfSize;
+            int n = 0;
+            while (n < halfSize && !exhausted2) {
+                uint64_t fileIdx = 0, inFileOff = 0, ioLen = 0;
+                for (;;) {
+                    if (!gen->next(bspec)) { exhausted2 = true; break; }
+                    if (mapBlock(bspec, fileIdx, inFileOff, ioLen)) break;
+                }
+                if (exhausted2) break;
+                rateLimiter.wait(ioLen);
+                int slot = base + n;
+                ops[slot] = {inFileOff, ioLen,
+                             lat ? Clock::now() : Clock::time_point(), (int)fileIdx};
+                if (eng.opsLog.isEnabled())
+                    eng.opsLog.log(globalRank, "uring_read", cfg.paths[fileIdx],
+                                   inFileOff, ioLen, true, false);
+                int rfd = ring.hasFixedFiles() ? (int)fileIdx : fg.fds[fileIdx];
+                if (!ring.prep(false, rfd, hostBufs[slot], ioLen, inFileOff,
+                               (uint64_t)slot, ring.hasFixedBuffers() ? slot : -1,
+                               ring.hasFixedFiles()))
+                    throw WorkerError("io_uring SQ unexpectedly full");
+                n++;
+            }
+            submitted[h] = n;
+            done[h] = 0;
+            return n;
+        };
+
+        int inFlightB = fillHalf(0) + fillHalf(1);
+        uint64_t opCountB = 0;
+        while (inFlightB > 0) {
+            checkInterrupt();
+            ring.submitAndWait(1);
+            unsigned n = ring.reap(bcomps.data(), depth);
+            for (unsigned i = 0; i < n; i++) {
+                int slot = (int)bcomps[i].userData;
+                const Op& op = ops[slot];
+                if (eng.opsLog.isEnabled())
+                    eng.opsLog.log(globalRank, "uring_read", cfg.paths[op.fileIdx],
+                                   op.off, op.len, false, bcomps[i].res < 0);
+                if (bcomps[i].res < 0)
+                    throw WorkerError(std::string("async read failed. Path: ") +
+                                      cfg.paths[op.fileIdx] + "; SysErr: " +
+                                      strerror(-bcomps[i].res));
+                if ((uint64_t)bcomps[i].res != op.len)
+                    throw WorkerError("unexpected short async read. Path: " +
+                                      cfg.paths[op.fileIdx]);
+                if (lat)
+                    addIoLat((uint64_t)std::chrono::duration_cast<
+                        std::chrono::microseconds>(Clock::now() - op.start).count());
+                liveOps.bytes.fetch_add(op.len, std::memory_order_relaxed);
+                liveOps.iops.fetch_add(1, std::memory_order_relaxed);
+                done[slot / halfSize]++;
+                inFlightB--;
+                if ((opCountB++ % INTERRUPT_CHECK_INTERVAL) == 0) checkInterrupt();
+            }
+            for (int h = 0; h < 2; h++) {
+                if (submitted[h] && done[h] == submitted[h]) {
+                    int base = h * halfSize;
+                    gpu->copyH2DRangeAsync(base, submitted[h]);
+                    gpu->recordSlotEvent(base);
+                    if (!exhausted2) {
+                        // short PCIe wait before the slots' host bufs are
+                        // reused; the other half's reads stay in flight
+                        gpu->waitSlotEvent(base);
+                        inFlightB += fillHalf(h);
+                    } else {
+                        submitted[h] = 0; // drained
+                    }
+                }
+            }
+        }
+        gpu->syncStream();
+        return;
+    }
+
     struct SlotState {
         uint64_t inFileOff = 0;
         uint64_t len = 0;

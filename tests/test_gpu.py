@@ -207,3 +207,32 @@ def test_gpu_s3_on_gpu_verify(core, tmp_path):
         assert rc == 1
     finally:
         server.shutdown()
+
+
+@pytest.mark.gpu
+def test_gpu_uring_batched_small_block_read(core, tmp_path):
+    """4K random reads at io_uring QD16 with GPU staging take the half-ring
+    batched H2D path (one ranged copy per completed half instead of one tiny
+    copy per block)."""
+    p = str(tmp_path / "gpu_uring_batch")
+    size = 8 * 1024 * 1024
+    wcfg = dict(paths=[p], path_type="file", threads=2, num_dataset_threads=2,
+                file_size=size, block_size=1 << 20, verify_salt=7)
+    eng = core.Engine(wcfg)
+    eng.prepare()
+    eng.start_phase(core.PHASES["WRITE"])
+    assert eng.wait_phase_done(120_000)
+    assert not [r["error"] for r in eng.finish_phase() if r["error"]]
+
+    rcfg = dict(paths=[p], path_type="file", threads=2, num_dataset_threads=2,
+                file_size=size, block_size=4096, iodepth=16, random=True,
+                gpu_ids=[0])
+    eng = core.Engine(rcfg)
+    eng.prepare()
+    eng.start_phase(core.PHASES["READ"])
+    assert eng.wait_phase_done(120_000)
+    res = eng.finish_phase()
+    assert not [r["error"] for r in res if r["error"]]
+    # random fair-share partitioning covers the full dataset in blocks
+    assert sum(r["iops"] for r in res) == size // 4096
+    assert sum(r["bytes"] for r in res) == size
