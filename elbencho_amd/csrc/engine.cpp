@@ -1092,9 +1092,17 @@ void Worker::fileModeBlocksUring(bool isWrite)
     // ranged H2D covers all its contiguous slots while the other half's
     // storage reads stay in flight (same geometry as the sync path's
     // half-ring batching above).
+    // measured (profiles/r01_uring_iops.md): at QD>=32 the per-block
+    // event-pipelined staging below overlaps storage and PCIe better than
+    // half-ring batching (5.1M vs 3.4M IOPS warm), so batching here is
+    // opt-in (EB_GPU_URING_BATCH=1) — unlike the sync path where it wins
     const int gpuBatch = gpu ? gpuBatchSlots(bs) : 1;
     const bool batchedRead = gpu && !isWrite && cfg.verifySalt < 0 &&
-                             gpuBatch >= 2 && depth >= 4;
+                             gpuBatch >= 2 && depth >= 4 &&
+                             getenv("EB_GPU_URING_BATCH");
+    if (getenv("EB_DEBUG_PATH"))
+        fprintf(stderr, "[eb] uring path: gpu=%d batchedRead=%d batch=%d depth=%d\n",
+                gpu ? 1 : 0, (int)batchedRead, gpuBatch, depth);
     if (batchedRead) {
         const int halfSize = std::min(depth / 2, 64);
         struct Op {
