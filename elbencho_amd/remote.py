@@ -320,6 +320,10 @@ class RemoteRunner:
             host_workers = []
             for w in res["workers"]:
                 ws = WorkerStats(**{k: w[k] for k in w if k in WorkerStats.__dataclass_fields__})
+                if ws.error:
+                    # frame remote errors with the origin host (reference
+                    # RemoteWorker::frameHostErrorMsg, RemoteWorker.cpp:650)
+                    ws.error = f"[{hs.client.hostport}] {ws.error}"
                 host_workers.append(ws)
             all_workers.extend(host_workers)
             self.last_service_elapsed.append(

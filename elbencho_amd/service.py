@@ -20,6 +20,7 @@ import sys
 import threading
 import time
 import urllib.parse
+from collections import deque
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 
 from elbencho_amd import HTTP_PROTOCOL_VERSION, VERSION
@@ -42,6 +43,9 @@ class ServiceState:
         self.phase_name = "IDLE"
         self.results: list[dict] | None = None
         self.error: str = ""
+        # recent error lines for master-side reporting (reference Logger
+        # error history, Logger.h:33+)
+        self.err_history: deque[str] = deque(maxlen=16)
         self.cpu = CpuUtil()
         self.quit_requested = threading.Event()
         self.auth = _auth_hash(self._read_pw(base_cfg.svc_pw_file)) if base_cfg.svc_pw_file \
@@ -98,6 +102,7 @@ class ServiceState:
             p["phase_name"] = self.phase_name
             p["cpu_util_pct"] = self.cpu.percent_since_last()
             p["idle"] = False
+            p["error_history"] = list(self.err_history)
             return p
 
     def bench_result(self) -> dict:
@@ -108,8 +113,12 @@ class ServiceState:
                 self.runner.wait(-1)
                 workers = self.runner.finish()
                 self.results = [w.__dict__ for w in workers]
+                for w in workers:
+                    if w.error:
+                        self.err_history.append(f"{self.phase_name}: {w.error}")
                 self.phase_name = "IDLE"
-            return {"bench_id": self.bench_id, "workers": self.results}
+            return {"bench_id": self.bench_id, "workers": self.results,
+                    "error_history": list(self.err_history)}
 
     def interrupt(self) -> None:
         with self.lock:

@@ -189,3 +189,16 @@ def test_distributed_custom_tree(services, tmp_path):
     assert res.returncode == 0, res.stdout + res.stderr
     assert not (base / "a").exists()
     assert not (base / "bigshared").exists()
+
+
+def test_service_error_history_framed(tmp_path, services):
+    """Worker errors surface in the service error history (/benchresult) and
+    the master frames them with the origin host (reference Logger error
+    history + RemoteWorker::frameHostErrorMsg)."""
+    port = services[0]
+    res = run_master(["--hosts", f"127.0.0.1:{port}", "-r", "-t", "1",
+                      "-s", "1m", "-b", "1m", "--nolive",
+                      str(tmp_path / "missing_file")])
+    out = res.stdout + res.stderr
+    assert res.returncode != 0
+    assert f"[127.0.0.1:{port}]" in out  # framed host in the error line
