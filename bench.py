@@ -81,6 +81,20 @@ def main() -> int:
     os.makedirs(args.dir, exist_ok=True)
     path = os.path.join(args.dir, f"bench_r{rank}.bin")
 
+    # EB_BENCH_BIND=1 pins each rank's workers to distinct physical cores
+    # (even CPU indices skip SMT siblings). Measured both ways on the 4K
+    # staging path: single-engine micro-benches prefer binding, but across
+    # fresh boxes the scheduler's own placement is steadier (5.2-5.7M IOPS
+    # unbound vs 3.7-6.1M with a fixed list — profiles/r01_uring_iops.md),
+    # so unbound is the default.
+    cores: list[int] = []
+    ncpu = os.cpu_count() or 0
+    if os.environ.get("EB_BENCH_BIND"):
+        first = (rank * args.threads) * 2
+        want = [first + 2 * i for i in range(args.threads)]
+        if all(c < ncpu for c in want):
+            cores = want
+
     base_cfg = dict(
         path_type="file",
         threads=args.threads,
@@ -92,6 +106,7 @@ def main() -> int:
         lat=False,
         blockvar_pct=0,  # setup fill is random already; steps measure I/O, not RNG
         bench_seed=0x9E3779B97F4A7C15 ^ rank,
+        cores=cores,
     )
 
     # setup: create the synthetic file (not timed)
