@@ -252,6 +252,9 @@ class BenchConfig:
     s3_virt_addr: bool = False     # --s3virtaddr (virtual-hosted addressing)
     s3_log: int = 0                # --s3log (client trace level, 0=off)
     s3_log_prefix: str = ""        # --s3logprefix (trace file prefix)
+    # master-precreated shared-MPU uploadIds ("bucket/key" -> id), sent on
+    # the wire so every service adds parts to the same upload (internal)
+    s3_mpu_upload_ids: dict = field(default_factory=dict)
 
     # --- misc ---
     config_file: str = ""          # -c/--configfile

@@ -478,6 +478,14 @@ class SharedUploadStore:
                 self.uploads[(bucket, key)] = ent
             return ent["id"]
 
+    def seed(self, bucket: str, key: str, upload_id: str) -> None:
+        """Pre-register a master-created uploadId (cross-service
+        --s3mpusharing): this instance adds parts but never completes or
+        aborts the upload (total unreachable, external owner)."""
+        with self.lock:
+            self.uploads[(bucket, key)] = {"id": upload_id, "parts": [],
+                                           "total": 1 << 62, "external": True}
+
     def add_part(self, bucket: str, key: str, part_num: int, etag: str) -> bool:
         """Record a completed part; True when all parts are done (the caller
         that gets True completes the upload — 'whoever finishes last')."""
@@ -493,6 +501,8 @@ class SharedUploadStore:
     def abort_unfinished(self, client: S3Client):
         with self.lock:
             for (bucket, key), ent in self.uploads.items():
+                if ent.get("external"):
+                    continue  # master-owned shared upload: never abort here
                 if len(ent["parts"]) < ent["total"]:
                     try:
                         client.abort_multipart(bucket, key, ent["id"])
@@ -771,19 +781,29 @@ class S3Worker(threading.Thread):
                         break
         elif ph == "S3MPUCOMPLETE":
             # complete multipart uploads left open by an earlier --s3nompucompl
-            # run — possibly by ANOTHER instance: uploadIds and part ETags are
-            # rediscovered from S3 itself (reference S3MPUCOMPLETE phase +
-            # shared MPU store, Common.h:196 / S3UploadStore)
-            for name in self._object_names():
-                self._check_interrupt()
-                bucket = self._bucket()
-                uploads = [u for k, u in
-                           self.client.list_multipart_uploads(bucket, prefix=name)
-                           if k == name]
-                for upload_id in uploads:
-                    parts = self.client.list_parts(bucket, name, upload_id)
-                    self.client.complete_multipart(bucket, name, upload_id, parts)
-                    self.ops.entries += 1
+            # or --s3mpusharing run — possibly by ANOTHER instance: uploadIds
+            # and part ETags are rediscovered from S3 itself (reference
+            # S3MPUCOMPLETE phase + shared MPU store, Common.h:196)
+            if cfg.files:
+                for name in self._object_names():
+                    self._check_interrupt()
+                    bucket = self._bucket()
+                    uploads = [u for k, u in
+                               self.client.list_multipart_uploads(bucket, prefix=name)
+                               if k == name]
+                    for upload_id in uploads:
+                        parts = self.client.list_parts(bucket, name, upload_id)
+                        self.client.complete_multipart(bucket, name, upload_id, parts)
+                        self.ops.entries += 1
+            elif self.local_rank == 0:
+                # no object layout given (bucket-only path, e.g. after an
+                # --s3mpusharing run): complete every open upload per bucket
+                for bucket in self.r.buckets:
+                    self._check_interrupt()
+                    for key, upload_id in self.client.list_multipart_uploads(bucket):
+                        parts = self.client.list_parts(bucket, key, upload_id)
+                        self.client.complete_multipart(bucket, key, upload_id, parts)
+                        self.ops.entries += 1
         else:
             raise S3Error(f"S3 phase not supported: {ph}")
 
@@ -1143,6 +1163,10 @@ class S3Runner:
             blk = blk + blk  # bytes concat (a bytearray cannot resize itself)
         self.rand_block = blk[:cfg.block_size]
         self.upload_store = SharedUploadStore()
+        # cross-service sharing: uploadIds pre-created by the master
+        for objpath, upload_id in (cfg.s3_mpu_upload_ids or {}).items():
+            b, _, k = objpath.partition("/")
+            self.upload_store.seed(b, k, upload_id)
 
     # --- runner interface ---
     def start(self, phase_name: str) -> None:

@@ -74,7 +74,11 @@ class ServiceState:
                 idx = int(wire_cfg.get("service_index", 0))
                 cfg.gpu_ids = [cfg.gpu_ids[idx % len(cfg.gpu_ids)]]
             self.cfg = cfg
-            self.runner = LocalRunner(cfg)
+            if cfg.bench_mode == "s3":
+                from elbencho_amd.s3 import S3Runner
+                self.runner = S3Runner(cfg)
+            else:
+                self.runner = LocalRunner(cfg)
             self.results = None
             self.error = ""
             self.phase_name = "IDLE"

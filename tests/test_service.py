@@ -202,3 +202,40 @@ def test_service_error_history_framed(tmp_path, services):
     out = res.stdout + res.stderr
     assert res.returncode != 0
     assert f"[127.0.0.1:{port}]" in out  # framed host in the error line
+
+
+def test_s3_mpu_sharing_across_services(services, tmp_path):
+    """Cross-service --s3mpusharing: the master pre-creates the multipart
+    uploads and all services add disjoint parts to the SAME uploadId; a
+    later --s3mpucompl run completes, then a read verifies the data."""
+    from tests.s3mock import ACCESS_KEY, SECRET_KEY, S3Handler, start_mock
+
+    server, port = start_mock()
+    try:
+        ep = f"http://127.0.0.1:{port}"
+        hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+        s3args = ["--s3endpoints", ep, "--s3key", ACCESS_KEY,
+                  "--s3secret", SECRET_KEY]
+        res = run_master(s3args + ["--hosts", hosts, "-d", "-w", "-t", "2",
+                                   "-s", "512k", "-b", "64k", "--s3mpusharing",
+                                   "--verify", "8", "shsvc/obj1", "shsvc/obj2"])
+        assert res.returncode == 0, res.stdout + res.stderr
+        with S3Handler.store.lock:
+            # one shared upload per object, parts from all 4 ranks, still open
+            assert len(S3Handler.store.uploads) == 2
+            assert all(len(parts) == 8 for parts in
+                       S3Handler.store.uploads.values())
+            assert S3Handler.store.buckets["shsvc"] == {}
+        # completion from a separate instance (standalone, no services)
+        res = run_master(s3args + ["--s3mpucompl", "-t", "1", "-s", "512k",
+                                   "-b", "64k", "s3://shsvc"])
+        assert res.returncode == 0, res.stdout + res.stderr
+        with S3Handler.store.lock:
+            assert not S3Handler.store.uploads
+            assert len(S3Handler.store.buckets["shsvc"]["obj1"]) == 512 * 1024
+        res = run_master(s3args + ["--hosts", hosts, "-r", "-t", "2",
+                                   "-s", "512k", "-b", "64k", "--s3mpusharing",
+                                   "--verify", "8", "shsvc/obj1", "shsvc/obj2"])
+        assert res.returncode == 0, res.stdout + res.stderr
+    finally:
+        server.shutdown()
