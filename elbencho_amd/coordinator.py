@@ -109,14 +109,12 @@ class Coordinator:
         if cfg.hosts:
             from elbencho_amd.remote import RemoteRunner
             self.runner = RemoteRunner(cfg)
-        elif cfg.bench_mode == "s3":
-            from elbencho_amd.s3 import S3Runner
-            self.runner = S3Runner(cfg)
         else:
             import os
             if "RANK" in os.environ and "WORLD_SIZE" in os.environ:
                 # launched under torch.distributed.run: one instance per GPU,
-                # RCCL barrier + all-reduce replace the HTTP master poll
+                # RCCL barrier + all-reduce replace the HTTP master poll —
+                # applies to the posix engine AND the S3 object engine
                 from elbencho_amd import parallel
                 self.dist = parallel.init_from_env()
                 if self.dist:
@@ -125,11 +123,16 @@ class Coordinator:
                         local = int(os.environ.get("LOCAL_RANK", rank))
                         cfg.gpu_ids = [cfg.gpu_ids[local % len(cfg.gpu_ids)]]
                     cfg.rank_offset = rank * cfg.threads
-                    if not cfg.no_svc_share and cfg.path_type != PATH_DIR:
+                    if cfg.bench_mode == "s3" or (
+                            not cfg.no_svc_share and cfg.path_type != PATH_DIR):
                         cfg.num_dataset_threads = cfg.threads * world
                     else:
                         cfg.num_dataset_threads = cfg.threads
-            self.runner = LocalRunner(cfg)
+            if cfg.bench_mode == "s3":
+                from elbencho_amd.s3 import S3Runner
+                self.runner = S3Runner(cfg)
+            else:
+                self.runner = LocalRunner(cfg)
 
         try:
             if cfg.dryrun:

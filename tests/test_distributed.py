@@ -93,3 +93,32 @@ def test_gloo_two_ranks(tmp_path):
     data = (tmp_path / "shared_file").read_bytes()
     assert len(data) == size
     assert core.verify_checksum(data, 0, 11) == 2**64 - 1
+
+
+def test_gloo_s3_two_ranks(tmp_path):
+    """S3 engine under torch.distributed.run: per-rank object namespaces
+    (rank_offset), barrier'd phases, aggregated results on rank 0."""
+    from tests.s3mock import ACCESS_KEY, SECRET_KEY, S3Handler, start_mock
+
+    server, port = start_mock()
+    try:
+        env = dict(os.environ, PYTHONPATH=REPO, MASTER_ADDR="127.0.0.1",
+                   EB_DIST_BACKEND="gloo")
+        res = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", "29519", "-m", "elbencho_amd",
+             "--s3endpoints", f"http://127.0.0.1:{port}",
+             "--s3key", ACCESS_KEY, "--s3secret", SECRET_KEY, "--nolive",
+             "-d", "-w", "-r", "-t", "2", "-N", "2", "-s", "64k", "-b", "64k",
+             "--verify", "4", "s3://distbkt"],
+            env=env, capture_output=True, text=True, timeout=300)
+        assert res.returncode == 0, res.stdout + res.stderr
+        with S3Handler.store.lock:
+            objs = set(S3Handler.store.buckets["distbkt"])
+        # 2 ranks x 2 threads x 2 files in distinct global-rank namespaces
+        assert objs == {f"r{r}-f{f}" for r in range(4) for f in range(2)}
+        # aggregated total on rank 0: 8 objects
+        assert "Objects total" in res.stdout
+    finally:
+        server.shutdown()
