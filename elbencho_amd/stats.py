@@ -423,13 +423,70 @@ def append_csv_result(cfg: BenchConfig, r: PhaseResults, path: str) -> None:
 # ---------------------------------------------------------------------------
 
 def phase_results_json(cfg: BenchConfig, r: PhaseResults) -> dict[str, Any]:
+    """Reference-schema per-phase JSON document (--jsonfile): key names and
+    nesting follow Statistics::printPhaseResultsAsJSON
+    (/root/reference/source/Statistics.cpp:2485-2770) — first_done/last_done
+    subtrees with entries/s, iops, bytes/s, cpu%, nested rwmix_read and
+    latency — plus additive keys (mib_per_sec, percentiles) kept for the
+    bundled summarize tool."""
     mib = 1024 * 1024
     start_dt = datetime.datetime.fromtimestamp(r.start_time).astimezone()
+
+    def done_tree(elapsed_usec, entries, iops, nbytes, cpu):
+        t: dict[str, Any] = {"elapsed_time_ms": elapsed_usec // 1000}
+        per_sec = (lambda v: v * 1_000_000 // elapsed_usec) if elapsed_usec \
+            else (lambda v: 0)
+        if r.entries:
+            t["entries/s"] = per_sec(entries)
+            t["entries"] = entries
+        if r.iops:
+            t["iops"] = per_sec(iops)
+        if r.bytes:
+            t["bytes/s"] = per_sec(nbytes)
+            t["bytes"] = nbytes
+            t["mib_per_sec"] = per_sec(nbytes) // mib
+        t["entries_per_sec"] = t.get("entries/s", 0)
+        t["cpu%"] = cpu
+        return t
+
+    first = done_tree(r.first_finish_usec, r.sw_entries, r.sw_iops, r.sw_bytes,
+                      r.cpu_first)
+    last = done_tree(r.last_finish_usec, r.entries, r.iops, r.bytes, r.cpu_last)
+
+    if r.rm_bytes or r.rm_iops:
+        first["rwmix_read"] = {
+            "iops": r.per_sec_first(r.rm_sw_iops),
+            "bytes/s": r.per_sec_first(r.rm_sw_bytes),
+            "bytes": r.rm_sw_bytes,
+        }
+        last["rwmix_read"] = {
+            "iops": r.per_sec_last(r.rm_iops),
+            "bytes/s": r.per_sec_last(r.rm_bytes),
+            "bytes": r.rm_bytes,
+        }
+
+    latency: dict[str, Any] = {}
+    if r.entry_lat.num_values:
+        latency["entries"] = _lat_json(r.entry_lat)
+    if r.io_lat.num_values:
+        latency["IO"] = _lat_json(r.io_lat)
+    if r.io_lat_rm.num_values:
+        latency.setdefault("IO", {})["rwmix_read"] = _lat_json(r.io_lat_rm)
+    if cfg.lat_histo:
+        if r.entry_lat.num_values:
+            latency["entries"]["histogram"] = r.entry_lat.nonzero_buckets()
+        if r.io_lat.num_values:
+            latency["IO"]["histogram"] = r.io_lat.nonzero_buckets()
+    if latency:
+        last["latency"] = latency
+
     doc: dict[str, Any] = {
         "phase_type": r.phase_name,
         "phase_id": r.phase_id,
         "iso_start_date": start_dt.isoformat(timespec="milliseconds"),
         "config": {
+            "version": VERSION,
+            "command": " ".join(sys.argv),
             "path_type": cfg.path_type,
             "paths": len(cfg.paths),
             "hosts": len(cfg.hosts) if cfg.hosts else 1,
@@ -439,48 +496,18 @@ def phase_results_json(cfg: BenchConfig, r: PhaseResults) -> dict[str, Any]:
             "file_size": cfg.file_size,
             "block_size": cfg.block_size,
             "direct_io": cfg.direct,
-            "random": cfg.random,
+            "random_offsets": cfg.random,
+            "random_aligned": cfg.rand_aligned,
+            "truncate_files": cfg.truncate,
+            "shared_service_paths": not cfg.no_svc_share,
             "io_depth": cfg.iodepth,
             "gpu_ids": cfg.gpu_ids,
         },
-        "first_done": {
-            "elapsed_ms": r.first_finish_usec // 1000,
-            "entries_per_sec": r.per_sec_first(r.sw_entries),
-            "iops": r.per_sec_first(r.sw_iops),
-            "mib_per_sec": r.per_sec_first(r.sw_bytes) // mib,
-            "entries": r.sw_entries,
-            "mib": r.sw_bytes // mib,
-            "bytes": r.sw_bytes,
-            "cpu_util_pct": r.cpu_first,
-        },
-        "last_done": {
-            "elapsed_ms": r.last_finish_usec // 1000,
-            "entries_per_sec": r.per_sec_last(r.entries),
-            "iops": r.per_sec_last(r.iops),
-            "mib_per_sec": r.per_sec_last(r.bytes) // mib,
-            "entries": r.entries,
-            "mib": r.bytes // mib,
-            "bytes": r.bytes,
-            "cpu_util_pct": r.cpu_last,
-        },
+        "first_done": first,
+        "last_done": last,
     }
     if cfg.label:
         doc["label"] = cfg.label
-    if r.rm_bytes or r.rm_iops:
-        doc["first_done_rwmix_read"] = {
-            "iops": r.per_sec_first(r.rm_sw_iops),
-            "mib_per_sec": r.per_sec_first(r.rm_sw_bytes) // mib,
-            "bytes": r.rm_sw_bytes,
-        }
-        doc["last_done_rwmix_read"] = {
-            "iops": r.per_sec_last(r.rm_iops),
-            "mib_per_sec": r.per_sec_last(r.rm_bytes) // mib,
-            "bytes": r.rm_bytes,
-        }
-    if r.entry_lat.num_values:
-        doc["entries_latency"] = _lat_json(r.entry_lat)
-    if r.io_lat.num_values:
-        doc["iops_latency"] = _lat_json(r.io_lat)
     if r.errors:
         doc["errors"] = r.errors
     return doc
