@@ -256,6 +256,20 @@ class Coordinator:
             live_csv = None
 
         if self.dist:
+            # cross-rank --s3mpusharing: rank 0 pre-creates the shared
+            # multipart uploads, all ranks receive the uploadIds (same
+            # mechanism as the HTTP master's wire distribution)
+            if (cfg.bench_mode == "s3" and cfg.s3_mpu_sharing and
+                    name == "WRITE" and not cfg.s3_mpu_upload_ids):
+                import torch.distributed as torch_dist
+
+                from elbencho_amd.s3 import precreate_upload_ids
+                box = [precreate_upload_ids(cfg) if self.dist.rank == 0 else None]
+                torch_dist.broadcast_object_list(box, src=0)
+                cfg.s3_mpu_upload_ids = box[0]
+                for objpath, upload_id in cfg.s3_mpu_upload_ids.items():
+                    b, _, k = objpath.partition("/")
+                    self.runner.upload_store.seed(b, k, upload_id)
             self.dist.barrier()  # lockstep phase start across GPU ranks
 
         if cfg.log_level >= 1:

@@ -174,23 +174,8 @@ class RemoteRunner:
         # --s3mpucompl run does (or --s3nompucompl semantics apply).
         if (cfg.bench_mode == "s3" and cfg.s3_mpu_sharing and
                 phase_name == "WRITE" and not cfg.s3_mpu_upload_ids):
-            from elbencho_amd.s3 import S3Client
-            part_size = cfg.s3_mpu_split or cfg.block_size
-            if cfg.file_size <= part_size:
-                raise RuntimeError("S3 MPU sharing mode selected but object size "
-                                   "is not larger than part block size")
-            client = S3Client(cfg.s3_endpoints[0], cfg.s3_key, cfg.s3_secret,
-                              cfg.s3_region, session_token=cfg.s3_session_token,
-                              virtual_addressing=cfg.s3_virt_addr)
-            ids: dict[str, str] = {}
-            for p in cfg.paths:
-                p = p[len("s3://"):] if p.startswith("s3://") else p
-                bucket, _, key = p.partition("/")
-                if not key:
-                    raise RuntimeError("--s3mpusharing requires bucket/object paths")
-                ids[f"{bucket}/{key}"] = client.create_multipart(bucket, key)
-            client.close()
-            cfg.s3_mpu_upload_ids = ids
+            from elbencho_amd.s3 import precreate_upload_ids
+            cfg.s3_mpu_upload_ids = precreate_upload_ids(cfg)
 
         # netbench role assignment: hosts named in --servers act as servers,
         # the rest are clients (reference scale-out semantics)

@@ -412,6 +412,31 @@ _ACL_PERM_HEADERS = {"READ": "x-amz-grant-read", "WRITE": "x-amz-grant-write",
                      "FULL_CONTROL": "x-amz-grant-full-control"}
 
 
+def precreate_upload_ids(cfg: BenchConfig) -> dict[str, str]:
+    """Create one multipart upload per shared object ("bucket/key" paths) for
+    cross-instance --s3mpusharing; the ids are distributed to every instance
+    so all of them add parts to the SAME upload (reference
+    ProgArgs::precreateMpuIDs, ProgArgs.cpp:2950-2990)."""
+    part_size = cfg.s3_mpu_split or cfg.block_size
+    if cfg.file_size <= part_size:
+        raise S3Error("S3 MPU sharing mode selected but object size is not "
+                      "larger than part block size")
+    client = S3Client(cfg.s3_endpoints[0], cfg.s3_key, cfg.s3_secret,
+                      cfg.s3_region, session_token=cfg.s3_session_token,
+                      virtual_addressing=cfg.s3_virt_addr)
+    ids: dict[str, str] = {}
+    try:
+        for p in cfg.paths:
+            p = p[len("s3://"):] if p.startswith("s3://") else p
+            bucket, _, key = p.partition("/")
+            if not key:
+                raise S3Error("--s3mpusharing requires bucket/object paths")
+            ids[f"{bucket}/{key}"] = client.create_multipart(bucket, key)
+    finally:
+        client.close()
+    return ids
+
+
 def build_put_headers(cfg: BenchConfig) -> dict[str, str]:
     """Upload headers from SSE (--s3sse/--s3sseckey/--s3ssekmskey) and
     inline-ACL (--s3aclputinl) options."""

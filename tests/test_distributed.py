@@ -122,3 +122,32 @@ def test_gloo_s3_two_ranks(tmp_path):
         assert "Objects total" in res.stdout
     finally:
         server.shutdown()
+
+
+def test_gloo_s3_mpu_sharing_two_ranks(tmp_path):
+    """--s3mpusharing under torch.distributed.run: rank 0 pre-creates the
+    shared uploads and broadcasts the ids; both ranks add disjoint parts."""
+    from tests.s3mock import ACCESS_KEY, SECRET_KEY, S3Handler, start_mock
+
+    server, port = start_mock()
+    try:
+        env = dict(os.environ, PYTHONPATH=REPO, MASTER_ADDR="127.0.0.1",
+                   EB_DIST_BACKEND="gloo")
+        res = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", "29523", "-m", "elbencho_amd",
+             "--s3endpoints", f"http://127.0.0.1:{port}",
+             "--s3key", ACCESS_KEY, "--s3secret", SECRET_KEY, "--nolive",
+             "-d", "-w", "-t", "2", "-s", "512k", "-b", "64k",
+             "--s3mpusharing", "--verify", "3", "distsh/obj1"],
+            env=env, capture_output=True, text=True, timeout=300)
+        assert res.returncode == 0, res.stdout + res.stderr
+        with S3Handler.store.lock:
+            # ONE shared upload with all 8 parts from both ranks, left open
+            # for a later --s3mpucompl (cross-instance semantics)
+            assert len(S3Handler.store.uploads) == 1
+            (parts,) = S3Handler.store.uploads.values()
+            assert sorted(parts) == list(range(1, 9))
+    finally:
+        server.shutdown()
