@@ -408,10 +408,30 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--hdfs", action="store_true", help=argparse.SUPPRESS)
     g.add_argument("-c", "--configfile", default="", metavar="PATH",
                    help="Read options from a config file (key=value lines).")
-    g.add_argument("--version", action="version",
-                   version=f"elbencho-amd {VERSION} (MI355X/gfx950 native)")
+    class _VersionAction(argparse.Action):
+        def __call__(self, parser, ns, values, option_string=None):
+            print(_version_text())  # verbatim (argparse "version" reflows)
+            parser.exit()
+
+    g.add_argument("--version", action=_VersionAction, nargs=0,
+                   help="Show version and included build features.")
 
     return p
+
+
+def _version_text() -> str:
+    """Version + build-feature matrix (reference
+    ProgArgs::printVersionAndBuildInfo, ProgArgs.cpp:3646-3740)."""
+    from elbencho_amd import HTTP_PROTOCOL_VERSION
+    included = ["hip/gfx950", "rccl-xgmi", "io_uring", "io_uring-fixedbufs",
+                "s3", "netbench", "corebind", "libnuma", "backtrace",
+                "althttpsvc", "mmap-zerocopy"]
+    excluded = ["cuda", "cufile/gds", "hdfs", "libaio", "s3crt", "mimalloc"]
+    return (f"elbencho-amd\n"
+            f" * Version: {VERSION} (MI355X/gfx950 native)\n"
+            f" * Net protocol version: {HTTP_PROTOCOL_VERSION}\n"
+            f" * Included optional build features: {' '.join(included)}\n"
+            f" * Excluded optional build features: {' '.join(excluded)}")
 
 
 def args_to_config(args: argparse.Namespace) -> BenchConfig:
