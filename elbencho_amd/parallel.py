@@ -30,8 +30,14 @@ def init_from_env(device: Optional[torch.device] = None) -> "PhaseSync | None":
     if "RANK" not in os.environ or "WORLD_SIZE" not in os.environ:
         return None
     if not dist.is_initialized():
-        backend = os.environ.get(
-            "EB_DIST_BACKEND", "nccl" if torch.cuda.is_available() else "gloo")
+        # RCCL refuses two ranks on one device ("Duplicate GPU detected"):
+        # when the job is oversubscribed (more ranks than GPUs), coordinate
+        # over gloo instead — the I/O staging itself still uses the GPUs
+        world = int(os.environ.get("WORLD_SIZE", "1"))
+        usable_nccl = torch.cuda.is_available() and \
+            world <= max(torch.cuda.device_count(), 1)
+        backend = os.environ.get("EB_DIST_BACKEND",
+                                 "nccl" if usable_nccl else "gloo")
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29511")
         dist.init_process_group(
