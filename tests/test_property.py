@@ -19,7 +19,7 @@ def core():
     return load_core()
 
 
-@settings(max_examples=25, deadline=None,
+@settings(max_examples=30, deadline=None,
           suppress_health_check=[HealthCheck.function_scoped_fixture])
 @given(
     threads=st.integers(1, 4),
@@ -27,15 +27,18 @@ def core():
     block_kib=st.sampled_from([4, 16, 64, 256]),
     pattern=st.sampled_from(["seq", "backward", "random", "strided"]),
     num_files=st.integers(1, 3),
+    iodepth=st.sampled_from([1, 8]),
 )
 def test_file_mode_roundtrip(core, tmp_path_factory, threads, file_size,
-                             block_kib, pattern, num_files):
+                             block_kib, pattern, num_files, iodepth):
+    if pattern == "backward" and iodepth > 1:
+        iodepth = 1  # async engine orders by completion; backward is sync-only
     tmp_path = tmp_path_factory.mktemp("prop")
     paths = [str(tmp_path / f"f{i}") for i in range(num_files)]
     bs = block_kib * 1024
     cfg = dict(paths=paths, path_type="file", threads=threads,
                num_dataset_threads=threads, file_size=file_size,
-               block_size=bs, verify_salt=13,
+               block_size=bs, verify_salt=13, iodepth=iodepth,
                backward=(pattern == "backward"),
                random=(pattern == "random"),
                strided=(pattern == "strided"))

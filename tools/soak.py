@@ -37,7 +37,8 @@ def cli(args: list[str]) -> None:
 def main() -> int:
     minutes = float(sys.argv[1]) if len(sys.argv) > 1 else 5.0
     base = "/dev/shm/soak"
-    os.makedirs(base, exist_ok=True)
+    for sub in ("", "dirs", "ct"):
+        os.makedirs(os.path.join(base, sub), exist_ok=True)
 
     from s3mock import ACCESS_KEY, SECRET_KEY, start_mock
     server, port = start_mock()
