@@ -395,6 +395,14 @@ class BenchConfig:
                                   "file/bdev mode (--nodiocheck to skip this check)")
         if self.verify >= 0 and self.random and not self.rand_aligned:
             raise ConfigError("--verify cannot be used with unaligned random offsets")
+        # reference ProgArgs.cpp:1548-1556: readback verification and inline
+        # reads are sync-engine features
+        if self.verify_direct and (self.verify < 0 or not self.run_write):
+            raise ConfigError("--verifydirect requires --verify and -w/--write")
+        if self.verify_direct and self.iodepth > 1:
+            raise ConfigError("--verifydirect cannot be used together with --iodepth")
+        if self.read_inline and self.iodepth > 1:
+            raise ConfigError("--readinline cannot be used together with --iodepth")
         if self.verify >= 0 and self.blockvar_pct and False:
             pass  # verify overrides block variance; no error
 

@@ -65,7 +65,10 @@ def main() -> int:
         cli(["-r", "-t", "8", "--iodepth", "32", "-b", "4k", "--rand",
              "--randamount", "128m", "--gpuids", "0",
              os.path.join(base, "big")])
-        # rwmix with dedicated readers
+        # rwmix with dedicated readers (file pre-written: mix reads need data)
+        if rounds == 0:
+            cli(["-w", "-t", "4", "-b", "1m", "-s", "64m",
+                 os.path.join(base, "mix")])
         cli(["-w", "-t", "4", "--rwmixthr", "2", "-b", "1m", "-s", "64m",
              os.path.join(base, "mix")])
         # custom tree round-robin with verify
