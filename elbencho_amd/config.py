@@ -177,7 +177,10 @@ class BenchConfig:
     quit_services: bool = False    # --quit
     rank_offset: int = 0           # --rankoffset
     dir_sharing: bool = False      # --dirsharing
-    no_fd_sharing: bool = False    # --nofdsharing
+    # --nofdsharing: accepted for CLI compatibility; this engine always opens
+    # per-thread fds (the reference's non-shared mode) — pread/pwrite carry
+    # the offset, so fd sharing is only an open()-count optimization there
+    no_fd_sharing: bool = False
 
     # --- custom tree ---
     treefile: str = ""             # --treefile

@@ -565,17 +565,13 @@ class S3Worker(threading.Thread):
         key, secret = cfg.s3_key, cfg.s3_secret
         if runner.credentials:  # --s3credfile/--s3credlist round-robin
             key, secret = runner.credentials[self.rank % len(runner.credentials)]
+        self._client_args = (ep, key, secret)
+        self._pool = None  # lazy --iodepth pipelining pool
         if cfg.s3_single:
             # --s3single: one shared client instance for all worker threads
             self.client = runner.shared_client
         else:
-            self.client = S3Client(ep, key, secret, cfg.s3_region,
-                                   extra_put_headers=build_put_headers(cfg),
-                                   session_token=cfg.s3_session_token,
-                                   virtual_addressing=cfg.s3_virt_addr,
-                                   checksum_algo=cfg.s3_chksum_algo,
-                                   trace=runner.trace,
-                                   sign_payload=cfg.s3_sign_policy != 2)
+            self.client = self._new_client()
         self.core = load_core()
         # --gpuids + --verify: verify/generate object data with the gfx950
         # kernels in HBM (BASELINE config 5); one persistent context per worker
@@ -585,6 +581,32 @@ class S3Worker(threading.Thread):
             self.gpu = self.core.GpuBufferOps(dev, cfg.block_size)
         elif cfg.gpu_ids and self.core.gpu_device_count() <= 0:
             raise S3Error("GPU requested (gpuids) but no HIP device is available")
+
+    def _new_client(self) -> S3Client:
+        cfg = self.r.cfg
+        ep, key, secret = self._client_args
+        return S3Client(ep, key, secret, cfg.s3_region,
+                        extra_put_headers=build_put_headers(cfg),
+                        session_token=cfg.s3_session_token,
+                        virtual_addressing=cfg.s3_virt_addr,
+                        checksum_algo=cfg.s3_chksum_algo,
+                        trace=self.r.trace,
+                        sign_payload=cfg.s3_sign_policy != 2)
+
+    def _pipeline(self):
+        """(executor, client queue, depth) for --iodepth S3 pipelining: up to
+        iodepth part uploads / ranged downloads of one object in flight per
+        worker, each pool thread with its own connection (reference async
+        multipart pipelining, LocalWorker.cpp:5155/:6280)."""
+        if self._pool is None:
+            import queue
+            from concurrent.futures import ThreadPoolExecutor
+            n = min(self.r.cfg.iodepth, 16)
+            clients: queue.Queue = queue.Queue()
+            for _ in range(n):
+                clients.put(self._new_client())
+            self._pool = (ThreadPoolExecutor(max_workers=n), clients, n)
+        return self._pool
 
     # --- object name layout mirrors dir mode: r{rank}/d{dir}/r{rank}-f{file} ---
     def _object_names(self):
@@ -637,6 +659,12 @@ class S3Worker(threading.Thread):
             self.error = str(e)
             self.r.interrupt_flag.set()
         finally:
+            if self._pool is not None:
+                ex, clients, _ = self._pool
+                ex.shutdown(wait=True)
+                while not clients.empty():
+                    clients.get_nowait().close()
+                self._pool = None
             self.elapsed_us = self.elapsed_us or int(
                 (time.monotonic() - self.r.phase_start) * 1e6)
             self.r.on_worker_done(self)
@@ -886,18 +914,23 @@ class S3Worker(threading.Thread):
                 self.ops.iops += 1
             else:  # multipart: block size = part size (reference -b semantics)
                 upload_id = self.client.create_multipart(bucket, name)
-                parts = []
                 try:
-                    for part_num, (off, ln) in enumerate(
-                            self._part_sizes(size, self.rank), 1):
-                        self._check_interrupt()
-                        t0 = time.monotonic()
-                        etag = self.client.upload_part(bucket, name, upload_id,
-                                                       part_num, self._make_block(ln, off))
-                        self.io_lat.vec = _add_lat(self.io_lat, t0)
-                        parts.append((part_num, etag))
-                        self.ops.bytes += ln
-                        self.ops.iops += 1
+                    if cfg.iodepth > 1 and not cfg.s3_single:
+                        parts = self._put_parts_pipelined(bucket, name, upload_id,
+                                                          size)
+                    else:
+                        parts = []
+                        for part_num, (off, ln) in enumerate(
+                                self._part_sizes(size, self.rank), 1):
+                            self._check_interrupt()
+                            t0 = time.monotonic()
+                            etag = self.client.upload_part(
+                                bucket, name, upload_id, part_num,
+                                self._make_block(ln, off))
+                            self.io_lat.vec = _add_lat(self.io_lat, t0)
+                            parts.append((part_num, etag))
+                            self.ops.bytes += ln
+                            self.ops.iops += 1
                     if not cfg.s3_no_mpu_compl:
                         self.client.complete_multipart(bucket, name, upload_id, parts)
                     # else: left open for a later S3MPUCOMPLETE phase
@@ -906,6 +939,48 @@ class S3Worker(threading.Thread):
                     raise
             self.entry_lat.vec = _add_lat(self.entry_lat, te)
             self.ops.entries += 1
+
+    def _put_parts_pipelined(self, bucket: str, name: str, upload_id: str,
+                             size: int) -> list[tuple[int, str]]:
+        """--iodepth > 1: keep up to iodepth part uploads of one object in
+        flight (blocks are generated in the worker thread, uploads overlap
+        on per-slot connections)."""
+        import threading as _threading
+        from concurrent.futures import as_completed
+
+        ex, clients, depth = self._pipeline()
+        inflight = _threading.Semaphore(depth * 2)  # bound queued block memory
+
+        def task(pn: int, body: bytes):
+            try:
+                c = clients.get()
+                try:
+                    t0 = time.monotonic()
+                    etag = c.upload_part(bucket, name, upload_id, pn, body)
+                    return pn, etag, len(body), int((time.monotonic() - t0) * 1e6)
+                finally:
+                    clients.put(c)
+            finally:
+                inflight.release()
+
+        futs = []
+        try:
+            for part_num, (off, ln) in enumerate(self._part_sizes(size, self.rank), 1):
+                self._check_interrupt()
+                inflight.acquire()
+                futs.append(ex.submit(task, part_num, self._make_block(ln, off)))
+            parts = []
+            for f in as_completed(futs):
+                pn, etag, ln, lat_us = f.result()
+                parts.append((pn, etag))
+                self.io_lat.vec = _add_lat_us(self.io_lat, lat_us)
+                self.ops.bytes += ln
+                self.ops.iops += 1
+            return parts
+        except BaseException:
+            for f in futs:
+                f.cancel()
+            raise
 
     def _put_objects_shared(self):
         """--s3mpusharing: every worker uploads its round-robin share of parts
@@ -992,36 +1067,92 @@ class S3Worker(threading.Thread):
             return self._get_objects_shared()
         if cfg.s3_rand_obj:
             return self._get_random_objects()
+        pipelined = cfg.iodepth > 1 and not cfg.s3_single and size > bs
         for name in self._object_names():
             self._check_interrupt()
             te = time.monotonic()
             bucket = self._bucket()
+            if pipelined:
+                self._get_ranges_pipelined(bucket, name, size, bs)
+            else:
+                off = 0
+                while off < size:
+                    ln = min(bs, size - off)
+                    t0 = time.monotonic()
+                    self._oplog("GetObject", name, off, ln, True)
+                    data = self.client.get_object(bucket, name, (off, off + ln - 1))
+                    self._oplog("GetObject", name, off, ln, False)
+                    self.io_lat.vec = _add_lat(self.io_lat, t0)
+                    if len(data) != ln:
+                        raise S3Error(f"short ranged read of {name}: "
+                                      f"{len(data)} != {ln}")
+                    if cfg.verify >= 0 and not cfg.s3_fastget:
+                        self._verify_block(name, data, off)
+                    self.ops.bytes += ln
+                    self.ops.iops += 1
+                    off += ln
+            self.entry_lat.vec = _add_lat(self.entry_lat, te)
+            self.ops.entries += 1
+
+    def _verify_block(self, name: str, data: bytes, off: int) -> None:
+        cfg = self.r.cfg
+        if self.gpu and off % 8 == 0 and len(data) % 16 == 0:
+            nbad, first = self.gpu.verify(data, off, cfg.verify)
+            if nbad:
+                raise S3Error(f"S3 data verification failed (GPU) for "
+                              f"{name} at object offset {first}")
+        else:
+            bad = self.core.verify_checksum(data, off, cfg.verify)
+            if bad != 2**64 - 1:
+                raise S3Error(f"S3 data verification failed for {name} at "
+                              f"object offset {bad}")
+
+    def _get_ranges_pipelined(self, bucket: str, name: str, size: int,
+                              bs: int) -> None:
+        """--iodepth > 1: up to iodepth ranged GETs of one object in flight;
+        verification happens in the worker thread as downloads complete
+        (reference async download pipelining, LocalWorker.cpp:6280)."""
+        import threading as _threading
+        from concurrent.futures import as_completed
+
+        cfg = self.r.cfg
+        ex, clients, depth = self._pipeline()
+        inflight = _threading.Semaphore(depth * 2)
+
+        def task(off: int, ln: int):
+            try:
+                c = clients.get()
+                try:
+                    t0 = time.monotonic()
+                    data = c.get_object(bucket, name, (off, off + ln - 1))
+                    return off, ln, data, int((time.monotonic() - t0) * 1e6)
+                finally:
+                    clients.put(c)
+            finally:
+                inflight.release()
+
+        futs = []
+        try:
             off = 0
             while off < size:
+                self._check_interrupt()
                 ln = min(bs, size - off)
-                t0 = time.monotonic()
-                self._oplog("GetObject", name, off, ln, True)
-                data = self.client.get_object(bucket, name, (off, off + ln - 1))
-                self._oplog("GetObject", name, off, ln, False)
-                self.io_lat.vec = _add_lat(self.io_lat, t0)
+                inflight.acquire()
+                futs.append(ex.submit(task, off, ln))
+                off += ln
+            for f in as_completed(futs):
+                off, ln, data, lat_us = f.result()
                 if len(data) != ln:
                     raise S3Error(f"short ranged read of {name}: {len(data)} != {ln}")
                 if cfg.verify >= 0 and not cfg.s3_fastget:
-                    if self.gpu and off % 8 == 0 and ln % 16 == 0:
-                        nbad, first = self.gpu.verify(data, off, cfg.verify)
-                        if nbad:
-                            raise S3Error(f"S3 data verification failed (GPU) for "
-                                          f"{name} at object offset {first}")
-                    else:
-                        bad = self.core.verify_checksum(data, off, cfg.verify)
-                        if bad != 2**64 - 1:
-                            raise S3Error(f"S3 data verification failed for {name} at "
-                                          f"object offset {bad}")
+                    self._verify_block(name, data, off)
+                self.io_lat.vec = _add_lat_us(self.io_lat, lat_us)
                 self.ops.bytes += ln
                 self.ops.iops += 1
-                off += ln
-            self.entry_lat.vec = _add_lat(self.entry_lat, te)
-            self.ops.entries += 1
+        except BaseException:
+            for f in futs:
+                f.cancel()
+            raise
 
     def _get_random_objects(self):
         """--s3randobj: ranged reads at random offsets of random objects
@@ -1088,7 +1219,10 @@ class S3Worker(threading.Thread):
 
 
 def _add_lat(h: Histogram, t0: float) -> list[int]:
-    us = int((time.monotonic() - t0) * 1e6)
+    return _add_lat_us(h, int((time.monotonic() - t0) * 1e6))
+
+
+def _add_lat_us(h: Histogram, us: int) -> list[int]:
     idx = _bucket_index(us)
     v = h.vec
     v[0] += 1

@@ -367,3 +367,19 @@ def test_s3_fastput_unsigned_payload(mock_s3, capsys):
     c.create_bucket("fp2")
     c.put_object("fp2", "o", b"x" * 100)
     assert c.get_object("fp2", "o") == b"x" * 100
+
+
+def test_s3_iodepth_pipelined(mock_s3, capsys):
+    """--iodepth pipelines part uploads and ranged downloads per worker
+    (reference async multipart pipelining)."""
+    rc = _cli(mock_s3, ["-d", "-w", "-r", "-t", "2", "-N", "2", "-s", "256k",
+                        "-b", "32k", "--iodepth", "4", "--verify", "9",
+                        "--lat", "s3://pipebkt"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    for line in out.splitlines():
+        if "Objects total" in line:
+            assert line.split()[-1] == "4"
+            break
+    else:
+        raise AssertionError("no Objects total row")
