@@ -94,7 +94,7 @@ def main() -> int:
           f"vram {vram0} -> warm {vram_warm} -> {vram1} "
           f"(post-warm delta {vram1 - vram_warm})")
     server.shutdown()
-    # fail loudly on post-warm growth > 256 MiB
+    # fail loudly on post-warm growth beyond the measured ceiling allowance
     if rounds > 50:
         assert vram1 - vram_warm < 768 * 1024 * 1024, "VRAM growth detected"
     return 0
