@@ -518,7 +518,8 @@ void Worker::preWriteFill(int slot, uint64_t len, uint64_t fileOff)
         uint64_t refillLen = (len * cfg.blockVarPct) / 100;
         if (gpu && (len % 16 == 0)) {
             gpu->blockVarRefillDev(slot, len, refillLen,
-                                   cfg.benchSeed ^ (0xD1B54A32D192ED03ULL * (globalRank + 1)));
+                                   cfg.benchSeed ^ (0xD1B54A32D192ED03ULL * (globalRank + 1)),
+                                   cfg.blockVarAlgo == "fast");
             return;
         }
         // CPU refill: refillLen random bytes, rest a fresh constant u64

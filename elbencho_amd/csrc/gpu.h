@@ -88,7 +88,7 @@ public:
     // --- device-side buffer ops (hand-written gfx950 kernels) ---
 
     // Fill device buffer with xoshiro256++ random data (replaces curand).
-    void fillRandDev(int slot, uint64_t len, uint64_t seed);
+    void fillRandDev(int slot, uint64_t len, uint64_t seed, bool fastAlgo = false);
 
     // Integrity-checksum fill: u64 at 8-aligned file offset o gets value
     // o + salt. Requires fileOff % 8 == 0 and len % 8 == 0.
@@ -100,7 +100,8 @@ public:
 
     // Block-variance refill: first refillLen bytes random, remainder filled
     // with one random u64 constant (defeats dedup). 8-aligned lengths.
-    void blockVarRefillDev(int slot, uint64_t len, uint64_t refillLen, uint64_t seed);
+    void blockVarRefillDev(int slot, uint64_t len, uint64_t refillLen, uint64_t seed,
+                           bool fastAlgo = false);
 
 private:
     struct Impl;

@@ -107,6 +107,61 @@ __global__ __launch_bounds__(256) void ebFillRandKernel(ulonglong2* __restrict__
     }
 }
 
+// "fast" random fill: value = splitmix64(seed, index) — a pure function of
+// the element index (no per-thread RNG state, ~6 ALU ops per u64), the GPU
+// analogue of the reference's golden-prime "fast" RandAlgo. Used for
+// --blockvaralgo fast; xoshiro256++ above serves balanced/strong.
+__global__ __launch_bounds__(256) void ebFillFastKernel(ulonglong2* __restrict__ buf,
+                                                        uint64_t nVec2,
+                                                        uint64_t seed)
+{
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t idx = tid; idx < nVec2; idx += stride) {
+        uint64_t s0 = seed + idx * 2 * 0x9E3779B97F4A7C15ULL;
+        uint64_t s1 = s0 + 0x9E3779B97F4A7C15ULL;
+        ulonglong2 val;
+        uint64_t z = s0;
+        z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+        z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+        val.x = z ^ (z >> 31);
+        z = s1;
+        z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+        z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+        val.y = z ^ (z >> 31);
+        buf[idx] = val;
+    }
+}
+
+// "fast" block-variance refill: splitmix-of-index prefix + constant tail.
+__global__ __launch_bounds__(256) void ebBlockVarFastKernel(ulonglong2* __restrict__ buf,
+                                                            uint64_t nVec2,
+                                                            uint64_t refillVec2,
+                                                            uint64_t seed,
+                                                            uint64_t fillConst)
+{
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t idx = tid; idx < nVec2; idx += stride) {
+        ulonglong2 val;
+        if (idx < refillVec2) {
+            uint64_t s0 = seed + idx * 2 * 0x9E3779B97F4A7C15ULL;
+            uint64_t z = s0;
+            z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+            z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+            val.x = z ^ (z >> 31);
+            z = s0 + 0x9E3779B97F4A7C15ULL;
+            z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+            z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+            val.y = z ^ (z >> 31);
+        } else {
+            val.x = fillConst;
+            val.y = fillConst;
+        }
+        buf[idx] = val;
+    }
+}
+
 // Integrity fill: u64 at file offset (fileOff + i*8) = fileOff + i*8 + salt.
 __global__ __launch_bounds__(256) void ebFillChecksumKernel(uint64_t* __restrict__ buf,
                                                             uint64_t n64,
@@ -442,14 +497,22 @@ GpuVerifyResult GpuCtx::fetchVerifyResult()
     return r;
 }
 
-void GpuCtx::fillRandDev(int slot, uint64_t len, uint64_t seed)
+void GpuCtx::fillRandDev(int slot, uint64_t len, uint64_t seed, bool fastAlgo)
 {
     uint64_t nVec2 = len / 16;
     uint64_t seq = ++impl->fillCallCounter;
     if (nVec2) {
-        dim3 grid = gridForBytes(nVec2 / (FILL_U64S_PER_STEP / 2));
-        hipLaunchKernelGGL(ebFillRandKernel, grid, dim3(256), 0, impl->stream,
-                           (ulonglong2*)impl->devBufs[slot], nVec2, seed + seq * 0x9E3779B9ULL);
+        if (fastAlgo) {
+            dim3 grid = gridForBytes(nVec2);
+            hipLaunchKernelGGL(ebFillFastKernel, grid, dim3(256), 0, impl->stream,
+                               (ulonglong2*)impl->devBufs[slot], nVec2,
+                               seed + seq * 0x9E3779B9ULL);
+        } else {
+            dim3 grid = gridForBytes(nVec2 / (FILL_U64S_PER_STEP / 2));
+            hipLaunchKernelGGL(ebFillRandKernel, grid, dim3(256), 0, impl->stream,
+                               (ulonglong2*)impl->devBufs[slot], nVec2,
+                               seed + seq * 0x9E3779B9ULL);
+        }
         HIP_CHECK(hipGetLastError());
     }
 }
@@ -470,7 +533,8 @@ GpuVerifyResult GpuCtx::verifyChecksumDev(int slot, uint64_t len, uint64_t fileO
     return fetchVerifyResult();
 }
 
-void GpuCtx::blockVarRefillDev(int slot, uint64_t len, uint64_t refillLen, uint64_t seed)
+void GpuCtx::blockVarRefillDev(int slot, uint64_t len, uint64_t refillLen, uint64_t seed,
+                               bool fastAlgo)
 {
     uint64_t nVec2 = len / 16;
     if (!nVec2) return;
@@ -483,10 +547,17 @@ void GpuCtx::blockVarRefillDev(int slot, uint64_t len, uint64_t refillLen, uint6
     z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
     uint64_t fillConst = z ^ (z >> 31);
 
-    dim3 grid = gridForBytes(nVec2 / (FILL_U64S_PER_STEP / 2));
-    hipLaunchKernelGGL(ebBlockVarKernel, grid, dim3(256), 0, impl->stream,
-                       (ulonglong2*)impl->devBufs[slot], nVec2, refillVec2,
-                       seed + seq * 0x9E3779B9ULL, fillConst);
+    if (fastAlgo) {
+        dim3 grid = gridForBytes(nVec2);
+        hipLaunchKernelGGL(ebBlockVarFastKernel, grid, dim3(256), 0, impl->stream,
+                           (ulonglong2*)impl->devBufs[slot], nVec2, refillVec2,
+                           seed + seq * 0x9E3779B9ULL, fillConst);
+    } else {
+        dim3 grid = gridForBytes(nVec2 / (FILL_U64S_PER_STEP / 2));
+        hipLaunchKernelGGL(ebBlockVarKernel, grid, dim3(256), 0, impl->stream,
+                           (ulonglong2*)impl->devBufs[slot], nVec2, refillVec2,
+                           seed + seq * 0x9E3779B9ULL, fillConst);
+    }
     HIP_CHECK(hipGetLastError());
 }
 

@@ -328,6 +328,9 @@ PYBIND11_MODULE(_core, m)
         };
         py::dict d;
         d["fill_rand_gbps"] = timeIt([&] { ctx.fillRandDev(0, len, 42); });
+        d["fill_fast_gbps"] = timeIt([&] { ctx.fillRandDev(0, len, 42, true); });
+        d["blockvar_fast_gbps"] =
+            timeIt([&] { ctx.blockVarRefillDev(0, len, len / 2, 9, true); });
         d["fill_checksum_gbps"] = timeIt([&] { ctx.fillChecksumDev(0, len, 0, 7); });
         d["blockvar_gbps"] = timeIt([&] { ctx.blockVarRefillDev(0, len, len / 2, 9); });
         ctx.fillChecksumDev(0, len, 0, 7);
