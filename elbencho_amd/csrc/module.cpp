@@ -293,24 +293,26 @@ PYBIND11_MODULE(_core, m)
         return py::make_tuple(r.numMismatches, r.firstBadFileOffset);
     }, py::arg("data"), py::arg("file_off"), py::arg("salt"), py::arg("dev") = 0);
 
-    m.def("gpu_fill_rand", [](uint64_t len, uint64_t seed, int dev) {
+    m.def("gpu_fill_rand", [](uint64_t len, uint64_t seed, int dev, bool fast) {
         GpuCtx ctx(dev, 1, len, true);
-        ctx.fillRandDev(0, len, seed);
+        ctx.fillRandDev(0, len, seed, fast);
         ctx.copyD2HAsync(0, len);
         ctx.syncStream();
         return py::bytes(ctx.hostBuf(0), len);
-    }, py::arg("len"), py::arg("seed"), py::arg("dev") = 0);
+    }, py::arg("len"), py::arg("seed"), py::arg("dev") = 0, py::arg("fast") = false);
 
-    m.def("gpu_blockvar_refill", [](uint64_t len, uint64_t refillLen, uint64_t seed, int dev) {
+    m.def("gpu_blockvar_refill", [](uint64_t len, uint64_t refillLen, uint64_t seed,
+                                    int dev, bool fast) {
         GpuCtx ctx(dev, 1, len, true);
         std::memset(ctx.hostBuf(0), 0, len);
         ctx.copyH2DAsync(0, len);
         ctx.syncStream();
-        ctx.blockVarRefillDev(0, len, refillLen, seed);
+        ctx.blockVarRefillDev(0, len, refillLen, seed, fast);
         ctx.copyD2HAsync(0, len);
         ctx.syncStream();
         return py::bytes(ctx.hostBuf(0), len);
-    }, py::arg("len"), py::arg("refill_len"), py::arg("seed"), py::arg("dev") = 0);
+    }, py::arg("len"), py::arg("refill_len"), py::arg("seed"), py::arg("dev") = 0,
+       py::arg("fast") = false);
 
     // Kernel micro-benchmark: bandwidth of the gfx950 fill/verify kernels and
     // the staging copies on one device buffer (GB/s, stream-synchronized).

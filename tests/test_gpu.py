@@ -236,3 +236,22 @@ def test_gpu_uring_batched_small_block_read(core, tmp_path):
     # random fair-share partitioning covers the full dataset in blocks
     assert sum(r["iops"] for r in res) == size // 4096
     assert sum(r["bytes"] for r in res) == size
+
+
+def test_gpu_fill_fast_quality(core):
+    """The splitmix-of-index "fast" fill (--blockvaralgo fast on GPU) is
+    distinct per seed and roughly uniform, like the xoshiro path."""
+    import collections
+
+    data = core.gpu_fill_rand(1 << 20, 777, 0, True)
+    assert len(data) == 1 << 20
+    assert data != core.gpu_fill_rand(1 << 20, 778, 0, True)
+    assert data != core.gpu_fill_rand(1 << 20, 777)  # differs from xoshiro
+    counts = collections.Counter(data)
+    assert len(counts) == 256
+    mean = (1 << 20) / 256
+    assert all(0.8 * mean < c < 1.2 * mean for c in counts.values())
+    # blockvar fast refill: prefix random, tail constant
+    out = core.gpu_blockvar_refill(1 << 16, 1 << 15, 5, 0, True)
+    tail = out[1 << 15:]
+    assert tail == tail[:8] * (len(tail) // 8)
