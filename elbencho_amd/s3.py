@@ -601,7 +601,9 @@ class S3Worker(threading.Thread):
         if self._pool is None:
             import queue
             from concurrent.futures import ThreadPoolExecutor
-            n = min(self.r.cfg.iodepth, 16)
+            cfg = self.r.cfg
+            # --s3maxconns caps the per-worker connection count
+            n = min(cfg.iodepth, cfg.s3_max_conns or 16, 16)
             clients: queue.Queue = queue.Queue()
             for _ in range(n):
                 clients.put(self._new_client())
