@@ -739,3 +739,8 @@ def main(argv: list[str] | None = None) -> int:
 
 if __name__ == "__main__":
     sys.exit(main())
+
+
+def entrypoint() -> None:
+    """console_scripts entry (pyproject.toml)."""
+    sys.exit(main())
