@@ -224,6 +224,11 @@ class RemoteRunner:
                 # rewrite servers list to hostport form for the engine
                 srv_list = sorted(server_hosts)
                 wire["servers"] = srv_list
+            elif cfg.no_svc_share:
+                # private whole dataset per host: every host runs ranks
+                # 0..threads-1 over the full work (a global offset would
+                # push later hosts past the dataset into empty slices)
+                wire["rank_offset"] = 0
             else:
                 wire["rank_offset"] = i * cfg.threads
             hs.done = False

@@ -239,3 +239,71 @@ def test_s3_mpu_sharing_across_services(services, tmp_path):
         assert res.returncode == 0, res.stdout + res.stderr
     finally:
         server.shutdown()
+
+
+def test_svcpwfile_auth(tmp_path):
+    """--svcpwfile: master requests must carry the shared-secret hash; a
+    master without the password is rejected (reference HashTk + svcpwfile)."""
+    pw = tmp_path / "secret.txt"
+    pw.write_text("hunter2\n")
+    port = free_port()
+    env = dict(os.environ, PYTHONPATH=REPO)
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "elbencho_amd", "--service", "--foreground",
+         "--port", str(port), "--svcpwfile", str(pw)],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    try:
+        deadline = time.monotonic() + 40
+        while True:
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{port}/protocolversion", timeout=1) as r:
+                    r.read()
+                break
+            except urllib.error.HTTPError:
+                break  # 403 = up, but auth required (expected)
+            except OSError:
+                if time.monotonic() > deadline:
+                    raise
+                time.sleep(0.1)
+        bench = tmp_path / "f1"
+        res = run_master(["--hosts", f"127.0.0.1:{port}", "-w", "-t", "1",
+                          "-s", "64k", "-b", "64k", str(bench)])
+        assert res.returncode != 0  # no password -> rejected
+        res = run_master(["--hosts", f"127.0.0.1:{port}", "--svcpwfile", str(pw),
+                          "-w", "-t", "1", "-s", "64k", "-b", "64k", str(bench)])
+        assert res.returncode == 0, res.stdout + res.stderr
+    finally:
+        proc.terminate()
+        proc.wait(5)
+
+
+def test_numhosts_limits_services(services, tmp_path):
+    """--numhosts 1: only the first of the two services runs the phase."""
+    hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+    res = run_master(["--hosts", hosts, "--numhosts", "1", "-d", "-w", "-t", "2",
+                      "-n", "1", "-N", "2", "-s", "16k", "-F", "-D",
+                      str(tmp_path)])
+    assert res.returncode == 0, res.stdout + res.stderr
+    for line in res.stdout.splitlines():
+        if "Files total" in line:
+            # 1 host x 2 threads x 1 dir x 2 files (2 hosts would give 8)
+            assert line.split()[-1] == "4"
+            break
+    else:
+        raise AssertionError("no Files total row:\n" + res.stdout)
+
+
+def test_nosvcshare_whole_dataset_per_host(services, tmp_path):
+    """--nosvcshare: every service host writes the whole dataset instead of
+    a partition (total bytes = hosts x size)."""
+    hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+    res = run_master(["--hosts", hosts, "--nosvcshare", "-w", "-t", "2",
+                      "-s", "2m", "-b", "1m", str(tmp_path / "shared")])
+    assert res.returncode == 0, res.stdout + res.stderr
+    for line in res.stdout.splitlines():
+        if "Total MiB" in line:
+            assert line.split()[-1] == "4"  # 2 hosts x 2 MiB
+            break
+    else:
+        raise AssertionError("no Total MiB row:\n" + res.stdout)
