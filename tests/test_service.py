@@ -307,3 +307,22 @@ def test_nosvcshare_whole_dataset_per_host(services, tmp_path):
             break
     else:
         raise AssertionError("no Total MiB row:\n" + res.stdout)
+
+
+def test_gpuperservice_override_unit():
+    """--gpuperservice: each service keeps one GPU (set) from the list,
+    selected by its service index (reference ProgArgs.h:378)."""
+    from elbencho_amd.config import BenchConfig
+    from elbencho_amd.service import ServiceState
+
+    st = ServiceState(BenchConfig())
+    cfg = BenchConfig()
+    cfg.threads = 1
+    cfg.gpu_ids = [0, 1, 2, 3]
+    cfg.gpu_per_service = True
+    cfg.paths = []
+    wire = cfg.to_wire()
+    wire["service_index"] = 2
+    # prepare builds the runner; no paths and no phases -> engine with no work
+    st.prepare_phase(wire)
+    assert st.cfg.gpu_ids == [2]
