@@ -113,8 +113,10 @@ def build_parser() -> argparse.ArgumentParser:
                    help="Percentage of each block refilled with random data between writes. "
                         "(Default: 100)")
     g.add_argument("--blockvaralgo", default="fast", metavar="ALGO",
-                   help="RNG for --blockvarpct: fast, balanced, strong. On GPUs the "
-                        "gfx950 xoshiro256++ fill kernel is used. (Default: fast)")
+                   help="RNG for --blockvarpct: fast, balanced, strong. On GPUs, "
+                        "\"fast\" runs the splitmix-of-index gfx950 kernel "
+                        "(6.4 TB/s) and the others the xoshiro256++ kernel. "
+                        "(Default: fast)")
 
     g = p.add_argument_group("GPU (MI355X)")
     g.add_argument("--gpuids", default="", metavar="IDS",
@@ -236,7 +238,9 @@ def build_parser() -> argparse.ArgumentParser:
                    help="Shared-secret file to authorize master<->service communication.")
     g.add_argument("--svcwait", action="store_true",
                    help="Wait indefinitely for services to become reachable.")
-    g.add_argument("--svcping", action="store_true", help=argparse.SUPPRESS)
+    g.add_argument("--svcping", action="store_true",
+                   help="Show per-service /status round-trip latency in the "
+                        "fullscreen live dashboard.")
     g.add_argument("--interrupt", action="store_true",
                    help="Interrupt the current phase on the given service hosts.")
     g.add_argument("--quit", action="store_true",
@@ -540,6 +544,7 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     cfg.svc_elapsed = args.svcelapsed
     cfg.svc_pw_file = args.svcpwfile
     cfg.svc_wait = args.svcwait
+    cfg.svc_ping = args.svcping
     cfg.interrupt_services = args.interrupt
     cfg.quit_services = args.quit
     cfg.rank_offset = args.rankoffset

@@ -173,6 +173,7 @@ class BenchConfig:
     svc_elapsed: bool = False      # --svcelapsed
     svc_pw_file: str = ""          # --svcpwfile
     svc_wait: bool = False         # --svcwait
+    svc_ping: bool = False         # --svcping (service RTT in the dashboard)
     interrupt_services: bool = False  # --interrupt
     quit_services: bool = False    # --quit
     rank_offset: int = 0           # --rankoffset

@@ -60,15 +60,20 @@ class FullscreenLiveStats:
         self.last_total = (poll["entries"], poll["bytes"], poll["iops"])
 
         if worker_rows:
-            lines.append(f"{'RANK':>5} {'ENTRIES':>10} {'MiB':>10} {'MiB/s':>10} "
-                         f"{'IOPS':>10}")
+            with_ping = any("ping_us" in row for row in worker_rows)
+            hdr = (f"{'RANK':>5} {'ENTRIES':>10} {'MiB':>10} {'MiB/s':>10} "
+                   f"{'IOPS':>10}")
+            lines.append(hdr + (f" {'PING us':>9}" if with_ping else ""))
             for row in worker_rows[:40]:  # cap at terminal-ish height
                 rank = row["rank"]
                 prev = self.last_rows.get(rank, (0, 0, 0))
                 wbps = int((row["bytes"] - prev[1]) / dt)
                 wiops = int((row["iops"] - prev[2]) / dt)
-                lines.append(f"{rank:>5} {row['entries']:>10} "
-                             f"{row['bytes'] // mib:>10} {wbps // mib:>10} {wiops:>10}")
+                line = (f"{rank:>5} {row['entries']:>10} "
+                        f"{row['bytes'] // mib:>10} {wbps // mib:>10} {wiops:>10}")
+                if with_ping:
+                    line += f" {row.get('ping_us', 0):>9}"
+                lines.append(line)
                 self.last_rows[rank] = (row["entries"], row["bytes"], row["iops"])
 
         # redraw in place

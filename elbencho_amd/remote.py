@@ -91,6 +91,7 @@ class HostState:
     last_status: dict = field(default_factory=dict)
     done: bool = False
     error: str = ""
+    ping_us: int = 0  # --svcping: last /status round-trip
 
 
 class RemoteRunner:
@@ -250,7 +251,11 @@ class RemoteRunner:
         def poll(i: int, hs: HostState):
             if hs.done:
                 return
+            t0 = time.monotonic()
             st = hs.client.get("/status")
+            # --svcping: /status round-trip per service for the dashboard
+            # (reference RemoteWorker pingMicroSecs, RemoteWorker.cpp:479)
+            hs.ping_us = int((time.monotonic() - t0) * 1e6)
             if st.get("bench_id") and st["bench_id"] != self._bench_id:
                 raise RuntimeError(
                     f"Service {hs.client.hostport} reports foreign benchmark ID "
@@ -305,8 +310,11 @@ class RemoteRunner:
             st = hs.last_status
             if not st or st.get("idle"):
                 continue
-            rows.append({"rank": i, "entries": st.get("entries", 0),
-                         "bytes": st.get("bytes", 0), "iops": st.get("iops", 0)})
+            row = {"rank": i, "entries": st.get("entries", 0),
+                   "bytes": st.get("bytes", 0), "iops": st.get("iops", 0)}
+            if self.cfg.svc_ping:
+                row["ping_us"] = hs.ping_us
+            rows.append(row)
         return rows
 
     # ------------------------------------------------------------------
