@@ -65,3 +65,19 @@ def test_livecsvex(tmp_path):
     content = path.read_text().splitlines()
     assert content[0].startswith("ISO date")
     assert len(content) == 1 + 4  # header + 2 ticks x 2 workers
+
+
+def test_fullscreen_svcping_column():
+    """--svcping adds a PING us column to per-service rows."""
+    cfg = BenchConfig()
+    cfg.svc_ping = True
+    out = FakeTty()
+    fs = FullscreenLiveStats(cfg, "WRITE", planned_bytes=4 << 20,
+                             planned_entries=0, out=out)
+    rows = [{"rank": 0, "entries": 5, "bytes": 1 << 19, "iops": 8,
+             "ping_us": 321}]
+    fs.update(_poll(), rows)
+    fs.finish()
+    s = out.getvalue()
+    assert "PING us" in s
+    assert "321" in s
