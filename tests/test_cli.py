@@ -159,3 +159,24 @@ def test_numservers_limits_servers():
         ["--netbench", "-w", "-s", "1m", "-b", "64k",
          "--hosts", "h1,h2,h3", "--servers", "h1,h2", "--numservers", "1"]))
     assert cfg.servers == ["h1:1611"] or cfg.servers == ["h1"]
+
+
+def test_start_time_sync(tmp_path):
+    """--start waits for the synchronized epoch start (reference --start)."""
+    import time as _time
+    f = tmp_path / "st"
+    t0 = _time.time()
+    rc = main(["-w", "-t", "1", "-s", "64k", "-b", "64k", "--nolive",
+               "--start", str(int(t0) + 2), str(f)])
+    assert rc == 0
+    assert _time.time() - t0 >= 1.0  # waited for the start time
+
+
+def test_phasedelay(tmp_path):
+    import time as _time
+    f = tmp_path / "pd"
+    t0 = _time.time()
+    rc = main(["-w", "-r", "-t", "1", "-s", "64k", "-b", "64k", "--nolive",
+               "--phasedelay", "1", str(f)])
+    assert rc == 0
+    assert _time.time() - t0 >= 1.0  # delay between WRITE and READ
