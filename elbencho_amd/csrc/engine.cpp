@@ -125,7 +125,9 @@ static uint64_t gpuBatchBytes()
 {
     static const uint64_t v = [] {
         const char* e = getenv("EB_GPU_BATCH_BYTES");
-        return e ? (uint64_t)atoll(e) : (256ULL << 10);
+        // swept on MI355X (gpurun_out/batch_sweep2.log): 4K random read hits
+        // 9.4M IOPS with 1 MiB half-rings (256 slots) vs 5.3M at 256 KiB
+        return e ? (uint64_t)atoll(e) : (1ULL << 20);
     }();
     return v;
 }
@@ -142,7 +144,7 @@ static int gpuBatchSlots(uint64_t blockSize)
     // cap the ring (2 batches) at 64 MiB per worker
     uint64_t maxBatch = (32ULL << 20) / std::max<uint64_t>(blockSize, 1);
     batch = std::min(batch, std::max<uint64_t>(maxBatch, 1));
-    return (int)std::min<uint64_t>(std::max<uint64_t>(batch, 1), 64);
+    return (int)std::min<uint64_t>(std::max<uint64_t>(batch, 1), 512);
 }
 
 static int gpuSlotCount(int ioDepth, uint64_t blockSize)
