@@ -49,11 +49,11 @@ def parse_args():
 def main() -> int:
     args = parse_args()
 
-    # zero-copy D2H writes peak with a 4-stream staging pool (45-48 GiB/s vs
+    # zero-copy D2H writes peak with a 2-3 stream staging pool (47+ GiB/s vs
     # 40-44 at the default 8; gpurun_out/stream_sweep3.log) — reads are flat
     # 4-12. Must be set before the first GpuCtx is created.
     if args.workload == "seqwrite":
-        os.environ.setdefault("EB_GPU_SHARED_STREAMS", "4")
+        os.environ.setdefault("EB_GPU_SHARED_STREAMS", "3")
 
     import torch
 
