@@ -15,7 +15,7 @@ import os
 import re
 import stat as statmod
 from dataclasses import dataclass, field
-from typing import Any, Optional
+from typing import Any
 
 from elbencho_amd.units import parse_size
 

@@ -12,13 +12,11 @@ from __future__ import annotations
 import csv
 import datetime
 import os
-import shutil
 import sys
 import time
-from typing import Any, Callable, Optional
+from typing import Any, Optional
 
 from elbencho_amd.config import BenchConfig
-from elbencho_amd.units import elapsed_ms_to_human
 
 
 class FullscreenLiveStats:

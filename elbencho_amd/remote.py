@@ -16,7 +16,7 @@ import threading
 import time
 import urllib.parse
 from dataclasses import dataclass, field
-from typing import Any, Optional
+from typing import Any
 
 from elbencho_amd import HTTP_PROTOCOL_VERSION
 from elbencho_amd.config import BenchConfig

@@ -18,7 +18,7 @@ import os
 import sys
 import time
 from dataclasses import dataclass, field
-from typing import Any, Optional
+from typing import Any
 
 from elbencho_amd import VERSION
 from elbencho_amd.config import BenchConfig, PATH_DIR
