@@ -326,3 +326,19 @@ def test_gpuperservice_override_unit():
     # prepare builds the runner; no paths and no phases -> engine with no work
     st.prepare_phase(wire)
     assert st.cfg.gpu_ids == [2]
+
+
+def test_service_reuse_after_interrupt(services, tmp_path):
+    """A service stays usable after a phase is interrupted mid-flight
+    (reference /interruptphase semantics: reset, no quit)."""
+    hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+    env = dict(os.environ, PYTHONPATH=REPO)
+    # long-running write (64 GiB would take minutes) with a 1s timelimit ->
+    # master interrupts the phase and reports failure
+    res = run_master(["--hosts", hosts, "-w", "-t", "1", "-s", "8g", "-b", "1m",
+                      "--timelimit", "1", str(tmp_path / "big")])
+    assert res.returncode != 0
+    # the same services then run a normal benchmark cleanly
+    res = run_master(["--hosts", hosts, "-w", "-r", "-t", "1", "-s", "1m",
+                      "-b", "1m", "--verify", "2", str(tmp_path / "small")])
+    assert res.returncode == 0, res.stdout + res.stderr
