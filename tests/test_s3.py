@@ -383,3 +383,14 @@ def test_s3_iodepth_pipelined(mock_s3, capsys):
             break
     else:
         raise AssertionError("no Objects total row")
+
+
+def test_s3_worker_error_does_not_hang_pipeline(mock_s3):
+    """A failing pipelined download (missing object) aborts cleanly without
+    hanging the pool or the phase."""
+    c = S3Client(mock_s3, ACCESS_KEY, SECRET_KEY)
+    c.create_bucket("errbkt")
+    c.put_object("errbkt", "r0-f0", b"x" * (64 * 1024))  # only one of two
+    rc = _cli(mock_s3, ["-r", "-t", "1", "-N", "2", "-s", "64k", "-b", "16k",
+                        "--iodepth", "4", "s3://errbkt"])
+    assert rc != 0  # r0-f1 is missing -> phase fails, process exits promptly
