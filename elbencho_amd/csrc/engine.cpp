@@ -1874,6 +1874,7 @@ void Worker::netbenchClient()
         cfg.netDevs.empty() ? "" : cfg.netDevs[globalRank % cfg.netDevs.size()];
     int fd = netConnect(host, port, bindDev, 30, eng.interruptFlag);
     setSockBufs(fd, cfg.sendBufSize, cfg.recvBufSize);
+    setRecvTimeout(fd, 5); // a dead server can't hang the client worker
 
     std::vector<char> resp(cfg.respSize);
     uint64_t sent = 0;
@@ -1886,7 +1887,8 @@ void Worker::netbenchClient()
             auto t0 = lat ? Clock::now() : Clock::time_point();
             if (!sendExact(fd, hostBufs[0], bs))
                 throw WorkerError("netbench: block send failed");
-            if (!recvExact(fd, resp.data(), cfg.respSize))
+            if (!recvExactInterruptible(fd, resp.data(), cfg.respSize,
+                                        eng.interruptFlag))
                 throw WorkerError("netbench: response recv failed");
             if (lat)
                 addIoLat((uint64_t)std::chrono::duration_cast<std::chrono::microseconds>(

@@ -121,4 +121,32 @@ inline bool recvExact(int fd, char* buf, uint64_t len)
     return true;
 }
 
+// recvExact with a socket receive timeout (set SO_RCVTIMEO first): a dead
+// peer cannot hang the worker past the timeout window — EAGAIN checks the
+// interrupt flag and keeps waiting (reference BasicSocket::recvExactT)
+inline bool recvExactInterruptible(int fd, char* buf, uint64_t len,
+                                   const std::atomic<bool>& interruptFlag)
+{
+    while (len) {
+        ssize_t n = recv(fd, buf, len, 0);
+        if (n <= 0) {
+            if (n < 0 && (errno == EINTR || errno == EAGAIN || errno == EWOULDBLOCK)) {
+                if (interruptFlag.load(std::memory_order_relaxed))
+                    throw InterruptedError();
+                continue;
+            }
+            return false;
+        }
+        buf += n;
+        len -= (uint64_t)n;
+    }
+    return true;
+}
+
+inline void setRecvTimeout(int fd, int secs)
+{
+    struct timeval tv{secs, 0};
+    setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+}
+
 } // namespace eb
