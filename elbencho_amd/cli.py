@@ -409,7 +409,10 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--s3nocompress", action="store_true", help=argparse.SUPPRESS)
 
     g = p.add_argument_group("misc")
-    g.add_argument("--hdfs", action="store_true", help=argparse.SUPPRESS)
+    g.add_argument("--hdfs", action="store_true",
+                   help="Benchmark HDFS via the WebHDFS REST protocol (bench "
+                        "paths: hdfs://namenode:port/base; no libhdfs/JVM "
+                        "needed).")
     g.add_argument("-c", "--configfile", default="", metavar="PATH",
                    help="Read options from a config file (key=value lines).")
     class _VersionAction(argparse.Action):
@@ -428,9 +431,9 @@ def _version_text() -> str:
     ProgArgs::printVersionAndBuildInfo, ProgArgs.cpp:3646-3740)."""
     from elbencho_amd import HTTP_PROTOCOL_VERSION
     included = ["hip/gfx950", "rccl-xgmi", "io_uring", "io_uring-fixedbufs",
-                "s3", "netbench", "corebind", "libnuma", "backtrace",
-                "althttpsvc", "mmap-zerocopy"]
-    excluded = ["cuda", "cufile/gds", "hdfs", "libaio", "s3crt", "mimalloc"]
+                "s3", "hdfs/webhdfs", "netbench", "corebind", "libnuma",
+                "backtrace", "althttpsvc", "mmap-zerocopy"]
+    excluded = ["cuda", "cufile/gds", "libaio", "s3crt", "mimalloc"]
     return (f"elbencho-amd\n"
             f" * Version: {VERSION} (MI355X/gfx950 native)\n"
             f" * Net protocol version: {HTTP_PROTOCOL_VERSION}\n"
@@ -638,10 +641,7 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
 
     cfg.config_file = args.configfile
 
-    if args.hdfs:
-        raise ConfigError("this build does not include Hadoop HDFS support "
-                          "(libhdfs is not available in the target image; "
-                          "matching the reference's optional-feature behavior)")
+    cfg.hdfs = args.hdfs
 
     cfg.finalize()
     return cfg

@@ -64,7 +64,8 @@ class BenchConfig:
     # --- paths & mode ---
     paths: list[str] = field(default_factory=list)
     path_type: str = PATH_FILE  # dir|file|bdev (auto-inferred)
-    bench_mode: str = "posix"  # posix|s3|netbench
+    bench_mode: str = "posix"  # posix|s3|hdfs|netbench
+    hdfs: bool = False             # --hdfs (WebHDFS REST engine)
 
     # --- phases ---
     run_mkdirs: bool = False       # -d/--mkdirs
@@ -285,6 +286,8 @@ class BenchConfig:
             self.bench_mode = "s3"
         elif self.s3_endpoints:
             self.bench_mode = "s3"
+        elif self.hdfs or (self.paths and self.paths[0].startswith("hdfs://")):
+            self.bench_mode = "hdfs"
         elif self.netbench:
             self.bench_mode = "netbench"
 
@@ -351,6 +354,12 @@ class BenchConfig:
                 raise ConfigError("--netbench requires service mode (--hosts)")
             if not self.servers:
                 raise ConfigError("--netbench requires --servers")
+        if self.bench_mode == "hdfs":
+            if not self.paths or not self.paths[0].startswith("hdfs://"):
+                raise ConfigError("HDFS mode requires an hdfs://namenode:port/base "
+                                  "bench path (WebHDFS)")
+            if (self.run_write or self.run_read) and self.files < 1:
+                raise ConfigError("HDFS read/write requires -N/--files >= 1")
         if self.bench_mode == "s3":
             if not self.s3_endpoints:
                 raise ConfigError("S3 mode requires --s3endpoints")

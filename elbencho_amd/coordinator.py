@@ -131,6 +131,9 @@ class Coordinator:
             if cfg.bench_mode == "s3":
                 from elbencho_amd.s3 import S3Runner
                 self.runner = S3Runner(cfg)
+            elif cfg.bench_mode == "hdfs":
+                from elbencho_amd.hdfs import HdfsRunner
+                self.runner = HdfsRunner(cfg)
             else:
                 self.runner = LocalRunner(cfg)
 
