@@ -342,3 +342,28 @@ def test_service_reuse_after_interrupt(services, tmp_path):
     res = run_master(["--hosts", hosts, "-w", "-r", "-t", "1", "-s", "1m",
                       "-b", "1m", "--verify", "2", str(tmp_path / "small")])
     assert res.returncode == 0, res.stdout + res.stderr
+
+
+def test_distributed_hdfs(services, tmp_path):
+    """Services route HDFS configs to the WebHDFS engine."""
+    from tests.webhdfsmock import WebHdfsHandler, start_mock
+
+    server, port = start_mock()
+    try:
+        hosts = ",".join(f"127.0.0.1:{p}" for p in services)
+        res = run_master(["--hosts", hosts, "-d", "-w", "-r", "-F", "-D",
+                          "-t", "2", "-n", "1", "-N", "2", "-s", "32k",
+                          "-b", "32k", "--verify", "6",
+                          f"hdfs://127.0.0.1:{port}/dbench"])
+        assert res.returncode == 0, res.stdout + res.stderr
+        for line in res.stdout.splitlines():
+            if "Files total" in line:
+                # 2 services x 2 threads x 1 dir x 2 files
+                assert line.split()[-1] == "8"
+                break
+        else:
+            raise AssertionError("no Files total row:\n" + res.stdout)
+        with WebHdfsHandler.store.lock:
+            assert not WebHdfsHandler.store.files
+    finally:
+        server.shutdown()
