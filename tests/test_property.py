@@ -96,3 +96,30 @@ def test_dir_mode_roundtrip(core, tmp_path_factory, threads, dirs, files,
     assert sum(r["entries"] for r in res) == dirs * files * threads
     run_phase(core, eng, "RMDIRS")
     assert not any(os.scandir(tmp_path))
+
+
+@settings(max_examples=200, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(v=st.integers(0, 2**63 - 1))
+def test_histogram_bucket_roundtrip(core, v):
+    """bucketLowerBound is the inverse of bucketIndex: every value falls in
+    the bucket whose bounds contain it, and bounds are monotonic."""
+    idx = None
+    # compute index via the native hook (exposed for tests)
+    lower = core.hist_bucket_lower_bound
+    n = core.hist_num_buckets()
+    # binary property: lower(i) <= v < lower(i+1) for the bucket v maps to
+    # (find the bucket by scanning bounds — bounds are strictly increasing)
+    lo = 0
+    hi = n - 1
+    while lo < hi:
+        mid = (lo + hi + 1) // 2
+        if lower(mid) <= v:
+            lo = mid
+        else:
+            hi = mid - 1
+    idx = lo
+    assert lower(idx) <= v
+    if idx + 1 < n:
+        assert v < lower(idx + 1)
+        assert lower(idx + 1) > lower(idx)
