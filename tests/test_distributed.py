@@ -151,3 +151,28 @@ def test_gloo_s3_mpu_sharing_two_ranks(tmp_path):
             assert sorted(parts) == list(range(1, 9))
     finally:
         server.shutdown()
+
+
+def test_bench_contract_two_ranks(tmp_path):
+    """bench.py under torch.distributed.run (gloo, CPU): rank 0 prints ONE
+    valid JSON line with the driver-contract fields; value aggregates both
+    ranks (whole-job), max-over-ranks timing."""
+    env = dict(os.environ, PYTHONPATH=REPO, MASTER_ADDR="127.0.0.1",
+               EB_DIST_BACKEND="gloo", EB_BENCH_DIR=str(tmp_path))
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29527", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--filesize", str(32 * 1024 * 1024)],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert res.returncode == 0, res.stdout + res.stderr
+    lines = [ln for ln in res.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1  # only rank 0 prints
+    doc = json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in doc, key
+    assert doc["config"]["parallelism"] == "dp2"
+    assert doc["value"] > 0
