@@ -204,25 +204,21 @@ def test_rank_offset_partitioning(core, tmp_path):
         assert core.verify_checksum(f.read(), 0, 6) == 2**64 - 1
 
 
-def test_write_to_readonly_dir_fails_loudly(core, tmp_path):
-    """Worker errors carry the failing path + errno and interrupt peers."""
-    import os as _os
-    ro = tmp_path / "ro"
-    ro.mkdir()
-    _os.chmod(ro, 0o555)
-    try:
-        cfg = dict(paths=[str(ro / "f")], path_type="file", threads=2,
-                   num_dataset_threads=2, file_size=1 << 20, block_size=1 << 20)
-        eng = core.Engine(cfg)
-        eng.prepare()
-        eng.start_phase(core.PHASES["WRITE"])
-        assert eng.wait_phase_done(60_000)
-        res = eng.finish_phase()
-        errs = [r["error"] for r in res if r["error"]]
-        assert errs, "expected open failure"
-        assert any("Permission denied" in e or "open" in e for e in errs)
-    finally:
-        _os.chmod(ro, 0o755)
+def test_write_to_invalid_parent_fails_loudly(core, tmp_path):
+    """Worker errors carry the failing path + errno and stop the phase
+    (running as root, so ENOTDIR stands in for permission failures)."""
+    blocker = tmp_path / "afile"
+    blocker.write_bytes(b"x")
+    cfg = dict(paths=[str(blocker / "f")], path_type="file", threads=2,
+               num_dataset_threads=2, file_size=1 << 20, block_size=1 << 20)
+    eng = core.Engine(cfg)
+    eng.prepare()
+    eng.start_phase(core.PHASES["WRITE"])
+    assert eng.wait_phase_done(60_000)
+    res = eng.finish_phase()
+    errs = [r["error"] for r in res if r["error"]]
+    assert errs, "expected open failure"
+    assert any("open" in e and "afile" in e for e in errs)
 
 
 def test_read_missing_file_fails_loudly(core, tmp_path):
