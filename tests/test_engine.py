@@ -205,20 +205,16 @@ def test_rank_offset_partitioning(core, tmp_path):
 
 
 def test_write_to_invalid_parent_fails_loudly(core, tmp_path):
-    """Worker errors carry the failing path + errno and stop the phase
-    (running as root, so ENOTDIR stands in for permission failures)."""
+    """An unusable bench path is rejected at prepare() with the failing
+    path and errno in the message (running as root, ENOTDIR stands in for
+    permission failures)."""
     blocker = tmp_path / "afile"
     blocker.write_bytes(b"x")
     cfg = dict(paths=[str(blocker / "f")], path_type="file", threads=2,
                num_dataset_threads=2, file_size=1 << 20, block_size=1 << 20)
     eng = core.Engine(cfg)
-    eng.prepare()
-    eng.start_phase(core.PHASES["WRITE"])
-    assert eng.wait_phase_done(60_000)
-    res = eng.finish_phase()
-    errs = [r["error"] for r in res if r["error"]]
-    assert errs, "expected open failure"
-    assert any("open" in e and "afile" in e for e in errs)
+    with pytest.raises(RuntimeError, match="Not a directory"):
+        eng.prepare()
 
 
 def test_read_missing_file_fails_loudly(core, tmp_path):
