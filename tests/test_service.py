@@ -367,3 +367,20 @@ def test_distributed_hdfs(services, tmp_path):
             assert not WebHdfsHandler.store.files
     finally:
         server.shutdown()
+
+
+def test_protocol_version_mismatch_rejected(services):
+    """A master with a different wire version is refused at /preparephase
+    (reference protocol-version check, HTTPServiceSWS.cpp:268)."""
+    import http.client
+
+    port = services[0]
+    conn = http.client.HTTPConnection("127.0.0.1", port, timeout=10)
+    body = json.dumps({"protocol_version": "99.0.0", "config": {}}).encode()
+    conn.request("POST", "/preparephase", body=body,
+                 headers={"Content-Type": "application/json"})
+    resp = conn.getresponse()
+    data = resp.read()
+    conn.close()
+    assert resp.status == 400
+    assert b"protocol version mismatch" in data
