@@ -561,6 +561,15 @@ class S3Worker(threading.Thread):
         self.error = ""
         self.elapsed_us = 0
         cfg = runner.cfg
+        # --rwmixthr: the first K threads of a WRITE phase read instead
+        # (reference s3ModeIterateObjects isRWMixedReader); their results go
+        # to the rwmix-read counter set
+        self.is_mix_reader = (phase == "WRITE" and
+                              local_rank < cfg._rwmix_threads_effective())
+        self.rm_ops = _Counters()
+        self.rm_sw: Optional[_Counters] = None
+        self.io_lat_rm = Histogram()
+        self.entry_lat_rm = Histogram()
         ep = cfg.s3_endpoints[self.rank % len(cfg.s3_endpoints)]
         key, secret = cfg.s3_key, cfg.s3_secret
         if runner.credentials:  # --s3credfile/--s3credlist round-robin
@@ -686,7 +695,21 @@ class S3Worker(threading.Thread):
                     self.client.delete_bucket(b)
                     self.ops.entries += 1
         elif ph == "WRITE":
-            self._put_objects()
+            if self.is_mix_reader:
+                # account into the rwmix-read counters via a counter swap
+                # (the read loop only touches ops/io_lat/entry_lat)
+                self.ops, self.rm_ops = self.rm_ops, self.ops
+                self.io_lat, self.io_lat_rm = self.io_lat_rm, self.io_lat
+                self.entry_lat, self.entry_lat_rm = self.entry_lat_rm, self.entry_lat
+                try:
+                    self._get_objects()
+                finally:
+                    self.ops, self.rm_ops = self.rm_ops, self.ops
+                    self.io_lat, self.io_lat_rm = self.io_lat_rm, self.io_lat
+                    self.entry_lat, self.entry_lat_rm = \
+                        self.entry_lat_rm, self.entry_lat
+            else:
+                self._put_objects()
         elif ph == "READ":
             self._get_objects()
         elif ph == "STAT":  # HEADOBJ
@@ -1344,10 +1367,13 @@ class S3Runner:
     def on_worker_done(self, w: S3Worker) -> None:
         with self.done_cv:
             if not self.stonewalled and not w.error and (
-                    w.ops.bytes or w.ops.entries or w.ops.iops):
+                    w.ops.bytes or w.ops.entries or w.ops.iops or
+                    w.rm_ops.bytes or w.rm_ops.iops):
                 elapsed = int((time.monotonic() - self.phase_start) * 1e6)
                 for peer in self.workers:
                     peer.sw = _Counters(peer.ops.entries, peer.ops.bytes, peer.ops.iops)
+                    peer.rm_sw = _Counters(peer.rm_ops.entries, peer.rm_ops.bytes,
+                                           peer.rm_ops.iops)
                     peer.sw_elapsed_us = elapsed
                 self.stonewalled = True
             self.done_count += 1
@@ -1378,12 +1404,34 @@ class S3Runner:
     def interrupt(self) -> None:
         self.interrupt_flag.set()
 
+    def trigger_stonewall(self) -> None:
+        """Remote stonewall propagation (/triggerstonewall): snapshot every
+        worker's live counters now (reference RemoteWorker poll loop)."""
+        with self.done_cv:
+            if not self.stonewalled:
+                elapsed = int((time.monotonic() - self.phase_start) * 1e6)
+                for peer in self.workers:
+                    peer.sw = _Counters(peer.ops.entries, peer.ops.bytes,
+                                        peer.ops.iops)
+                    peer.rm_sw = _Counters(peer.rm_ops.entries,
+                                           peer.rm_ops.bytes, peer.rm_ops.iops)
+                    peer.sw_elapsed_us = elapsed
+                self.stonewalled = True
+
+    def poll_workers(self):
+        """Per-worker rows for the fullscreen dashboard / --livecsvex."""
+        return [{"rank": w.rank,
+                 "entries": w.ops.entries + w.rm_ops.entries,
+                 "bytes": w.ops.bytes + w.rm_ops.bytes,
+                 "iops": w.ops.iops + w.rm_ops.iops} for w in self.workers]
+
     def finish(self) -> list[WorkerStats]:
         for w in self.workers:
             w.join()
         out = []
         for w in self.workers:
             sw = w.sw or w.ops
+            rm_sw = w.rm_sw or w.rm_ops
             out.append(WorkerStats(
                 rank=w.rank,
                 elapsed_usec=w.elapsed_us,
@@ -1391,7 +1439,13 @@ class S3Runner:
                 stonewall_elapsed_usec=w.sw_elapsed_us or w.elapsed_us,
                 stonewall_entries=sw.entries, stonewall_bytes=sw.bytes,
                 stonewall_iops=sw.iops,
+                rm_entries=w.rm_ops.entries, rm_bytes=w.rm_ops.bytes,
+                rm_iops=w.rm_ops.iops,
+                rm_stonewall_entries=rm_sw.entries,
+                rm_stonewall_bytes=rm_sw.bytes, rm_stonewall_iops=rm_sw.iops,
                 io_lat=list(w.io_lat.vec), entry_lat=list(w.entry_lat.vec),
+                io_lat_rm=list(w.io_lat_rm.vec),
+                entry_lat_rm=list(w.entry_lat_rm.vec),
                 error=w.error))
         return out
 

@@ -394,3 +394,26 @@ def test_s3_worker_error_does_not_hang_pipeline(mock_s3):
     rc = _cli(mock_s3, ["-r", "-t", "1", "-N", "2", "-s", "64k", "-b", "16k",
                         "--iodepth", "4", "s3://errbkt"])
     assert rc != 0  # r0-f1 is missing -> phase fails, process exits promptly
+
+
+def test_s3_rwmix_dedicated_readers(mock_s3, capsys):
+    """--rwmixthr in the S3 WRITE phase: the first K threads download the
+    pre-written objects while the rest upload (reference
+    s3ModeIterateObjects isRWMixedReader); read results appear in the rwmix
+    read columns."""
+    # pre-write the reader threads' objects (mix reads need existing data)
+    rc = _cli(mock_s3, ["-d", "-w", "-t", "4", "-N", "2", "-s", "64k",
+                        "-b", "64k", "--verify", "4", "s3://mixbkt"])
+    assert rc == 0
+    capsys.readouterr()  # drop the pre-write output
+    rc = _cli(mock_s3, ["-w", "-t", "4", "--rwmixthr", "2", "-N", "2",
+                        "-s", "64k", "-b", "64k", "--verify", "4",
+                        "s3://mixbkt"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "MiB/s read" in out  # rwmix read rows present
+    for line in out.splitlines():
+        if "Objects total" in line:
+            # 2 writer threads x 2 objects written; readers tracked separately
+            assert line.split()[-1] == "4"
+            break
