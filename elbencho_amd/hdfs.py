@@ -150,6 +150,14 @@ class HdfsWorker(threading.Thread):
         self.entry_lat = Histogram()
         self.error = ""
         self.elapsed_us = 0
+        # --rwmixthr parity with the reference's hdfs isRWMixedReader
+        # (LocalWorker.cpp:7631): first K threads of a WRITE phase read
+        self.is_mix_reader = (phase == "WRITE" and
+                              local_rank < runner.cfg._rwmix_threads_effective())
+        self.rm_ops = _Counters()
+        self.rm_sw: Optional[_Counters] = None
+        self.io_lat_rm = Histogram()
+        self.entry_lat_rm = Histogram()
         host, port, _ = runner.endpoint
         self.client = WebHdfsClient(host, port)
         self.core = load_core()
@@ -216,6 +224,19 @@ class HdfsWorker(threading.Thread):
                 self.ops.entries += 1
             # rank dir itself
             self.client.delete(f"{base}/r{self.rank}", recursive=True)
+        elif ph == "WRITE" and self.is_mix_reader:
+            self.ops, self.rm_ops = self.rm_ops, self.ops
+            self.io_lat, self.io_lat_rm = self.io_lat_rm, self.io_lat
+            self.entry_lat, self.entry_lat_rm = self.entry_lat_rm, self.entry_lat
+            try:
+                self.phase = "READ"
+                self._run_phase()
+            finally:
+                self.phase = "WRITE"
+                self.ops, self.rm_ops = self.rm_ops, self.ops
+                self.io_lat, self.io_lat_rm = self.io_lat_rm, self.io_lat
+                self.entry_lat, self.entry_lat_rm = \
+                    self.entry_lat_rm, self.entry_lat
         elif ph == "WRITE":
             size, bs = cfg.file_size, cfg.block_size
             for path in self._file_paths():
@@ -321,11 +342,14 @@ class HdfsRunner:
     def on_worker_done(self, w: HdfsWorker) -> None:
         with self.done_cv:
             if not self.stonewalled and not w.error and (
-                    w.ops.bytes or w.ops.entries or w.ops.iops):
+                    w.ops.bytes or w.ops.entries or w.ops.iops or
+                    w.rm_ops.bytes or w.rm_ops.iops):
                 elapsed = int((time.monotonic() - self.phase_start) * 1e6)
                 for peer in self.workers:
                     peer.sw = _Counters(peer.ops.entries, peer.ops.bytes,
                                         peer.ops.iops)
+                    peer.rm_sw = _Counters(peer.rm_ops.entries,
+                                           peer.rm_ops.bytes, peer.rm_ops.iops)
                     peer.sw_elapsed_us = elapsed
                 self.stonewalled = True
             self.done_count += 1
@@ -347,9 +371,9 @@ class HdfsRunner:
                "lat_num_ios": 0, "lat_sum_ios": 0, "lat_num_entries": 0,
                "lat_sum_entries": 0}
         for w in self.workers:
-            agg["entries"] += w.ops.entries
-            agg["bytes"] += w.ops.bytes
-            agg["iops"] += w.ops.iops
+            agg["entries"] += w.ops.entries + w.rm_ops.entries
+            agg["bytes"] += w.ops.bytes + w.rm_ops.bytes
+            agg["iops"] += w.ops.iops + w.rm_ops.iops
         return agg
 
     def poll_workers(self):
@@ -374,13 +398,20 @@ class HdfsRunner:
         out = []
         for w in self.workers:
             sw = w.sw or _Counters()
+            rm_sw = w.rm_sw or w.rm_ops
             out.append(WorkerStats(
                 rank=w.rank, elapsed_usec=w.elapsed_us,
                 entries=w.ops.entries, bytes=w.ops.bytes, iops=w.ops.iops,
                 stonewall_elapsed_usec=w.sw_elapsed_us or w.elapsed_us,
                 stonewall_entries=sw.entries, stonewall_bytes=sw.bytes,
                 stonewall_iops=sw.iops,
+                rm_entries=w.rm_ops.entries, rm_bytes=w.rm_ops.bytes,
+                rm_iops=w.rm_ops.iops,
+                rm_stonewall_entries=rm_sw.entries,
+                rm_stonewall_bytes=rm_sw.bytes, rm_stonewall_iops=rm_sw.iops,
                 io_lat=list(w.io_lat.vec), entry_lat=list(w.entry_lat.vec),
+                io_lat_rm=list(w.io_lat_rm.vec),
+                entry_lat_rm=list(w.entry_lat_rm.vec),
                 error=w.error))
         return out
 

@@ -85,3 +85,17 @@ def test_hdfs_requires_hdfs_path(tmp_path):
     rc = main(["--hdfs", "-w", "-t", "1", "-N", "1", "-s", "4k", "--nolive",
                str(tmp_path)])
     assert rc != 0  # clear config error, not a crash
+
+
+def test_hdfs_rwmix_dedicated_readers(mock_hdfs, capsys):
+    rc = main(["-d", "-w", "-t", "2", "-N", "2", "-s", "32k", "-b", "32k",
+               "--verify", "4", "--nolive",
+               f"hdfs://127.0.0.1:{mock_hdfs}/mix"])
+    assert rc == 0
+    capsys.readouterr()
+    rc = main(["-w", "-t", "2", "--rwmixthr", "1", "-N", "2", "-s", "32k",
+               "-b", "32k", "--verify", "4", "--nolive",
+               f"hdfs://127.0.0.1:{mock_hdfs}/mix"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "MiB/s read" in out or "IOPS read" in out
