@@ -124,3 +124,20 @@ def test_histogram_bucket_roundtrip(core, v):
     if idx + 1 < n:
         assert v < lower(idx + 1)
         assert lower(idx + 1) > lower(idx)
+
+
+@settings(max_examples=80, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(nblocks=st.integers(1, 5000), seed=st.integers(0, 2**31),
+       tail=st.integers(0, 4095))
+def test_full_coverage_lcg_property(core, nblocks, seed, tail):
+    """The full-coverage random-aligned generator (Hull-Dobell LCG +
+    cycle-walking) visits every block exactly once for ANY block count and
+    tail size, not just the handwritten cases."""
+    bs = 4096
+    offs, total = core.gen_offsets("full_coverage", bs, 0,
+                                   bs * nblocks + tail, seed=seed)
+    starts = sorted(o for o, _ in offs)
+    expected_blocks = nblocks + (1 if tail else 0)
+    assert starts == [bs * i for i in range(expected_blocks)]
+    assert total == sum(ln for _, ln in offs)
