@@ -141,3 +141,43 @@ def test_full_coverage_lcg_property(core, nblocks, seed, tail):
     expected_blocks = nblocks + (1 if tail else 0)
     assert starts == [bs * i for i in range(expected_blocks)]
     assert total == sum(ln for _, ln in offs)
+
+
+@settings(max_examples=100, deadline=None)
+@given(n=st.integers(0, 2**40), suffix=st.sampled_from(
+    ["", "k", "K", "m", "M", "g", "G", "kb", "KB", "mib", "MiB", "gib"]))
+def test_parse_size_suffixes(n, suffix):
+    from elbencho_amd.units import parse_size
+    mult = {"": 1, "k": 1 << 10, "m": 1 << 20, "g": 1 << 30}[
+        suffix[:1].lower() if suffix else ""]
+    assert parse_size(f"{n}{suffix}") == n * mult
+
+
+@settings(max_examples=50, deadline=None)
+@given(threads=st.integers(1, 64), bs=st.integers(1, 2**30),
+       label=st.text(alphabet=st.characters(codec="ascii",
+                                            exclude_characters="\x00"),
+                     max_size=24),
+       salt=st.integers(-1, 2**31))
+def test_config_wire_roundtrip(threads, bs, label, salt):
+    """to_wire -> JSON -> from_wire preserves every benchmark-relevant field
+    (the config system IS the wire schema)."""
+    import dataclasses
+    import json as _json
+
+    from elbencho_amd.config import BenchConfig
+    cfg = BenchConfig()
+    cfg.threads = threads
+    cfg.block_size = bs
+    cfg.label = label
+    cfg.verify = salt
+    cfg.gpu_ids = [0, 3]
+    cfg.s3_mpu_upload_ids = {"b/k": "id-1"}
+    wire = _json.loads(_json.dumps(cfg.to_wire()))
+    back = BenchConfig.from_wire(wire)
+    for f in dataclasses.fields(BenchConfig):
+        if f.name in ("hosts", "service_mode", "quit_services",
+                      "interrupt_services", "csv_file", "json_file",
+                      "res_file", "live_csv", "config_file"):
+            continue  # master-only, intentionally not on the wire
+        assert getattr(back, f.name) == getattr(cfg, f.name), f.name
