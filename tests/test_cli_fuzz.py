@@ -1,0 +1,67 @@
+"""CLI robustness fuzz: random combinations of posix-mode flags must either
+run cleanly or fail with a framed ERROR (rc != 0) — never an unhandled
+traceback. Deterministic seeds so failures reproduce."""
+
+import random
+
+import pytest
+
+from elbencho_amd.cli import main
+
+# (flag template, needs_value) — safe subset: no services/hosts, no external
+# endpoints, no infinite loops, tiny sizes, bounded time
+FLAG_POOL = [
+    ("-w", None), ("-r", None), ("--stat", None), ("-F", None),
+    ("-d", None), ("-D", None),
+    ("-t", ["1", "2", "3"]),
+    ("-n", ["0", "1", "2"]),
+    ("-N", ["0", "1", "2"]),
+    ("-s", ["0", "1", "4k", "64k", "1m", "100"]),
+    ("-b", ["1", "512", "4k", "64k", "1m"]),
+    ("--iodepth", ["1", "2", "8"]),
+    ("-i", ["1", "2"]),
+    ("--rand", None), ("--norandalign", None), ("--backward", None),
+    ("--strided", None), ("--randamount", ["0", "64k"]),
+    ("--randalgo", ["fast", "balanced", "balanced_single", "strong"]),
+    ("--trunc", None), ("--trunctosize", None), ("--preallocfile", None),
+    ("--verify", ["0", "7"]), ("--readinline", None), ("--statinline", None),
+    ("--mmap", None), ("--fadv", ["seq", "rand,willneed"]),
+    ("--flock", ["range", "full"]),
+    ("--blockvarpct", ["0", "50", "100"]),
+    ("--blockvaralgo", ["fast", "balanced", "strong"]),
+    ("--lat", None), ("--lathisto", None), ("--latpercent", None),
+    ("--allelapsed", None), ("--cpu", None), ("--dirstats", None),
+    ("--base10", None), ("--rwmixpct", ["0", "30"]),
+    ("--rwmixthr", ["0", "1"]),
+    ("--nodelerr", None), ("--no0usecerr", None), ("--nodiocheck", None),
+    ("--dirsharing", None), ("--sharesize", ["0", "4k"]),
+    ("--label", ["fuzz"]), ("--dryrun", None),
+]
+
+
+@pytest.mark.parametrize("seed", range(40))
+def test_fuzz_flag_combinations(tmp_path, seed, capsys):
+    rng = random.Random(seed)
+    argv = ["--nolive", "--timelimit", "20"]
+    for flag, values in FLAG_POOL:
+        if rng.random() < 0.25:
+            argv.append(flag)
+            if values:
+                argv.append(rng.choice(values))
+    # a path type at random: file, dir, or multiple files
+    kind = rng.randrange(3)
+    if kind == 0:
+        argv.append(str(tmp_path / "f1"))
+    elif kind == 1:
+        argv.append(str(tmp_path))
+    else:
+        argv += [str(tmp_path / "f1"), str(tmp_path / "f2")]
+
+    rc = main(argv)  # the property: never an unhandled traceback
+    assert isinstance(rc, int)
+    if rc != 0:
+        cap = capsys.readouterr()
+        # a diagnostic lands on stderr (config errors) or stdout (phase
+        # ERROR rows in the results table)
+        assert (cap.err + cap.out).strip(), (argv,
+                                             "non-zero exit with no diagnostic")
