@@ -65,3 +65,45 @@ def test_fuzz_flag_combinations(tmp_path, seed, capsys):
         # ERROR rows in the results table)
         assert (cap.err + cap.out).strip(), (argv,
                                              "non-zero exit with no diagnostic")
+
+
+S3_FLAG_POOL = [
+    ("-w", None), ("-r", None), ("--stat", None), ("-F", None),
+    ("-d", None), ("-D", None),
+    ("-t", ["1", "2"]), ("-n", ["0", "1"]), ("-N", ["1", "2"]),
+    ("-s", ["0", "4k", "64k", "192k"]), ("-b", ["4k", "64k"]),
+    ("--iodepth", ["1", "4"]),
+    ("--verify", ["0", "5"]), ("--s3fastget", None), ("--s3fastput", None),
+    ("--s3listobj", ["0", "10"]), ("--s3listverify", None),
+    ("--s3multidel", ["0", "2"]), ("--s3randobj", None),
+    ("--s3objprefix", ["p/"]), ("--s3sign", ["0", "2"]),
+    ("--s3chksumalgo", ["CRC32", "SHA256"]), ("--s3sse", None),
+    ("--s3aclput", None), ("--s3aclget", None), ("--s3statdirs", None),
+    ("--s3bversion", None), ("--s3olockcfg", None), ("--s3listobjpar", None),
+    ("--lat", None), ("--rwmixthr", ["0", "1"]),
+]
+
+
+@pytest.mark.parametrize("seed", range(20))
+def test_fuzz_s3_flag_combinations(seed, capsys):
+    from tests.s3mock import ACCESS_KEY, SECRET_KEY, start_mock
+
+    server, port = start_mock()
+    try:
+        rng = random.Random(seed)
+        argv = ["--nolive", "--timelimit", "30",
+                "--s3endpoints", f"http://127.0.0.1:{port}",
+                "--s3key", ACCESS_KEY, "--s3secret", SECRET_KEY]
+        for flag, values in S3_FLAG_POOL:
+            if rng.random() < 0.3:
+                argv.append(flag)
+                if values:
+                    argv.append(rng.choice(values))
+        argv.append(f"s3://fuzzbkt{seed}")
+        rc = main(argv)
+        assert isinstance(rc, int)
+        if rc != 0:
+            cap = capsys.readouterr()
+            assert (cap.err + cap.out).strip(), (argv, "silent nonzero")
+    finally:
+        server.shutdown()
