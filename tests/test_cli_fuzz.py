@@ -107,3 +107,32 @@ def test_fuzz_s3_flag_combinations(seed, capsys):
             assert (cap.err + cap.out).strip(), (argv, "silent nonzero")
     finally:
         server.shutdown()
+
+
+@pytest.mark.parametrize("seed", range(10))
+def test_fuzz_hdfs_flag_combinations(seed, capsys):
+    from tests.webhdfsmock import start_mock
+
+    server, port = start_mock()
+    try:
+        rng = random.Random(seed)
+        argv = ["--nolive", "--timelimit", "30"]
+        for flag, values in [
+                ("-w", None), ("-r", None), ("--stat", None), ("-F", None),
+                ("-d", None), ("-D", None), ("-t", ["1", "2"]),
+                ("-n", ["0", "1", "2"]), ("-N", ["1", "2"]),
+                ("-s", ["0", "4k", "96k"]), ("-b", ["4k", "32k"]),
+                ("--verify", ["0", "3"]), ("--lat", None),
+                ("--rwmixthr", ["0", "1"]), ("--nodelerr", None)]:
+            if rng.random() < 0.35:
+                argv.append(flag)
+                if values:
+                    argv.append(rng.choice(values))
+        argv.append(f"hdfs://127.0.0.1:{port}/fz{seed}")
+        rc = main(argv)
+        assert isinstance(rc, int)
+        if rc != 0:
+            cap = capsys.readouterr()
+            assert (cap.err + cap.out).strip(), (argv, "silent nonzero")
+    finally:
+        server.shutdown()
