@@ -384,3 +384,34 @@ def test_protocol_version_mismatch_rejected(services):
     conn.close()
     assert resp.status == 400
     assert b"protocol version mismatch" in data
+
+
+def test_daemonized_service_lifecycle(tmp_path):
+    """Default (non-foreground) service: double-fork daemonize, serve, obey
+    a master --quit (the suite otherwise only covers --foreground)."""
+    port = free_port()
+    env = dict(os.environ, PYTHONPATH=REPO, TMPDIR=str(tmp_path))
+    # parent exits immediately after daemonizing
+    res = subprocess.run([sys.executable, "-m", "elbencho_amd", "--service",
+                          "--port", str(port)], env=env, capture_output=True,
+                         text=True, timeout=60, cwd=str(tmp_path))
+    assert res.returncode == 0, res.stdout + res.stderr
+    try:
+        deadline = time.monotonic() + 30
+        while True:
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{port}/info", timeout=1) as r:
+                    assert b"elbencho-amd service" in r.read()
+                break
+            except OSError:
+                assert time.monotonic() < deadline, "daemon never came up"
+                time.sleep(0.2)
+        bench = run_master(["--hosts", f"127.0.0.1:{port}", "-w", "-r",
+                            "-t", "1", "-s", "64k", "-b", "64k",
+                            str(tmp_path / "f1")])
+        assert bench.returncode == 0, bench.stdout + bench.stderr
+    finally:
+        subprocess.run([sys.executable, "-m", "elbencho_amd",
+                        "--hosts", f"127.0.0.1:{port}", "--quit"],
+                       env=env, capture_output=True, timeout=30)
