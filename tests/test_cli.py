@@ -180,3 +180,32 @@ def test_phasedelay(tmp_path):
                "--phasedelay", "1", str(f)])
     assert rc == 0
     assert _time.time() - t0 >= 1.0  # delay between WRITE and READ
+
+
+def test_file_list_options(tmp_path):
+    """--hostsfile / --serversfile / --s3credfile file parsing."""
+    from elbencho_amd.cli import args_to_config, build_parser
+    p = build_parser()
+
+    hf = tmp_path / "hosts.txt"
+    hf.write_text("# comment\nnode1:1611\nnode2:1611\n\n")
+    cfg = args_to_config(p.parse_args(
+        ["--hostsfile", str(hf), "-w", "-s", "1m", str(tmp_path / "f")]))
+    assert cfg.hosts == ["node1:1611", "node2:1611"]
+
+    sf = tmp_path / "servers.txt"
+    sf.write_text("srv1\nsrv2\n")
+    cfg = args_to_config(p.parse_args(
+        ["--netbench", "-w", "-s", "1m", "-b", "64k",
+         "--hosts", "h1,h2,h3", "--serversfile", str(sf)]))
+    assert cfg.servers == ["srv1", "srv2"]
+
+    credf = tmp_path / "creds.txt"
+    credf.write_text("# creds\nkeyA:secretA\nkeyB:secretB\n")
+    cfg = args_to_config(p.parse_args(
+        ["--s3endpoints", "http://x:1", "-w", "-N", "1", "-s", "4k",
+         "--s3credfile", str(credf), "s3://b"]))
+    assert cfg.s3_cred_file == str(credf)
+    from elbencho_amd.s3 import S3Runner
+    runner = S3Runner(cfg)
+    assert runner.credentials == [("keyA", "secretA"), ("keyB", "secretB")]
