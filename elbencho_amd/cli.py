@@ -647,6 +647,31 @@ def args_to_config(args: argparse.Namespace) -> BenchConfig:
     return cfg
 
 
+def _intercept_bool_overrides(parser: argparse.ArgumentParser,
+                              argv: list[str]) -> list[str]:
+    """Reference bool-override interception (ProgArgs.cpp:1053): a flag
+    followed by an explicit "false" on the command line clears the flag a
+    config file turned on ("--direct false"); "true" keeps it."""
+    bool_flags = {s for a in parser._actions
+                  if isinstance(a, argparse._StoreTrueAction)
+                  for s in a.option_strings}
+    out: list[str] = []
+    i = 0
+    while i < len(argv):
+        tok = argv[i]
+        if tok in bool_flags and i + 1 < len(argv) and \
+                argv[i + 1].lower() in ("true", "false"):
+            if argv[i + 1].lower() == "true":
+                out.append(tok)
+            else:  # "false": drop this AND any earlier occurrence
+                out = [t for t in out if t != tok]
+            i += 2
+            continue
+        out.append(tok)
+        i += 1
+    return out
+
+
 def apply_config_file(argv: list[str]) -> list[str]:
     """Prepend options from -c/--configfile (key=value lines) to argv."""
     cfgpath = None
@@ -716,6 +741,7 @@ def main(argv: list[str] | None = None) -> int:
             return 0
     argv = apply_config_file(argv)
     parser = build_parser()
+    argv = _intercept_bool_overrides(parser, argv)
     args = parser.parse_args(argv)
 
     try:
