@@ -31,8 +31,6 @@ def core():
 )
 def test_file_mode_roundtrip(core, tmp_path_factory, threads, file_size,
                              block_kib, pattern, num_files, iodepth):
-    if pattern == "backward" and iodepth > 1:
-        iodepth = 1  # async engine orders by completion; backward is sync-only
     tmp_path = tmp_path_factory.mktemp("prop")
     paths = [str(tmp_path / f"f{i}") for i in range(num_files)]
     bs = block_kib * 1024
